@@ -1,0 +1,31 @@
+"""alpa_amd — MI355X-native auto-parallelization framework.
+
+A from-scratch, MI355X-first framework with the capabilities of Alpa
+(alpa-projects/alpa): automatic intra-operator sharding (ILP), inter-operator
+pipeline parallelism (DP stage slicing, 1F1B), ZeRO, MoE/expert parallelism,
+checkpointing, and auto-sharded serving — built on PyTorch-ROCm +
+hand-written CDNA4 (gfx950) HIP kernels + RCCL collectives over xGMI, one
+process per GPU.  No JAX/XLA, no Ray, no CUDA shims.
+"""
+
+from .api import TrainState, init, parallelize, shutdown
+from .global_env import global_config
+from .mesh import (DeviceMesh, VirtualMesh, device, full_mesh, full_virtual_mesh,
+                   get_device_mesh, init_distributed, local_rank, rank,
+                   world_size)
+from .optim import AdamW
+from .parallel_method import (AutoShardingOption, DataParallel,
+                              ParallelMethod, PipeshardParallel, ShardParallel,
+                              Zero2Parallel, Zero3Parallel,
+                              get_3d_parallel_method)
+
+__version__ = "0.1.0"
+
+__all__ = [
+    "init", "shutdown", "parallelize", "TrainState", "AdamW",
+    "DeviceMesh", "VirtualMesh", "device", "full_mesh", "full_virtual_mesh",
+    "get_device_mesh", "init_distributed", "local_rank", "rank", "world_size",
+    "ParallelMethod", "ShardParallel", "DataParallel", "Zero2Parallel",
+    "Zero3Parallel", "PipeshardParallel", "AutoShardingOption",
+    "get_3d_parallel_method", "global_config",
+]

@@ -1,0 +1,77 @@
+"""Global configuration singleton.
+
+MI355X-native analog of the reference's ``alpa/global_env.py:5`` GlobalConfig:
+a single mutable options object read throughout the framework, with env-var
+overrides.  Unlike the reference there is no Ray worker push — every rank
+constructs the same config from its own environment (one process per GPU).
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass, field
+
+
+def _env_bool(name: str, default: bool) -> bool:
+    v = os.environ.get(name)
+    if v is None:
+        return default
+    return v.lower() not in ("0", "false", "no", "off")
+
+
+@dataclass
+class GlobalConfig:
+    # ---------- backend ----------
+    #: "hip" on a GPU box, "cpu" for GPU-free tests (gloo collectives).
+    backend: str = "hip"
+    #: torch.distributed backend name. "nccl" IS RCCL on ROCm.
+    dist_backend: str = "auto"  # auto -> nccl if cuda available else gloo
+
+    # ---------- compute ----------
+    #: default compute dtype for models/benchmarks
+    compute_dtype: str = "bfloat16"
+    #: use hand-written HIP kernels when their extension is available
+    use_hip_kernels: bool = True
+    #: fail loudly if running on GPU without the HIP extension (anti-silent-fallback)
+    require_hip_kernels_on_gpu: bool = True
+    #: capture steady-state step in a hipGraph when possible
+    use_hip_graphs: bool = False
+
+    # ---------- collectives ----------
+    #: bucket size for gradient all-reduce / reduce-scatter (bytes).
+    #: xGMI rings are per-link bound (~153 GB/s); larger buckets amortize
+    #: launch + ring latency. 100 MiB default.
+    grad_bucket_bytes: int = 100 * 1024 * 1024
+    #: overlap grad collectives with backward
+    overlap_grad_sync: bool = True
+    #: alpha (s) and beta (s/byte) for the mesh cost model, calibrated for
+    #: single-node xGMI (ring all-reduce is bound by one ~153 GB/s link).
+    mesh_alpha: float = 1e-5
+    mesh_beta: float = 1.0 / (150e9)
+
+    # ---------- auto-sharding ----------
+    #: ILP time limit (s) for scipy/HiGHS
+    solver_timeout: float = 600.0
+    #: memory budget fraction of HBM used by the ILP memory constraint
+    memory_fraction: float = 0.9
+    #: print chosen strategies (debug)
+    print_strategy: bool = _env_bool("ALPA_AMD_DEBUG_PRINT_STRATEGY", False)
+
+    # ---------- pipeline ----------
+    pipeline_check_alive: bool = True
+    collect_trace: bool = False
+
+    # ---------- benchmark ----------
+    use_dummy_value_for_benchmarking: bool = False
+
+    # ---------- paths ----------
+    prof_database_path: str = field(
+        default_factory=lambda: os.environ.get("ALPA_AMD_PROF_DB", "prof_database.pkl"))
+
+    def resolved_dist_backend(self) -> str:
+        if self.dist_backend != "auto":
+            return self.dist_backend
+        import torch
+        return "nccl" if torch.cuda.is_available() else "gloo"
+
+
+global_config = GlobalConfig()
