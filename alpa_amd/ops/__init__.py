@@ -149,13 +149,54 @@ def softmax_cross_entropy(logits: torch.Tensor,
     return _SoftmaxCrossEntropy.apply(logits, targets)
 
 
+_ADAM_CHUNK = 16384  # must match ADAM_CHUNK in csrc/adamw.hip
+_ADAM_TABLE_CACHE: dict = {}
+
+
+def _adam_tables(params, grads, exp_avgs, exp_avg_sqs):
+    """Build (and cache) the device-side tensor-descriptor + chunk tables
+    consumed by the multi-tensor AdamW kernel (one launch per step)."""
+    key = tuple(p.data_ptr() for p in params) + \
+        tuple(g.data_ptr() for g in grads)
+    hit = _ADAM_TABLE_CACHE.get(key)
+    if hit is not None:
+        return hit
+    n = len(params)
+    descs = torch.empty(n, 6, dtype=torch.int64)
+    counts = []
+    for i, (p, g, m, v) in enumerate(zip(params, grads, exp_avgs,
+                                         exp_avg_sqs)):
+        assert p.is_contiguous() and g.is_contiguous()
+        assert m.dtype == torch.float32 and v.dtype == torch.float32
+        assert p.dtype == g.dtype and p.dtype in (torch.bfloat16,
+                                                  torch.float32)
+        descs[i, 0] = p.data_ptr()
+        descs[i, 1] = g.data_ptr()
+        descs[i, 2] = m.data_ptr()
+        descs[i, 3] = v.data_ptr()
+        descs[i, 4] = p.numel()
+        descs[i, 5] = 1 if p.dtype == torch.bfloat16 else 0
+        counts.append((p.numel() + _ADAM_CHUNK - 1) // _ADAM_CHUNK)
+    tensor_idx = torch.repeat_interleave(
+        torch.arange(n, dtype=torch.int32),
+        torch.tensor(counts, dtype=torch.int64))
+    chunk_idx = torch.cat([torch.arange(c, dtype=torch.int32)
+                           for c in counts])
+    chunks = torch.stack([tensor_idx, chunk_idx], dim=1).contiguous()
+    dev = params[0].device
+    tbl = (descs.to(dev), chunks.to(dev))
+    _ADAM_TABLE_CACHE[key] = tbl
+    return tbl
+
+
 @torch.no_grad()
 def fused_adamw(params: List[torch.Tensor], grads: List[torch.Tensor],
                 exp_avgs: List[torch.Tensor], exp_avg_sqs: List[torch.Tensor],
                 step: int, lr: float, beta1: float = 0.9, beta2: float = 0.95,
                 eps: float = 1e-8, weight_decay: float = 0.0,
                 grad_scale: float = 1.0) -> None:
-    """Multi-tensor AdamW update, in place.
+    """Multi-tensor AdamW update, in place — ONE kernel launch for all
+    params.
 
     Params may be bf16 (fp32 master math happens inside the kernel against
     the fp32 exp_avg/exp_avg_sq state).  Analog of the fused Adam the
@@ -164,8 +205,9 @@ def fused_adamw(params: List[torch.Tensor], grads: List[torch.Tensor],
     if not params:
         return
     if use_hip(params[0]):
-        hip_ops().adamw_step(params, grads, exp_avgs, exp_avg_sqs, step, lr,
-                             beta1, beta2, eps, weight_decay, grad_scale)
+        descs, chunks = _adam_tables(params, grads, exp_avgs, exp_avg_sqs)
+        hip_ops().adamw_step_raw(descs, chunks, step, lr, beta1, beta2, eps,
+                                 weight_decay, grad_scale)
     else:
         ref.adamw_step(params, grads, exp_avgs, exp_avg_sqs, step, lr, beta1,
                        beta2, eps, weight_decay, grad_scale)

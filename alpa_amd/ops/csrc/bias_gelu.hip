@@ -1,0 +1,111 @@
+// Fused bias + GeLU(tanh) forward/backward, bf16 (gfx950).
+//
+// The epilogue of the MLP up-projection GEMM (reference gets this from XLA
+// fusion; SURVEY.md §2.3 N13 "GEMM+bias+GeLU epilogue").  Memory-bound:
+// bf16x8 vector loads, grid-stride, dbias via deterministic striped
+// partials + column sum.
+#include "common.h"
+
+__device__ __forceinline__ float gelu_f(float x) {
+  const float k = 0.7978845608028654f;
+  return 0.5f * x * (1.0f + tanhf(k * (x + 0.044715f * x * x * x)));
+}
+
+__device__ __forceinline__ float gelu_grad_f(float x) {
+  const float k = 0.7978845608028654f;
+  float x2 = x * x;
+  float t = tanhf(k * (x + 0.044715f * x2 * x));
+  float dt = (1.0f - t * t) * k * (1.0f + 3.f * 0.044715f * x2);
+  return 0.5f * (1.0f + t) + 0.5f * x * dt;
+}
+
+// x [N, F] bf16, bias [F] -> y [N, F]
+__global__ void bias_gelu_fwd_kernel(const short* __restrict__ x,
+                                     const short* __restrict__ bias,
+                                     short* __restrict__ y, int64_t total,
+                                     int F) {
+  int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (; idx < total; idx += stride) {
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(x + idx);
+    bf16x8 bv = *reinterpret_cast<const bf16x8*>(bias + (idx % F));
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = f2bf(gelu_f(bf2f(v[j]) + bf2f(bv[j])));
+    *reinterpret_cast<bf16x8*>(y + idx) = o;
+  }
+}
+
+__global__ void bias_gelu_bwd_dx_kernel(const short* __restrict__ dy,
+                                        const short* __restrict__ x,
+                                        const short* __restrict__ bias,
+                                        short* __restrict__ dx, int64_t total,
+                                        int F) {
+  int64_t idx = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x * 8;
+  for (; idx < total; idx += stride) {
+    bf16x8 dv = *reinterpret_cast<const bf16x8*>(dy + idx);
+    bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + idx);
+    bf16x8 bv = *reinterpret_cast<const bf16x8*>(bias + (idx % F));
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float z = bf2f(xv[j]) + bf2f(bv[j]);
+      o[j] = f2bf(bf2f(dv[j]) * gelu_grad_f(z));
+    }
+    *reinterpret_cast<bf16x8*>(dx + idx) = o;
+  }
+}
+
+// dbias partials: stripe p sums rows p, p+P, ... of dx over column chunk.
+__global__ void bias_grad_partial_kernel(const short* __restrict__ dx,
+                                         float* __restrict__ part, int64_t N,
+                                         int F) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  const int p = blockIdx.y;
+  const int P = gridDim.y;
+  if (col >= F) return;
+  float s = 0.f;
+  for (int64_t row = p; row < N; row += P)
+    s += bf2f(dx[row * F + col]);
+  part[(int64_t)p * F + col] = s;
+}
+
+__global__ void bias_colsum_kernel(const float* __restrict__ part,
+                                   float* __restrict__ out, int P, int F) {
+  int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= F) return;
+  float s = 0.f;
+  for (int p = 0; p < P; ++p) s += part[(int64_t)p * F + col];
+  out[col] = s;
+}
+
+extern "C" {
+
+hipError_t launch_bias_gelu_fwd(const void* x, const void* bias, void* y,
+                                int64_t N, int64_t F, hipStream_t stream) {
+  int64_t total = N * F;
+  int blocks = (int)min((int64_t)2048, ceil_div(total, 256 * 8));
+  bias_gelu_fwd_kernel<<<dim3(blocks), dim3(256), 0, stream>>>(
+      (const short*)x, (const short*)bias, (short*)y, total, (int)F);
+  return hipGetLastError();
+}
+
+hipError_t launch_bias_gelu_bwd(const void* dy, const void* x,
+                                const void* bias, void* dx, float* db_part,
+                                float* db, int64_t N, int64_t F, int P,
+                                hipStream_t stream) {
+  int64_t total = N * F;
+  int blocks = (int)min((int64_t)2048, ceil_div(total, 256 * 8));
+  bias_gelu_bwd_dx_kernel<<<dim3(blocks), dim3(256), 0, stream>>>(
+      (const short*)dy, (const short*)x, (const short*)bias, (short*)dx,
+      total, (int)F);
+  dim3 grid((uint32_t)ceil_div(F, 256), P);
+  bias_grad_partial_kernel<<<grid, dim3(256), 0, stream>>>(
+      (const short*)dx, db_part, N, (int)F);
+  bias_colsum_kernel<<<dim3((uint32_t)ceil_div(F, 256)), dim3(256), 0,
+                       stream>>>(db_part, db, P, (int)F);
+  return hipGetLastError();
+}
+
+}  // extern "C"

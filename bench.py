@@ -1,0 +1,159 @@
+"""Flagship training benchmark — GPT-2.6B (BASELINE.md headline config).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+(N>1 launched via torch.distributed.run, one rank per GPU over RCCL.)
+
+Measures Alpa's headline metric on Alpa's headline config: model TFLOPS for
+GPT-2.6B (H=2560, L=32, heads=32, vocab 51200, seq 1024), batch 32 per GPU,
+num_micro_batches=4, bf16, synthetic data, random-init weights.  TFLOPS
+accounting follows the reference formula exactly
+(``benchmark/alpa/util.py:65-89``):
+  factor*B*S*H^2*L*(1+S/(6H)) + 6*B*S*H*V, factor=72 fwd+bwd (96 w/ remat).
+Baseline: 37.01 TFLOPS/GPU x 8 = 296 TFLOPS aggregate on 8x V100
+(benchmark/alpa/README.md:89-101).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+import alpa_amd as aa
+from alpa_amd.models.gpt import GPTConfig, GPTModel, gpt_config
+
+
+def model_tflops_per_step(cfg: GPTConfig, global_batch: int,
+                          remat: bool = False) -> float:
+    B, S, H, L, V = (global_batch, cfg.seq_len, cfg.hidden_size,
+                     cfg.num_layers, cfg.vocab_size)
+    factor = 96 if remat else 72
+    return (factor * B * S * H * H * L * (1 + S / (6 * H)) +
+            6 * B * S * H * V) / 1e12
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=5)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--model", type=str, default="auto",
+                   help="GPT spec name (125M..76B) or 'auto'")
+    p.add_argument("--batch-per-gpu", type=int, default=32)
+    p.add_argument("--nmb", type=int, default=4, help="num micro batches")
+    p.add_argument("--dp", type=int, default=0, help="data-parallel degree "
+                   "(0 = all GPUs)")
+    p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
+    p.add_argument("--seq", type=int, default=1024)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    aa.init()
+    on_gpu = torch.cuda.is_available()
+    n = aa.world_size()
+    assert n == args.gpus or args.gpus == 1 or not on_gpu, \
+        f"world_size {n} != --gpus {args.gpus}"
+    n = max(n, 1)
+
+    model_name = args.model
+    if model_name == "auto":
+        model_name = "2.6B" if on_gpu else "125M"
+    if on_gpu:
+        cfg = gpt_config(model_name, seq_len=args.seq)
+        batch_per_gpu = args.batch_per_gpu
+    else:
+        # CPU smoke: tiny shapes so the script works without a GPU
+        cfg = GPTConfig(hidden_size=128, num_layers=2, num_heads=4,
+                        seq_len=64, vocab_size=512)
+        batch_per_gpu = 4
+
+    dp = args.dp if args.dp > 0 else n // args.tp
+    tp = args.tp
+    assert dp * tp == n, (dp, tp, n)
+
+    method = aa.ShardParallel(num_micro_batches=args.nmb,
+                              logical_mesh_shape=(dp, tp))
+
+    def build(mesh=None, axis=1, dtype=torch.float32, device=None):
+        torch.manual_seed(1234)
+        return GPTModel(cfg, mesh, axis, dtype, device)
+
+    state = aa.TrainState.create(build, method, lr=1e-4, weight_decay=0.01)
+    step_fn = aa.parallelize(
+        lambda model, b: model.loss(b["ids"], b["labels"]), method=method)
+
+    # synthetic data of the benchmark shape, one local batch per dp rank
+    g = torch.Generator().manual_seed(4321 + aa.rank())
+    ids = torch.randint(0, cfg.vocab_size, (batch_per_gpu, cfg.seq_len),
+                        generator=g)
+    labels = torch.randint(0, cfg.vocab_size, (batch_per_gpu, cfg.seq_len),
+                           generator=g)
+    dev = aa.device()
+    batch = {"ids": ids.to(dev), "labels": labels.to(dev)}
+
+    mesh = state.mesh
+
+    def sync():
+        if mesh is not None:
+            mesh.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step_fn(state, batch)
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = step_fn(state, batch)
+    sync()
+    elapsed = time.perf_counter() - t0
+    # MAX over ranks
+    et = torch.tensor([elapsed], dtype=torch.float64)
+    if aa.world_size() > 1:
+        import torch.distributed as dist
+        et_d = et.to(dev) if on_gpu else et
+        dist.all_reduce(et_d, op=dist.ReduceOp.MAX)
+        elapsed = float(et_d.item())
+
+    ms_per_step = elapsed / args.steps * 1000
+    global_batch = batch_per_gpu * dp
+    tflops_step = model_tflops_per_step(cfg, global_batch)
+    value = tflops_step / (elapsed / args.steps)  # aggregate TFLOPS
+    baseline_aggregate_tflops = 296.0  # 37.01 TF/GPU x 8 V100
+
+    if aa.rank() == 0:
+        result = {
+            "metric": "model_tflops_aggregate",
+            "value": round(value, 2),
+            "unit": "TFLOPS",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(value / baseline_aggregate_tflops, 3)
+                           if on_gpu else None,
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"GPT-{model_name}" if on_gpu else "GPT-tiny-cpu",
+                "hidden": cfg.hidden_size, "layers": cfg.num_layers,
+                "heads": cfg.num_heads, "vocab": cfg.vocab_size,
+                "global_batch": global_batch, "seq_len": cfg.seq_len,
+                "num_micro_batches": args.nmb,
+                "parallelism": f"dp{dp}" + (f"tp{tp}" if tp > 1 else ""),
+                "loss": float(loss),
+            },
+        }
+        print(json.dumps(result))
+    aa.shutdown()
+
+
+if __name__ == "__main__":
+    main()
