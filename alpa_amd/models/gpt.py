@@ -88,15 +88,10 @@ class Attention(nn.Module):
                                      axis, dtype=dtype, device=device)
 
     def forward(self, x):
-        B, S, _ = x.shape
+        # packed per-head [q|k|v] projection; the strided attention kernel
+        # consumes/produces these layouts directly (no permute copies)
         qkv = self.qkv(x)  # [B, S, 3*H/tp]
-        qkv = qkv.view(B, S, self.heads_per_rank, 3 * self.head_dim)
-        q, k, v = qkv.chunk(3, dim=-1)
-        q = q.permute(0, 2, 1, 3).contiguous()  # [B, h/tp, S, d]
-        k = k.permute(0, 2, 1, 3).contiguous()
-        v = v.permute(0, 2, 1, 3).contiguous()
-        o = ops.flash_attention(q, k, v, causal=True)
-        o = o.permute(0, 2, 1, 3).reshape(B, S, -1)
+        o = ops.flash_attention_qkv(qkv, self.heads_per_rank, causal=True)
         return self.out(o)
 
 

@@ -22,8 +22,8 @@ from . import reference as ref
 from ._backend import hip_ops, hip_ops_available, use_hip
 
 __all__ = [
-    "layer_norm", "bias_gelu", "flash_attention", "softmax_cross_entropy",
-    "fused_adamw", "hip_ops_available",
+    "layer_norm", "bias_gelu", "flash_attention", "flash_attention_qkv",
+    "softmax_cross_entropy", "fused_adamw", "hip_ops_available",
 ]
 
 
@@ -115,6 +115,72 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     hand-written gfx950 kernel).
     """
     return _FlashAttention.apply(q, k, v, causal, scale)
+
+
+class _FlashAttentionQKV(torch.autograd.Function):
+    """Packed-qkv attention: input [B, S, h*3d] (per-head [q|k|v] layout),
+    output [B, S, h*d].  The strided gfx950 kernel reads/writes these
+    layouts directly — zero permute/contiguous copies on the hot path."""
+
+    @staticmethod
+    def forward(ctx, qkv, num_heads, causal, scale):
+        B, S, F = qkv.shape
+        h = num_heads
+        d = F // (3 * h)
+        if scale is None:
+            scale = 1.0 / math.sqrt(d)
+        qkv5 = qkv.view(B, S, h, 3, d)
+        q = qkv5[:, :, :, 0].permute(0, 2, 1, 3)  # [B,h,S,d] strided view
+        k = qkv5[:, :, :, 1].permute(0, 2, 1, 3)
+        v = qkv5[:, :, :, 2].permute(0, 2, 1, 3)
+        if use_hip(qkv):
+            o_buf = torch.empty(B, S, h * d, dtype=qkv.dtype,
+                                device=qkv.device)
+            o_view = o_buf.view(B, S, h, d).permute(0, 2, 1, 3)
+            lse = torch.empty(B, h, S, dtype=torch.float32,
+                              device=qkv.device)
+            hip_ops().attn_fwd_out(q, k, v, o_view, lse, causal, scale)
+        else:
+            o4, lse = ref.attention_fwd(q.contiguous(), k.contiguous(),
+                                        v.contiguous(), causal, scale)
+            o_buf = o4.permute(0, 2, 1, 3).reshape(B, S, h * d)
+        ctx.save_for_backward(qkv, o_buf, lse)
+        ctx.meta = (num_heads, causal, scale)
+        return o_buf
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, o_buf, lse = ctx.saved_tensors
+        h, causal, scale = ctx.meta
+        B, S, F = qkv.shape
+        d = F // (3 * h)
+        qkv5 = qkv.view(B, S, h, 3, d)
+        q = qkv5[:, :, :, 0].permute(0, 2, 1, 3)
+        k = qkv5[:, :, :, 1].permute(0, 2, 1, 3)
+        v = qkv5[:, :, :, 2].permute(0, 2, 1, 3)
+        do4 = do.view(B, S, h, d).permute(0, 2, 1, 3)
+        o4 = o_buf.view(B, S, h, d).permute(0, 2, 1, 3)
+        if use_hip(qkv):
+            dq, dk, dv = hip_ops().attn_bwd(do4, q, k, v, o4, lse, causal,
+                                            scale)
+        else:
+            dq, dk, dv = ref.attention_bwd(
+                do4.contiguous(), q.contiguous(), k.contiguous(),
+                v.contiguous(), o4.contiguous(), lse, causal, scale)
+        dqkv = torch.empty_like(qkv)
+        dqkv5 = dqkv.view(B, S, h, 3, d)
+        dqkv5[:, :, :, 0].copy_(dq.permute(0, 2, 1, 3))
+        dqkv5[:, :, :, 1].copy_(dk.permute(0, 2, 1, 3))
+        dqkv5[:, :, :, 2].copy_(dv.permute(0, 2, 1, 3))
+        return dqkv, None, None, None
+
+
+def flash_attention_qkv(qkv: torch.Tensor, num_heads: int,
+                        causal: bool = True,
+                        scale: Optional[float] = None) -> torch.Tensor:
+    """Attention on the packed qkv projection output [B, S, heads*3*d]
+    (per-head [q|k|v] blocks) -> [B, S, heads*d].  No layout copies."""
+    return _FlashAttentionQKV.apply(qkv, num_heads, causal, scale)
 
 
 class _SoftmaxCrossEntropy(torch.autograd.Function):

@@ -19,25 +19,38 @@
 #define ATTN_BLOCK_K 64
 #define ATTN_THREADS 256
 
+// Strides are in elements; the innermost (D) dim must be contiguous.
+// Strided addressing lets the packed qkv layout [B, S, heads, 3*D] feed the
+// kernel directly — no permute/contiguous copies on the hot path.
+struct AttnStrides {
+  int64_t qb, qh, qs;  // q batch/head/seq strides
+  int64_t kb, kh, ks;
+  int64_t vb, vh, vs;
+  int64_t ob, oh, os;
+};
+
 template <int Dp>
 __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
-    float* __restrict__ lse_out, int S, int Skv, int D, float scale,
-    int causal) {
+    float* __restrict__ lse_out, int H, int S, int Skv, int D, float scale,
+    int causal, AttnStrides st) {
   constexpr int KSTEPS_QK = Dp / 32;   // k-steps over head dim
   constexpr int NTILES = ATTN_BLOCK_K / 16;  // 4
   constexpr int DTILES = Dp / 16;
 
   const int qb = blockIdx.x;           // q block index
   const int bh = blockIdx.y;           // fused batch*head
+  const int batch = bh / H, head = bh % H;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int lo = lane & 15;            // fragment row/col low index
   const int hi = lane >> 4;            // fragment quartet index
 
-  const int64_t qoff = (int64_t)bh * S * D;
-  const int64_t kvoff = (int64_t)bh * Skv * D;
+  const short* qp = q + batch * st.qb + head * st.qh;
+  const short* kp = k + batch * st.kb + head * st.kh;
+  const short* vp = v + batch * st.vb + head * st.vh;
+  short* op = o + batch * st.ob + head * st.oh;
   const int q_row0 = qb * ATTN_BLOCK_Q + wave * 16;  // wave's first q row
 
   __shared__ short k_lds[ATTN_BLOCK_K][Dp];
@@ -54,7 +67,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
       int col = ks * 32 + hi * 8;
       if (col + 8 <= D) {
         q_frag[ks] =
-            *reinterpret_cast<const bf16x8*>(q + qoff + (int64_t)row * D + col);
+            *reinterpret_cast<const bf16x8*>(qp + (int64_t)row * st.qs + col);
       } else {
         bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
         q_frag[ks] = z;
@@ -91,10 +104,8 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
         bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
         int src = kvb + kvr;
         if (src < Skv && dg + 8 <= D) {
-          kv8 = *reinterpret_cast<const bf16x8*>(k + kvoff +
-                                                 (int64_t)src * D + dg);
-          vv8 = *reinterpret_cast<const bf16x8*>(v + kvoff +
-                                                 (int64_t)src * D + dg);
+          kv8 = *reinterpret_cast<const bf16x8*>(kp + (int64_t)src * st.ks + dg);
+          vv8 = *reinterpret_cast<const bf16x8*>(vp + (int64_t)src * st.vs + dg);
         }
         *reinterpret_cast<bf16x8*>(&k_lds[kvr][dg]) = kv8;
 #pragma unroll
@@ -202,7 +213,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     for (int dt = 0; dt < DTILES; ++dt) {
       int col = dt * 16 + lo;
       if (col < D)
-        o[qoff + (int64_t)q_idx * D + col] = f2bf(o_acc[dt][r] * inv_l);
+        op[(int64_t)q_idx * st.os + col] = f2bf(o_acc[dt][r] * inv_l);
     }
     if (lo == 0 && lse_out != nullptr)
       lse_out[(int64_t)bh * S + q_idx] =
@@ -215,21 +226,35 @@ extern "C" {
 hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
                            void* o, float* lse, int64_t B, int64_t H,
                            int64_t S, int64_t Skv, int64_t D, float scale,
-                           int causal, hipStream_t stream) {
+                           int causal, const int64_t* strides,
+                           hipStream_t stream) {
   dim3 grid((uint32_t)ceil_div(S, ATTN_BLOCK_Q), (uint32_t)(B * H));
   dim3 block(ATTN_THREADS);
+  AttnStrides st;
+  st.qb = strides[0];
+  st.qh = strides[1];
+  st.qs = strides[2];
+  st.kb = strides[3];
+  st.kh = strides[4];
+  st.ks = strides[5];
+  st.vb = strides[6];
+  st.vh = strides[7];
+  st.vs = strides[8];
+  st.ob = strides[9];
+  st.oh = strides[10];
+  st.os = strides[11];
   if (D <= 64) {
     attn_fwd_kernel<64><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)S, (int)Skv, (int)D, scale, causal);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
   } else if (D <= 96) {
     attn_fwd_kernel<96><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)S, (int)Skv, (int)D, scale, causal);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
   } else if (D <= 128) {
     attn_fwd_kernel<128><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)S, (int)Skv, (int)D, scale, causal);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
   } else {
     return hipErrorInvalidValue;
   }

@@ -40,7 +40,7 @@ hipError_t launch_ce_bwd(const float*, const void*, const int64_t*,
                          const float*, void*, int64_t, int64_t, hipStream_t);
 hipError_t launch_attn_fwd(const void*, const void*, const void*, void*,
                            float*, int64_t, int64_t, int64_t, int64_t,
-                           int64_t, float, int, hipStream_t);
+                           int64_t, float, int, const int64_t*, hipStream_t);
 hipError_t launch_mfma_probe(const void*, const void*, float*, hipStream_t);
 }
 
@@ -184,65 +184,90 @@ at::Tensor cross_entropy_bwd(const at::Tensor& dloss, const at::Tensor& logits,
 
 // ------------------------------ Attention --------------------------------
 
-std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
-                                 const at::Tensor& v, bool causal,
-                                 double scale) {
-  check_bf16_contig(q, "q");
-  check_bf16_contig(k, "k");
-  check_bf16_contig(v, "v");
-  TORCH_CHECK(q.dim() == 4, "q must be [B, H, S, D]");
+void check_bf16_strided4(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.dim() == 4, name, " must be 4-D [B,H,S,D]");
+  TORCH_CHECK(t.stride(3) == 1, name, " head-dim must be contiguous");
+}
+
+// q/k/v/o are logical [B, H, S, D] views with arbitrary B/H/S strides and a
+// contiguous D — so the packed qkv layout [B, S, heads, 3*D] feeds the
+// kernel directly, with no permute/contiguous copies on the hot path.
+at::Tensor attn_fwd_out(const at::Tensor& q, const at::Tensor& k,
+                        const at::Tensor& v, at::Tensor o, at::Tensor lse,
+                        bool causal, double scale) {
+  check_bf16_strided4(q, "q");
+  check_bf16_strided4(k, "k");
+  check_bf16_strided4(v, "v");
+  check_bf16_strided4(o, "o");
   int64_t B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   int64_t Skv = k.size(2);
   TORCH_CHECK(D % 16 == 0 && D <= 128, "head_dim must be <=128, mult of 16");
-  auto o = at::empty_like(q);
-  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  TORCH_CHECK(lse.is_contiguous() && lse.numel() == B * H * S);
+  int64_t strides[12] = {q.stride(0), q.stride(1), q.stride(2),
+                         k.stride(0), k.stride(1), k.stride(2),
+                         v.stride(0), v.stride(1), v.stride(2),
+                         o.stride(0), o.stride(1), o.stride(2)};
   HIP_OK(launch_attn_fwd(q.const_data_ptr(), k.const_data_ptr(),
                          v.const_data_ptr(), o.mutable_data_ptr(),
                          (float*)lse.mutable_data_ptr(), B, H, S, Skv, D,
-                         (float)scale, causal ? 1 : 0, cur_stream()));
+                         (float)scale, causal ? 1 : 0, strides,
+                         cur_stream()));
+  return o;
+}
+
+std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
+                                 const at::Tensor& v, bool causal,
+                                 double scale) {
+  int64_t B = q.size(0), H = q.size(1), S = q.size(2);
+  auto o = at::empty_like(q.contiguous());
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  attn_fwd_out(q, k, v, o, lse, causal, scale);
   return {o, lse};
 }
 
-// Attention backward: deterministic blocked recompute using library GEMMs
-// (hipBLASLt via at::matmul).  The q-block loop keeps the S x S score
-// matrix from materializing beyond one [Bq, Skv] block.  (A fully
-// hand-written HIP bwd kernel replaces this on the optimization path.)
-std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
-                                 const at::Tensor& k, const at::Tensor& v,
-                                 const at::Tensor& o, const at::Tensor& lse,
+// Attention backward: deterministic blocked recompute; all GEMMs in bf16
+// (hipBLASLt / MFMA), softmax math in fp32.  The q-block loop keeps the
+// S x S score matrix from materializing beyond one [Bq, Skv] block.
+// (A fully hand-written HIP bwd kernel replaces this on the optimization
+// path.)
+std::vector<at::Tensor> attn_bwd(const at::Tensor& dout_, const at::Tensor& q_,
+                                 const at::Tensor& k_, const at::Tensor& v_,
+                                 const at::Tensor& o_, const at::Tensor& lse,
                                  bool causal, double scale) {
+  auto q = q_.contiguous(), k = k_.contiguous(), v = v_.contiguous();
+  auto dout = dout_.contiguous(), o = o_.contiguous();
   int64_t B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   int64_t Skv = k.size(2);
-  auto kf = k.to(at::kFloat), vf = v.to(at::kFloat);
-  auto dq = at::zeros_like(q, q.options().dtype(at::kFloat));
-  auto dk = at::zeros_like(kf);
-  auto dv = at::zeros_like(vf);
-  const int64_t BQ = 256;  // q-block rows
+  auto dq = at::empty_like(q);
+  auto dk = at::zeros_like(k, k.options().dtype(at::kFloat));
+  auto dv = at::zeros_like(v, v.options().dtype(at::kFloat));
+  // delta = rowsum(dO * O), fp32, computed once
+  auto delta = (dout.to(at::kFloat) * o.to(at::kFloat)).sum(-1, true);
+  auto lse4 = lse.view({B, H, S, 1});
+  const int64_t BQ = 256;
   for (int64_t qs = 0; qs < S; qs += BQ) {
     int64_t qe = std::min(qs + BQ, S);
-    auto qb = q.slice(2, qs, qe).to(at::kFloat);
-    auto dob = dout.slice(2, qs, qe).to(at::kFloat);
-    auto ob = o.slice(2, qs, qe).to(at::kFloat);
-    auto lseb = lse.slice(2, qs, qe).unsqueeze(-1);
+    auto qb = q.slice(2, qs, qe);
+    auto dob = dout.slice(2, qs, qe);
     int64_t ke = causal ? std::min(qe, Skv) : Skv;
-    auto kb = kf.slice(2, 0, ke);
-    auto vb = vf.slice(2, 0, ke);
-    auto s = at::matmul(qb, kb.transpose(-1, -2)) * scale;
-    if (causal) {
-      auto mask = at::triu(at::ones({qe - qs, ke}, s.options().dtype(at::kBool)),
-                           /*diagonal=*/qs + 1);
-      s.masked_fill_(mask, -std::numeric_limits<float>::infinity());
-    }
-    auto p = at::exp(s - lseb);
-    dv.slice(2, 0, ke).add_(at::matmul(p.transpose(-1, -2), dob));
-    auto dp = at::matmul(dob, vb.transpose(-1, -2));
-    auto delta = (dob * ob).sum(-1, /*keepdim=*/true);
-    auto ds = p * (dp - delta) * scale;
-    dq.slice(2, qs, qe).add_(at::matmul(ds, kb));
-    dk.slice(2, 0, ke).add_(at::matmul(ds.transpose(-1, -2), qb));
+    auto kb = k.slice(2, 0, ke);
+    auto vb = v.slice(2, 0, ke);
+    // p = exp(s*scale - lse), zeroed above the causal diagonal via tril
+    auto s = at::matmul(qb, kb.transpose(-1, -2));  // bf16 GEMM
+    auto p = at::exp(s.to(at::kFloat) * scale - lse4.slice(2, qs, qe));
+    if (causal) p = p.tril_(qs);
+    auto p_bf = p.to(at::kBFloat16);
+    dv.slice(2, 0, ke).add_(at::matmul(p_bf.transpose(-1, -2), dob));
+    auto dp = at::matmul(dob, vb.transpose(-1, -2));  // bf16 GEMM
+    auto ds = p.mul_(dp.to(at::kFloat) - delta.slice(2, qs, qe))
+                  .mul_(scale);  // reuse p storage
+    auto ds_bf = ds.to(at::kBFloat16);
+    dq.slice(2, qs, qe).copy_(at::matmul(ds_bf, kb));
+    dk.slice(2, 0, ke).add_(at::matmul(ds_bf.transpose(-1, -2), qb));
   }
-  return {dq.to(q.scalar_type()), dk.to(k.scalar_type()),
-          dv.to(v.scalar_type())};
+  return {dq, dk.to(k.scalar_type()), dv.to(v.scalar_type())};
 }
 
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
@@ -264,6 +289,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE fwd (gfx950)");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE bwd (gfx950)");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)");
+  m.def("attn_fwd_out", &attn_fwd_out, "flash attention fwd, strided out");
   m.def("attn_bwd", &attn_bwd, "attention bwd (blocked, deterministic)");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
 }
