@@ -62,10 +62,17 @@ class TrainState:
             if torch.cuda.is_available() else torch.float32
         model = model_fn(mesh=mesh, axis=method.tp_axis, dtype=dtype,
                          device=device())
-        opt = optimizer_cls(model.parameters(), lr=lr, betas=betas,
-                            weight_decay=weight_decay)
-        state = cls(model, opt, method, mesh)
+        state = cls(model, None, method, mesh)
         state._install_grad_sync()
+        if method.zero_stage >= 2:
+            from .parallel.zero import ZeroOptimizer
+            state.optimizer = ZeroOptimizer(state.grad_sync, lr=lr,
+                                            betas=betas,
+                                            weight_decay=weight_decay)
+        else:
+            state.optimizer = optimizer_cls(model.parameters(), lr=lr,
+                                            betas=betas,
+                                            weight_decay=weight_decay)
         return state
 
     def _install_grad_sync(self):
@@ -124,11 +131,7 @@ class ParallelizedFunc:
         # grad scale: mean over microbatches; all-reduce over dp was a SUM
         dp = state.mesh.axis_size(m.dp_axis) if state.mesh is not None else 1
         scale = 1.0 / (nmb * (dp if dp > 1 else 1))
-        if m.zero_stage >= 2:
-            state.optimizer.step_sharded(gs, grad_scale=scale)
-        else:
-            state.optimizer.step(grads=[p.grad for p in state.optimizer.params],
-                                 grad_scale=scale)
+        state.optimizer.step(grad_scale=scale)
         state.step_count += 1
         return total_loss / nmb
 

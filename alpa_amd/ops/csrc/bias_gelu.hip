@@ -65,8 +65,9 @@ __global__ void bias_grad_partial_kernel(const short* __restrict__ dx,
   const int p = blockIdx.y;
   const int P = gridDim.y;
   if (col >= F) return;
+  const int64_t r0 = (int64_t)p * N / P, r1 = (int64_t)(p + 1) * N / P;
   float s = 0.f;
-  for (int64_t row = p; row < N; row += P)
+  for (int64_t row = r0; row < r1; ++row)
     s += bf2f(dx[row * F + col]);
   part[(int64_t)p * F + col] = s;
 }

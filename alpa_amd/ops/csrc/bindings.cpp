@@ -9,6 +9,7 @@
 #include <hip/hip_runtime.h>
 
 #include <vector>
+#include <algorithm>
 
 #define HIP_OK(expr)                                                       \
   do {                                                                     \
@@ -56,7 +57,13 @@ void check_bf16_contig(const at::Tensor& t, const char* name) {
   TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
 }
 
-constexpr int kStripes = 64;  // deterministic column-partial stripe count
+// Deterministic column-partial stripe count: enough (stripes x col-chunks)
+// blocks to fill 256 CUs, bounded by the row count.
+int stripes_for(int64_t N, int64_t cols) {
+  int64_t chunks = std::max<int64_t>(1, cols / 256);
+  int64_t p = std::max<int64_t>(64, 2048 / chunks);
+  return (int)std::min<int64_t>({p, N, 512});
+}
 
 // ------------------------------- LayerNorm -------------------------------
 
@@ -91,6 +98,7 @@ std::vector<at::Tensor> layer_norm_bwd(const at::Tensor& dy,
   auto f32 = x.options().dtype(at::kFloat);
   auto dw = at::empty({H}, f32);
   auto db = at::empty({H}, f32);
+  const int kStripes = stripes_for(N, H);
   auto dw_part = at::empty({kStripes, H}, f32);
   auto db_part = at::empty({kStripes, H}, f32);
   HIP_OK(launch_layer_norm_bwd_dx(
@@ -126,6 +134,7 @@ std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& dy,
   int64_t N = x.numel() / F;
   auto dx = at::empty_like(x);
   auto f32 = x.options().dtype(at::kFloat);
+  const int kStripes = stripes_for(N, F);
   auto db = at::empty({F}, f32);
   auto db_part = at::empty({kStripes, F}, f32);
   HIP_OK(launch_bias_gelu_bwd(dy.const_data_ptr(), x.const_data_ptr(),

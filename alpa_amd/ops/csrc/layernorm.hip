@@ -124,11 +124,13 @@ __global__ void layer_norm_bwd_dwdb_partial(const short* __restrict__ dy,
   const int p = blockIdx.y;  // stripe index
   const int P = gridDim.y;
   if (col >= H) return;
+  // contiguous row range per stripe (DRAM page locality)
+  const int64_t r0 = (int64_t)p * N / P, r1 = (int64_t)(p + 1) * N / P;
   float dw = 0.f, db = 0.f;
-  for (int row = p; row < N; row += P) {
+  for (int64_t row = r0; row < r1; ++row) {
     float mean = mean_in[row], rstd = rstd_in[row];
-    float d = bf2f(dy[(int64_t)row * H + col]);
-    float xv = bf2f(x[(int64_t)row * H + col]);
+    float d = bf2f(dy[row * H + col]);
+    float xv = bf2f(x[row * H + col]);
     dw += d * (xv - mean) * rstd;
     db += d;
   }

@@ -116,6 +116,31 @@ def test_grad_accumulation_matches_serial():
         torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-5)
 
 
+def _zero2_worker(rank, world_size):
+    method = aa.Zero2Parallel(num_micro_batches=2)
+    state = aa.TrainState.create(build_mlp, method, lr=1e-3)
+    step = aa.parallelize(loss_fn, method=method)
+    per = BATCH // world_size
+    idx = state.mesh.axis_index(0)
+    losses = []
+    for i in range(STEPS):
+        x, y = make_batch(i)
+        losses.append(float(step(state, (x[idx * per:(idx + 1) * per],
+                                         y[idx * per:(idx + 1) * per]))))
+    params = [p.detach().clone() for p in state.model.parameters()]
+    return losses, params
+
+
+def test_zero2_matches_serial():
+    """ZeRO-2 (reduce-scatter + sharded AdamW + param all-gather) must give
+    the same updates as plain DP / serial."""
+    serial_losses, serial_params = run_serial(num_micro_batches=2)
+    results = run_distributed(_zero2_worker, world_size=2)
+    for _, rank_params in results:
+        for sp, rp in zip(serial_params, rank_params):
+            torch.testing.assert_close(rp, sp, rtol=1e-4, atol=1e-5)
+
+
 def test_dp2_with_microbatches():
     serial_losses, serial_params = run_serial(num_micro_batches=2)
     results = run_distributed(_dp_worker, world_size=2, args=(2, (2, 1)))
