@@ -48,15 +48,26 @@ class TrainState:
         self.method = method
         self.mesh = mesh
         self.grad_sync: Optional[GradSynchronizer] = None
+        self.engine = None  # PipelineEngine for PipeshardParallel
         self.step_count = 0
 
     @classmethod
-    def create(cls, model_fn: Callable[..., torch.nn.Module],
-               method: ParallelMethod, lr: float = 1e-4, betas=(0.9, 0.95),
-               weight_decay: float = 0.0,
+    def create(cls, model_fn, method: ParallelMethod, lr: float = 1e-4,
+               betas=(0.9, 0.95), weight_decay: float = 0.0,
                optimizer_cls=AdamW) -> "TrainState":
         """model_fn(mesh, axis, dtype, device) -> nn.Module built directly
-        on the target device with method-resolved sharding."""
+        on the target device with method-resolved sharding; for
+        PipeshardParallel, model_fn is a PipelineModelSpec instead."""
+        from .parallel_method import PipeshardParallel
+        if isinstance(method, PipeshardParallel):
+            from .pipeline_parallel.compile import build_pipeline_state
+            stage_module, engine, opt, gs, meshes = build_pipeline_state(
+                model_fn, method, lr, betas, weight_decay)
+            state = cls(stage_module, opt, method, engine.mesh)
+            state.engine = engine
+            state.grad_sync = gs
+            state.stage_meshes = meshes
+            return state
         mesh = method.resolve_mesh()
         dtype = getattr(torch, global_config.compute_dtype) \
             if torch.cuda.is_available() else torch.float32
@@ -118,6 +129,8 @@ class ParallelizedFunc:
         m = self.method
         nmb = m.num_micro_batches
         micro = _split_microbatches(batch, nmb)
+        if state.engine is not None:
+            return self._pipeline_call(state, micro)
         gs = state.grad_sync
         gs.zero_grads()
         total_loss = None
@@ -134,6 +147,22 @@ class ParallelizedFunc:
         state.optimizer.step(grad_scale=scale)
         state.step_count += 1
         return total_loss / nmb
+
+    def _pipeline_call(self, state: TrainState, micro) -> torch.Tensor:
+        """Pipeshard path: 1F1B engine step + per-stage fused optimizer +
+        loss broadcast from the last stage."""
+        import torch.distributed as dist
+        m = self.method
+        engine = state.engine
+        loss = engine.train_step(micro)
+        dp = engine.mesh.axis_size(m.dp_axis)
+        scale = 1.0 / (m.num_micro_batches * (dp if dp > 1 else 1))
+        state.optimizer.step(grad_scale=scale)
+        state.step_count += 1
+        if is_distributed():
+            loss = loss.contiguous()
+            dist.broadcast(loss, src=engine.loss_src_rank)
+        return loss
 
 
 def parallelize(fn: Optional[Callable] = None, *,

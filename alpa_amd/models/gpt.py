@@ -23,7 +23,7 @@ import torch.nn as nn
 from .. import ops
 from ..mesh import DeviceMesh
 from ..parallel.layers import (ColumnParallelLinear, RowParallelLinear,
-                               VocabParallelEmbedding,
+                               VocabParallelEmbedding, tag_seed,
                                vocab_parallel_cross_entropy)
 
 
@@ -70,12 +70,22 @@ def gpt_config(name: str, seq_len: int = 1024,
                      seq_len=seq_len, vocab_size=vocab_size)
 
 
+def _wpe_init(cfg, dtype, device, init_seed: int):
+    gen_device = device if (device is not None and
+                            torch.device(device).type == "cuda") else "cpu"
+    g = torch.Generator(device=gen_device)
+    g.manual_seed(tag_seed(init_seed, "wpe"))
+    w = torch.empty(cfg.seq_len, cfg.hidden_size, dtype=torch.float32,
+                    device=gen_device).normal_(0, 0.02, generator=g)
+    return w.to(dtype=dtype, device=device)
+
+
 class Attention(nn.Module):
     """Multi-head attention: fused qkv column-split by heads over the tp
     axis, flash-attention kernel, output row-split + all-reduce."""
 
     def __init__(self, cfg: GPTConfig, mesh: Optional[DeviceMesh], axis: int,
-                 dtype, device):
+                 dtype, device, layer_idx: int = 0, init_seed: int = 0):
         super().__init__()
         self.cfg = cfg
         tp = mesh.axis_size(axis) if mesh is not None else 1
@@ -83,9 +93,13 @@ class Attention(nn.Module):
         self.heads_per_rank = cfg.num_heads // tp
         self.head_dim = cfg.head_dim
         self.qkv = ColumnParallelLinear(cfg.hidden_size, 3 * cfg.hidden_size,
-                                        mesh, axis, dtype=dtype, device=device)
+                                        mesh, axis, dtype=dtype, device=device,
+                                        init_seed=init_seed,
+                                        init_tag=f"b{layer_idx}.qkv")
         self.out = RowParallelLinear(cfg.hidden_size, cfg.hidden_size, mesh,
-                                     axis, dtype=dtype, device=device)
+                                     axis, dtype=dtype, device=device,
+                                     init_seed=init_seed,
+                                     init_tag=f"b{layer_idx}.out")
 
     def forward(self, x):
         # packed per-head [q|k|v] projection; the strided attention kernel
@@ -98,13 +112,17 @@ class Attention(nn.Module):
 class MLP(nn.Module):
 
     def __init__(self, cfg: GPTConfig, mesh: Optional[DeviceMesh], axis: int,
-                 dtype, device):
+                 dtype, device, layer_idx: int = 0, init_seed: int = 0):
         super().__init__()
         ffn = cfg.ffn_mult * cfg.hidden_size
         self.fc1 = ColumnParallelLinear(cfg.hidden_size, ffn, mesh, axis,
-                                        gelu=True, dtype=dtype, device=device)
+                                        gelu=True, dtype=dtype, device=device,
+                                        init_seed=init_seed,
+                                        init_tag=f"b{layer_idx}.fc1")
         self.fc2 = RowParallelLinear(ffn, cfg.hidden_size, mesh, axis,
-                                     dtype=dtype, device=device)
+                                     dtype=dtype, device=device,
+                                     init_seed=init_seed,
+                                     init_tag=f"b{layer_idx}.fc2")
 
     def forward(self, x):
         return self.fc2(self.fc1(x))
@@ -126,12 +144,14 @@ class LayerNorm(nn.Module):
 
 class Block(nn.Module):
 
-    def __init__(self, cfg: GPTConfig, mesh, axis, dtype, device):
+    def __init__(self, cfg: GPTConfig, mesh, axis, dtype, device,
+                 layer_idx: int = 0, init_seed: int = 0):
         super().__init__()
         self.ln1 = LayerNorm(cfg.hidden_size, cfg.layernorm_eps, dtype, device)
-        self.attn = Attention(cfg, mesh, axis, dtype, device)
+        self.attn = Attention(cfg, mesh, axis, dtype, device, layer_idx,
+                              init_seed)
         self.ln2 = LayerNorm(cfg.hidden_size, cfg.layernorm_eps, dtype, device)
-        self.mlp = MLP(cfg, mesh, axis, dtype, device)
+        self.mlp = MLP(cfg, mesh, axis, dtype, device, layer_idx, init_seed)
 
     def forward(self, x):
         x = x + self.attn(self.ln1(x))
@@ -144,19 +164,19 @@ class GPTModel(nn.Module):
     (dp, tp) mesh); dp replication is handled by the trainer."""
 
     def __init__(self, cfg: GPTConfig, mesh: Optional[DeviceMesh] = None,
-                 axis: int = 1, dtype=torch.float32, device=None):
+                 axis: int = 1, dtype=torch.float32, device=None,
+                 init_seed: int = 0):
         super().__init__()
         self.cfg = cfg
         self.mesh, self.axis = mesh, axis
         self.wte = VocabParallelEmbedding(cfg.vocab_size, cfg.hidden_size,
                                           mesh, axis, dtype=dtype,
-                                          device=device)
-        self.wpe = nn.Parameter(
-            torch.empty(cfg.seq_len, cfg.hidden_size, dtype=dtype,
-                        device=device).normal_(0, 0.02))
+                                          device=device, init_seed=init_seed,
+                                          init_tag="wte")
+        self.wpe = nn.Parameter(_wpe_init(cfg, dtype, device, init_seed))
         self.blocks = nn.ModuleList([
-            Block(cfg, mesh, axis, dtype, device)
-            for _ in range(cfg.num_layers)
+            Block(cfg, mesh, axis, dtype, device, i, init_seed)
+            for i in range(cfg.num_layers)
         ])
         self.ln_f = LayerNorm(cfg.hidden_size, cfg.layernorm_eps, dtype,
                               device)
@@ -164,7 +184,9 @@ class GPTModel(nn.Module):
         # via vocab-parallel cross-entropy (no logits gather).
         self.lm_head = ColumnParallelLinear(cfg.hidden_size, cfg.vocab_size,
                                             mesh, axis, bias=False,
-                                            dtype=dtype, device=device)
+                                            dtype=dtype, device=device,
+                                            init_seed=init_seed,
+                                            init_tag="lm_head")
         if cfg.tie_embeddings:
             self.lm_head.weight = self.wte.weight
 
@@ -187,3 +209,87 @@ class GPTModel(nn.Module):
         per_tok = vocab_parallel_cross_entropy(
             logits, labels.reshape(N), self.mesh, self.axis, vocab_start)
         return per_tok.mean()
+
+
+# --------------------------------------------------------------------------
+# Pipeline-parallel stage (inter-op parallelism; reference slices the jaxpr
+# at pipeline markers — here the model family builds its stage directly)
+# --------------------------------------------------------------------------
+
+
+class GPTStage(nn.Module):
+    """One pipeline stage of the GPT LM: [layer_range) blocks, plus the
+    embedding on the first stage and ln_f + LM head + loss on the last.
+
+    forward(x, microbatch): x = activations [B, S, H] from the previous
+    stage (None on the first stage); returns activations, or the scalar
+    microbatch loss on the last stage.
+    """
+
+    def __init__(self, cfg: GPTConfig, layer_range, is_first: bool,
+                 is_last: bool, mesh: Optional[DeviceMesh] = None,
+                 axis: int = 1, dtype=torch.float32, device=None,
+                 init_seed: int = 0):
+        super().__init__()
+        self.cfg = cfg
+        self.mesh, self.axis = mesh, axis
+        self.is_first, self.is_last = is_first, is_last
+        self.layer_range = tuple(layer_range)
+        if is_first:
+            self.wte = VocabParallelEmbedding(cfg.vocab_size, cfg.hidden_size,
+                                              mesh, axis, dtype=dtype,
+                                              device=device,
+                                              init_seed=init_seed,
+                                              init_tag="wte")
+            self.wpe = nn.Parameter(_wpe_init(cfg, dtype, device, init_seed))
+        self.blocks = nn.ModuleList([
+            Block(cfg, mesh, axis, dtype, device, i, init_seed)
+            for i in range(layer_range[0], layer_range[1])
+        ])
+        if is_last:
+            self.ln_f = LayerNorm(cfg.hidden_size, cfg.layernorm_eps, dtype,
+                                  device)
+            self.lm_head = ColumnParallelLinear(cfg.hidden_size,
+                                                cfg.vocab_size, mesh, axis,
+                                                bias=False, dtype=dtype,
+                                                device=device,
+                                                init_seed=init_seed,
+                                                init_tag="lm_head")
+
+    def forward(self, x, microbatch):
+        ids, labels = microbatch["ids"], microbatch.get("labels")
+        if self.is_first:
+            S = ids.shape[1]
+            x = self.wte(ids) + self.wpe[:S]
+        for blk in self.blocks:
+            x = blk(x)
+        if not self.is_last:
+            return x
+        x = self.ln_f(x)
+        logits = self.lm_head(x)
+        N = logits.shape[0] * logits.shape[1]
+        logits = logits.reshape(N, -1)
+        vocab_start = self.lm_head.mesh.axis_index(self.axis) * \
+            self.lm_head.out_per_rank if self.mesh is not None else 0
+        per_tok = vocab_parallel_cross_entropy(
+            logits, labels.reshape(N), self.mesh, self.axis, vocab_start)
+        return per_tok.mean()
+
+
+def gpt_pipeline_spec(cfg: GPTConfig):
+    """PipelineModelSpec for the GPT family (used by PipeshardParallel)."""
+    from ..pipeline_parallel.spec import PipelineModelSpec
+
+    def build_stage(layer_range, is_first, is_last, mesh, axis, dtype,
+                    device):
+        return GPTStage(cfg, layer_range, is_first, is_last, mesh, axis,
+                        dtype, device)
+
+    def act_shape(microbatch):
+        ids = microbatch["ids"]
+        return (ids.shape[0], ids.shape[1], cfg.hidden_size)
+
+    # per-block cost uniform; embedding/LM head are pinned to the
+    # first/last stage by construction
+    return PipelineModelSpec(num_layers=cfg.num_layers,
+                             build_stage=build_stage, act_shape=act_shape)

@@ -65,18 +65,30 @@ def reduce_from_tp(x, mesh, axis):
     return _ReduceFromParallel.apply(x, mesh, axis)
 
 
+def tag_seed(init_seed: int, tag: str) -> int:
+    """Deterministic, position-independent seed for a named parameter:
+    lets pipeline stages built on different ranks draw identical weights
+    for the same logical layer (pipeline-vs-serial oracle tests)."""
+    import zlib
+    return (zlib.crc32(tag.encode()) ^ (init_seed * 0x9E3779B1)) % (2**31 - 1)
+
+
 @torch.no_grad()
 def _sharded_normal_(w: torch.Tensor, full_shape, shard_dim: int,
-                     shard_idx: int, num_shards: int, std: float):
+                     shard_idx: int, num_shards: int, std: float,
+                     seed: int = None):
     """Initialize `w` as shard `shard_idx` (along `shard_dim`) of a full
     tensor drawn from N(0, std).
 
-    The seed is drawn from the *global* RNG — consumed identically on every
-    rank since all ranks build the same module structure — so the union of
-    shards equals the serial init exactly (serial-vs-parallel tests depend
-    on this; cf. the reference's oracle pattern, alpa/testing.py:233).
+    With seed=None the seed is drawn from the *global* RNG — consumed
+    identically on every rank since all ranks build the same module
+    structure — so the union of shards equals the serial init exactly
+    (serial-vs-parallel tests depend on this; cf. the reference's oracle
+    pattern, alpa/testing.py:233).  Pipeline stages pass an explicit
+    tag_seed instead (position-independent).
     """
-    seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+    if seed is None:
+        seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
     gen_device = w.device if w.device.type == "cuda" else "cpu"
     g = torch.Generator(device=gen_device)
     g.manual_seed(seed)
@@ -101,7 +113,8 @@ class ColumnParallelLinear(nn.Module):
     def __init__(self, in_features: int, out_features: int,
                  mesh: Optional[DeviceMesh] = None, axis: int = 1,
                  bias: bool = True, gelu: bool = False,
-                 dtype=torch.float32, device=None):
+                 dtype=torch.float32, device=None, init_seed=None,
+                 init_tag: str = None):
         super().__init__()
         self.mesh, self.axis = mesh, axis
         tp = mesh.axis_size(axis) if mesh is not None else 1
@@ -114,8 +127,9 @@ class ColumnParallelLinear(nn.Module):
             torch.empty(self.out_per_rank, in_features, dtype=dtype,
                         device=device))
         idx = mesh.axis_index(axis) if (mesh is not None and mesh.is_member) else 0
+        seed = tag_seed(init_seed, init_tag) if init_tag is not None else None
         _sharded_normal_(self.weight, (out_features, in_features), 0,
-                         max(idx, 0), tp, 1.0 / math.sqrt(in_features))
+                         max(idx, 0), tp, 1.0 / math.sqrt(in_features), seed)
         if bias:
             self.bias = nn.Parameter(
                 torch.zeros(self.out_per_rank, dtype=dtype, device=device))
@@ -139,7 +153,8 @@ class RowParallelLinear(nn.Module):
 
     def __init__(self, in_features: int, out_features: int,
                  mesh: Optional[DeviceMesh] = None, axis: int = 1,
-                 bias: bool = True, dtype=torch.float32, device=None):
+                 bias: bool = True, dtype=torch.float32, device=None,
+                 init_seed=None, init_tag: str = None):
         super().__init__()
         self.mesh, self.axis = mesh, axis
         tp = mesh.axis_size(axis) if mesh is not None else 1
@@ -151,8 +166,9 @@ class RowParallelLinear(nn.Module):
             torch.empty(out_features, self.in_per_rank, dtype=dtype,
                         device=device))
         idx = mesh.axis_index(axis) if (mesh is not None and mesh.is_member) else 0
+        seed = tag_seed(init_seed, init_tag) if init_tag is not None else None
         _sharded_normal_(self.weight, (out_features, in_features), 1,
-                         max(idx, 0), tp, 1.0 / math.sqrt(in_features))
+                         max(idx, 0), tp, 1.0 / math.sqrt(in_features), seed)
         if bias:
             self.bias = nn.Parameter(
                 torch.zeros(out_features, dtype=dtype, device=device))
@@ -175,7 +191,8 @@ class VocabParallelEmbedding(nn.Module):
 
     def __init__(self, num_embeddings: int, embedding_dim: int,
                  mesh: Optional[DeviceMesh] = None, axis: int = 1,
-                 dtype=torch.float32, device=None):
+                 dtype=torch.float32, device=None, init_seed=None,
+                 init_tag: str = None):
         super().__init__()
         self.mesh, self.axis = mesh, axis
         tp = mesh.axis_size(axis) if mesh is not None else 1
@@ -188,8 +205,9 @@ class VocabParallelEmbedding(nn.Module):
         self.weight = nn.Parameter(
             torch.empty(self.vocab_per_rank, embedding_dim, dtype=dtype,
                         device=device))
+        seed = tag_seed(init_seed, init_tag) if init_tag is not None else None
         _sharded_normal_(self.weight, (num_embeddings, embedding_dim), 0,
-                         idx, tp, 0.02)
+                         idx, tp, 0.02, seed)
 
     def forward(self, ids: torch.Tensor):
         if self.mesh is None or self.mesh.axis_size(self.axis) == 1:

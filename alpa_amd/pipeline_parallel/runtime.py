@@ -1,0 +1,234 @@
+"""Pipeline execution engine: static 1F1B/GPipe microbatch runtime over
+RCCL p2p (xGMI) between per-stage submeshes.
+
+MI355X-native replacement for the reference's instruction-list interpreter
+(``runtime_emitter.py:258`` emits RUN/SEND/RECV/FREE per worker;
+``pipeshard_executable.py:489`` interprets them).  Here every rank executes
+the same statically-known control flow for its stage; cross-stage traffic
+is fused send+recv pairs (one batch_isend_irecv per adjacent-stage
+exchange, both sides posting simultaneously — the deadlock-freedom
+discipline the reference gets from sorted resharding task order,
+SURVEY.md §5.2).
+
+Stage meshes are uniform (same (dp, tp) shape): the p2p peer of a rank is
+the same-coordinate rank of the adjacent stage mesh, so activations cross
+stages with zero resharding.  (Heterogeneous stage meshes go through the
+cross-mesh resharding module.)
+"""
+from __future__ import annotations
+
+from typing import Any, Callable, Dict, List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+from ..mesh import DeviceMesh, is_distributed, rank
+from ..parallel.grad_sync import GradSynchronizer
+from . import schedules
+
+
+class PipelineEngine:
+    """Runs one training (or inference) step for this rank's stage."""
+
+    def __init__(self, stage_module: torch.nn.Module, stage_idx: int,
+                 num_stages: int, stage_mesh: DeviceMesh,
+                 prev_peer: Optional[int], next_peer: Optional[int],
+                 num_microbatches: int, act_shape: Tuple[int, ...],
+                 act_dtype: torch.dtype, schedule: str = "1f1b",
+                 grad_sync: Optional[GradSynchronizer] = None,
+                 loss_scale_ranks: Optional[List[int]] = None):
+        self.stage = stage_module
+        self.s = stage_idx
+        self.P = num_stages
+        self.mesh = stage_mesh
+        self.prev_peer = prev_peer  # global rank or None
+        self.next_peer = next_peer
+        self.M = num_microbatches
+        self.act_shape = act_shape
+        self.act_dtype = act_dtype
+        self.schedule_name = schedule
+        self.grad_sync = grad_sync
+        self.is_first = stage_idx == 0
+        self.is_last = stage_idx == num_stages - 1
+        self.device = next(stage_module.parameters()).device
+        self._act_shape_fn = None  # lazy: microbatch -> act shape
+        self.loss_src_rank = 0
+
+    # ------------------------- p2p primitives -------------------------
+    def _p2p(self, ops: List[dist.P2POp]):
+        if not ops:
+            return
+        works = dist.batch_isend_irecv(ops)
+        for w in works:
+            w.wait()
+
+    def _recv_forward(self) -> torch.Tensor:
+        buf = torch.empty(self.act_shape, dtype=self.act_dtype,
+                          device=self.device)
+        self._p2p([dist.P2POp(dist.irecv, buf, self.prev_peer)])
+        return buf
+
+    def _send_forward(self, y: torch.Tensor):
+        self._p2p([dist.P2POp(dist.isend, y.contiguous(), self.next_peer)])
+
+    def _recv_backward(self) -> torch.Tensor:
+        buf = torch.empty(self.act_shape, dtype=self.act_dtype,
+                          device=self.device)
+        self._p2p([dist.P2POp(dist.irecv, buf, self.next_peer)])
+        return buf
+
+    def _send_backward(self, g: torch.Tensor):
+        self._p2p([dist.P2POp(dist.isend, g.contiguous(), self.prev_peer)])
+
+    def _send_forward_recv_backward(self, y: torch.Tensor) -> torch.Tensor:
+        buf = torch.empty(self.act_shape, dtype=self.act_dtype,
+                          device=self.device)
+        self._p2p([dist.P2POp(dist.isend, y.contiguous(), self.next_peer),
+                   dist.P2POp(dist.irecv, buf, self.next_peer)])
+        return buf
+
+    def _send_backward_recv_forward(self, g: torch.Tensor) -> torch.Tensor:
+        buf = torch.empty(self.act_shape, dtype=self.act_dtype,
+                          device=self.device)
+        self._p2p([dist.P2POp(dist.isend, g.contiguous(), self.prev_peer),
+                   dist.P2POp(dist.irecv, buf, self.prev_peer)])
+        return buf
+
+    # ------------------------- compute steps -------------------------
+    def _forward_step(self, mb_idx: int, x: Optional[torch.Tensor],
+                      microbatches: List[Any]):
+        """Returns (input, output): output is loss on the last stage."""
+        if not self.is_first:
+            x = x.requires_grad_(True)
+        out = self.stage(x, microbatches[mb_idx])
+        return x, out
+
+    def _backward_step(self, x, out, out_grad,
+                       is_last_bwd: bool) -> Optional[torch.Tensor]:
+        if self.grad_sync is not None:
+            self.grad_sync.begin_microbatch(is_last=is_last_bwd)
+        if self.is_last:
+            out.backward()  # out is the microbatch loss
+        else:
+            torch.autograd.backward(out, grad_tensors=out_grad)
+        return None if self.is_first else x.grad
+
+    # ------------------------- the step -------------------------
+    def train_step(self, microbatches: List[Any]) -> torch.Tensor:
+        """1F1B (or GPipe) over the local microbatch list.  Returns the mean
+        loss (valid on last-stage ranks; others get the broadcast value via
+        PipelineTrainer)."""
+        assert len(microbatches) == self.M
+        if self.act_shape is None:
+            self.act_shape = tuple(self._act_shape_fn(microbatches[0]))
+        P, M, s = self.P, self.M, self.s
+        if self.grad_sync is not None:
+            self.grad_sync.zero_grads()
+            self.grad_sync.begin_microbatch(is_last=False)
+
+        if self.schedule_name == "gpipe":
+            return self._gpipe_step(microbatches)
+
+        warmup = min(P - s - 1, M)
+        steady = M - warmup
+        fwd_i = 0
+        bwd_i = 0
+        pending: List[Tuple[Optional[torch.Tensor], torch.Tensor]] = []
+        losses = []
+
+        # ---- warmup forwards ----
+        for _ in range(warmup):
+            x = None if self.is_first else self._recv_forward()
+            x, out = self._forward_step(fwd_i, x, microbatches)
+            if not self.is_last:
+                self._send_forward(out)
+            else:
+                losses.append(out.detach())
+            pending.append((x, out))
+            fwd_i += 1
+
+        x_next = None
+        if steady > 0 and not self.is_first:
+            x_next = self._recv_forward()
+
+        # ---- steady 1F1B ----
+        for i in range(steady):
+            x, out = self._forward_step(fwd_i, x_next, microbatches)
+            fwd_i += 1
+            if self.is_last:
+                losses.append(out.detach())
+                out_grad = None
+            else:
+                out_grad = self._send_forward_recv_backward(out)
+            pending.append((x, out))
+            px, pout = pending.pop(0)
+            bwd_i += 1
+            in_grad = self._backward_step(px, pout, out_grad,
+                                          is_last_bwd=(bwd_i == M))
+            last_iter = (i == steady - 1)
+            if self.is_first:
+                pass
+            elif last_iter:
+                self._send_backward(in_grad)
+            else:
+                x_next = self._send_backward_recv_forward(in_grad)
+
+        # ---- cooldown backwards ----
+        for _ in range(warmup):
+            out_grad = None if self.is_last else self._recv_backward()
+            px, pout = pending.pop(0)
+            bwd_i += 1
+            in_grad = self._backward_step(px, pout, out_grad,
+                                          is_last_bwd=(bwd_i == M))
+            if not self.is_first:
+                self._send_backward(in_grad)
+
+        if self.grad_sync is not None:
+            self.grad_sync.finish()
+        if losses:
+            return torch.stack(losses).mean()
+        return torch.zeros((), device=self.device)
+
+    def _gpipe_step(self, microbatches) -> torch.Tensor:
+        M = self.M
+        pending = []
+        losses = []
+        for i in range(M):
+            x = None if self.is_first else self._recv_forward()
+            x, out = self._forward_step(i, x, microbatches)
+            if not self.is_last:
+                self._send_forward(out)
+            else:
+                losses.append(out.detach())
+            pending.append((x, out))
+        for i in reversed(range(M)):
+            out_grad = None if self.is_last else self._recv_backward()
+            px, pout = pending[i]
+            in_grad = self._backward_step(px, pout, out_grad,
+                                          is_last_bwd=(i == 0))
+            if not self.is_first:
+                self._send_backward(in_grad)
+        pending.clear()
+        if self.grad_sync is not None:
+            self.grad_sync.finish()
+        if losses:
+            return torch.stack(losses).mean()
+        return torch.zeros((), device=self.device)
+
+    # GPipe backward consumes grads in reverse mb order — receive order from
+    # the next stage is also reverse, consistent across stages.
+
+    def inference_step(self, microbatches: List[Any]) -> List[torch.Tensor]:
+        """Forward-only pipeline; returns last-stage outputs per microbatch."""
+        outs = []
+        if self.act_shape is None:
+            self.act_shape = tuple(self._act_shape_fn(microbatches[0]))
+        with torch.no_grad():
+            for i in range(self.M):
+                x = None if self.is_first else self._recv_forward()
+                out = self.stage(x, microbatches[i])
+                if not self.is_last:
+                    self._send_forward(out)
+                else:
+                    outs.append(out)
+        return outs

@@ -1,0 +1,35 @@
+"""Pipeline model specification: how a model family exposes its layer
+structure to the pipeline compiler.
+
+The reference slices a traced jaxpr at pipeline markers
+(``computation.py:387`` slice_closed_jaxpr_by_full_pipeline_marks); in the
+module-level world the model family provides a stage builder and the
+compiler decides the layer->stage assignment (auto clustering DP or
+manual).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Callable, List, Optional, Sequence, Tuple
+
+import torch
+
+
+@dataclass
+class PipelineModelSpec:
+    """Everything the pipeline compiler needs from a model family.
+
+    - ``num_layers``: number of cluster-able layers (transformer blocks).
+    - ``build_stage(layer_range, is_first, is_last, mesh, axis, dtype,
+      device) -> nn.Module``: builds this rank's stage module.  The stage
+      forward signature is ``forward(x, microbatch)`` where x is None on
+      the first stage; the last stage returns the microbatch loss.
+    - ``act_shape(microbatch) -> tuple``: shape of the activation crossing
+      stage boundaries for one microbatch.
+    - ``layer_costs``: optional per-layer FLOPs estimate for the
+      clustering DP (uniform if None).
+    """
+    num_layers: int
+    build_stage: Callable[..., torch.nn.Module]
+    act_shape: Callable[[Any], Tuple[int, ...]]
+    layer_costs: Optional[Sequence[float]] = None
