@@ -59,6 +59,10 @@ class ShardParallel(ParallelMethod):
     logical_mesh_shape: Optional[Tuple[int, int]] = None
     auto_sharding_option: AutoShardingOption = field(
         default_factory=AutoShardingOption)
+    #: model description for the auto-sharding ILP when no manual mesh
+    #: shape is given, e.g. {"family": "gpt", "hidden": 2560, "layers": 32,
+    #: "vocab": 51200, "tokens": <tokens per microbatch>}
+    model_hint: Optional[dict] = None
 
     def resolve_mesh(self) -> DeviceMesh:
         n = world_size()

@@ -24,8 +24,24 @@ def factorizations(n: int) -> List[Tuple[int, int]]:
 
 
 def choose_mesh_shape(method, n: int) -> Tuple[int, int]:
-    """Default: pure DP (batch-dim sharding) — the ILP-backed choice runs in
-    compile() when a graph is available."""
+    """ILP-backed (dp, tp) choice when the method carries a model hint;
+    pure DP otherwise."""
     if n == 1:
         return (1, 1)
+    hint = getattr(method, "model_hint", None)
+    if hint and hint.get("family") == "gpt":
+        from .auto_sharding import plan_to_logical_shape, solve_gpt_sharding
+        budget = hint.get("memory_budget")
+        # homogeneous layers: 2 fix the per-layer optimum — but the memory
+        # constraint needs the full depth, so keep all layers when a budget
+        # is set
+        layers = int(hint.get("layers", 2)) if budget else \
+            min(int(hint.get("layers", 2)), 2)
+        plan = solve_gpt_sharding(
+            n, hidden=hint["hidden"], layers=layers,
+            vocab=hint["vocab"], tokens=hint["tokens"],
+            memory_budget=budget,
+            force_data_parallel=method.auto_sharding_option
+            .force_data_parallel)
+        return plan_to_logical_shape(plan)
     return (n, 1)
