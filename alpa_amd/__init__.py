@@ -9,6 +9,8 @@ process per GPU.  No JAX/XLA, no Ray, no CUDA shims.
 """
 
 from .api import TrainState, init, parallelize, shutdown
+from .serialization import (restore_checkpoint, restore_train_state,
+                            save_checkpoint, save_train_state)
 from .global_env import global_config
 from .mesh import (DeviceMesh, VirtualMesh, device, full_mesh, full_virtual_mesh,
                    get_device_mesh, init_distributed, local_rank, rank,

@@ -1,0 +1,287 @@
+"""Checkpoint save/restore in the reference's on-disk format.
+
+Format (mirrors ``alpa/serialization.py:75-189`` + the shard writer in
+``device_mesh.py:302-338,1582-1616``):
+
+  <ckpt_dir>/checkpoint_<step>        msgpack-packed state tree whose
+                                      tensor leaves are replaced by
+                                      relative directory paths
+  <ckpt_dir>/<tensor_dir>/shard_<host>.<k>    .npy shard files
+  <ckpt_dir>/<tensor_dir>/metadata_<host>     pickle:
+      {"global_shape", "dtype", "shard_names", "shard_indices"}
+
+"host" here is the rank (one process per GPU).  Restore reassembles each
+rank's target slice from whichever saved shards overlap it — resharding
+on load, so a checkpoint written under one (dp, tp) placement loads under
+another (reference restore_checkpoint:137 guided by PlacementSpec).
+"""
+from __future__ import annotations
+
+import os
+import pickle
+from typing import Any, Dict, List, Optional, Sequence, Tuple
+
+import msgpack
+import numpy as np
+import torch
+
+from .mesh import is_distributed, rank, world_size
+
+
+class ShardSpec:
+    """This rank's slice of a global tensor: (global_shape, index)."""
+
+    def __init__(self, global_shape: Tuple[int, ...],
+                 index: Tuple[slice, ...], is_writer: bool = True):
+        self.global_shape = tuple(global_shape)
+        self.index = tuple(index)
+        self.is_writer = is_writer
+
+    @staticmethod
+    def full(shape, is_writer: bool = True) -> "ShardSpec":
+        return ShardSpec(tuple(shape), tuple(slice(0, s) for s in shape),
+                         is_writer)
+
+
+def _flatten(tree: Any, prefix: str = "") -> Dict[str, Any]:
+    out = {}
+    if isinstance(tree, dict):
+        for k, v in tree.items():
+            out.update(_flatten(v, f"{prefix}{k}." if prefix or True else k))
+    else:
+        out[prefix[:-1]] = tree
+    return out
+
+
+def _tensor_dir_name(path_key: str) -> str:
+    return path_key.replace("/", "_").replace(".", "_")
+
+
+def _replace_leaves(tree: Any, fn) -> Any:
+    if isinstance(tree, dict):
+        return {k: _replace_leaves(v, fn) for k, v in tree.items()}
+    return fn(tree)
+
+
+def save_checkpoint(ckpt_dir: str, state_tree: Any, step: int,
+                    shard_specs: Optional[Dict[str, ShardSpec]] = None
+                    ) -> None:
+    """Save a (possibly nested-dict) state tree.  Tensor leaves become
+    shard directories; non-tensors are stored inline in the msgpack tree.
+
+    shard_specs maps flattened leaf paths (dot-joined) to ShardSpec; absent
+    leaves are treated as replicated full tensors written by rank 0 only.
+    """
+    shard_specs = shard_specs or {}
+    host = rank()
+    os.makedirs(ckpt_dir, exist_ok=True)
+
+    flat_pos: List[str] = []
+
+    def leaf_to_entry(path: str, leaf):
+        if not torch.is_tensor(leaf):
+            return leaf
+        dirname = _tensor_dir_name(path)
+        spec = shard_specs.get(path)
+        if spec is None:
+            spec = ShardSpec.full(leaf.shape, is_writer=(host == 0 or
+                                                         not is_distributed()))
+        tdir = os.path.join(ckpt_dir, dirname)
+        os.makedirs(tdir, exist_ok=True)
+        if spec.is_writer:
+            arr = leaf.detach().to(torch.float32).cpu().numpy() \
+                if leaf.dtype == torch.bfloat16 \
+                else leaf.detach().cpu().numpy()
+            shard_name = f"shard_{host}.0"
+            np.save(os.path.join(tdir, shard_name + ".npy"), arr)
+            meta = {
+                "global_shape": spec.global_shape,
+                "dtype": str(leaf.dtype).replace("torch.", ""),
+                "shard_names": [shard_name],
+                "shard_indices": [tuple((s.start, s.stop)
+                                        for s in spec.index)],
+            }
+            with open(os.path.join(tdir, f"metadata_{host}"), "wb") as f:
+                pickle.dump(meta, f)
+        return {"__tensor_dir__": dirname}
+
+    def walk(tree, prefix=""):
+        if isinstance(tree, dict):
+            return {k: walk(v, f"{prefix}{k}.") for k, v in tree.items()}
+        return leaf_to_entry(prefix[:-1], tree)
+
+    packed_tree = walk(state_tree)
+    if host == 0 or not is_distributed():
+        with open(os.path.join(ckpt_dir, f"checkpoint_{step}"), "wb") as f:
+            f.write(msgpack.packb(packed_tree, use_bin_type=True))
+
+
+def _load_slice(tdir: str, want_index: Tuple[slice, ...],
+                global_shape, np_dtype) -> np.ndarray:
+    """Assemble want_index of the global tensor from saved shards
+    (resharding on load)."""
+    out_shape = tuple(s.stop - s.start for s in want_index)
+    out = np.zeros(out_shape, dtype=np_dtype)
+    filled = np.zeros(out_shape, dtype=bool) if out.size else None
+    for fname in sorted(os.listdir(tdir)):
+        if not fname.startswith("metadata_"):
+            continue
+        with open(os.path.join(tdir, fname), "rb") as f:
+            meta = pickle.load(f)
+        for shard_name, idx in zip(meta["shard_names"],
+                                   meta["shard_indices"]):
+            # intersection of shard idx with want_index
+            inter = []
+            src = []
+            dst = []
+            ok = True
+            for d, ((a0, a1), w) in enumerate(zip(idx, want_index)):
+                lo = max(a0, w.start)
+                hi = min(a1, w.stop)
+                if lo >= hi:
+                    ok = False
+                    break
+                src.append(slice(lo - a0, hi - a0))
+                dst.append(slice(lo - w.start, hi - w.start))
+            if not ok:
+                continue
+            arr = np.load(os.path.join(tdir, shard_name + ".npy"))
+            out[tuple(dst)] = arr[tuple(src)]
+            if filled is not None:
+                filled[tuple(dst)] = True
+    if filled is not None and not filled.all():
+        raise RuntimeError(f"checkpoint {tdir}: missing regions for "
+                           f"{want_index}")
+    return out
+
+
+def restore_checkpoint(ckpt_dir: str, step: int, target_tree: Any,
+                       shard_specs: Optional[Dict[str, ShardSpec]] = None
+                       ) -> Any:
+    """Load into `target_tree` (tensors are filled in place with this
+    rank's slice per shard_specs; non-tensor leaves are replaced).
+    Returns the tree."""
+    shard_specs = shard_specs or {}
+    with open(os.path.join(ckpt_dir, f"checkpoint_{step}"), "rb") as f:
+        packed_tree = msgpack.unpackb(f.read(), raw=False)
+
+    def walk(target, packed, prefix=""):
+        if isinstance(target, dict):
+            return {k: walk(v, packed[k], f"{prefix}{k}.")
+                    for k, v in target.items()}
+        path = prefix[:-1]
+        if torch.is_tensor(target):
+            assert isinstance(packed, dict) and "__tensor_dir__" in packed, \
+                f"{path}: checkpoint has no tensor dir"
+            tdir = os.path.join(ckpt_dir, packed["__tensor_dir__"])
+            spec = shard_specs.get(path) or ShardSpec.full(target.shape)
+            want = spec.index
+            np_dtype = np.float32 if target.dtype in (torch.bfloat16,
+                                                      torch.float32) \
+                else target.detach().cpu().numpy().dtype
+            arr = _load_slice(tdir, want, spec.global_shape, np_dtype)
+            with torch.no_grad():
+                target.copy_(torch.from_numpy(arr).to(target.device,
+                                                      target.dtype))
+            return target
+        return packed
+
+    return walk(target_tree, packed_tree)
+
+
+# ----------------------------------------------------------------------
+# TrainState integration
+# ----------------------------------------------------------------------
+
+
+def model_shard_specs(model: torch.nn.Module) -> Dict[str, ShardSpec]:
+    """Derive per-parameter ShardSpecs from the parallel layer modules
+    (Column/Row/VocabParallel know their global shape + slice; everything
+    else is replicated, written by the global writer rank of its dp
+    group)."""
+    from .parallel.layers import (ColumnParallelLinear, RowParallelLinear,
+                                  VocabParallelEmbedding)
+    specs: Dict[str, ShardSpec] = {}
+    for mod_name, mod in model.named_modules():
+        prefix = f"{mod_name}." if mod_name else ""
+        mesh = getattr(mod, "mesh", None)
+        axis = getattr(mod, "axis", 1)
+        tp = mesh.axis_size(axis) if mesh is not None else 1
+        idx = mesh.axis_index(axis) if (mesh is not None and mesh.is_member) \
+            else 0
+        # dp writer: only the first rank along every non-tp axis writes
+        dp_writer = True
+        if mesh is not None and mesh.is_member:
+            other = 1 - axis
+            dp_writer = mesh.axis_index(other) == 0
+        if isinstance(mod, ColumnParallelLinear) and tp > 1:
+            o = mod.out_per_rank
+            specs[prefix + "weight"] = ShardSpec(
+                (mod.out_features, mod.in_features),
+                (slice(idx * o, (idx + 1) * o), slice(0, mod.in_features)),
+                dp_writer)
+            if mod.bias is not None:
+                specs[prefix + "bias"] = ShardSpec(
+                    (mod.out_features,), (slice(idx * o, (idx + 1) * o),),
+                    dp_writer)
+        elif isinstance(mod, RowParallelLinear) and tp > 1:
+            i = mod.in_per_rank
+            specs[prefix + "weight"] = ShardSpec(
+                (mod.out_features, mod.in_features),
+                (slice(0, mod.out_features), slice(idx * i, (idx + 1) * i)),
+                dp_writer)
+            if mod.bias is not None:
+                specs[prefix + "bias"] = ShardSpec(
+                    (mod.out_features,), (slice(0, mod.out_features),),
+                    dp_writer and idx == 0)
+        elif isinstance(mod, VocabParallelEmbedding) and tp > 1:
+            v = mod.vocab_per_rank
+            emb = mod.weight.shape[1]
+            specs[prefix + "weight"] = ShardSpec(
+                (mod.num_embeddings, emb),
+                (slice(idx * v, (idx + 1) * v), slice(0, emb)), dp_writer)
+    # default for unsharded params: replicated; written once per dp group
+    for name, p in model.named_parameters():
+        if name not in specs:
+            writer = rank() == 0 or not is_distributed()
+            # in a TP group, tp rank 0 of dp rank 0 writes
+            specs[name] = ShardSpec.full(p.shape, writer)
+    return specs
+
+
+def _train_state_tree_and_specs(state):
+    """(tree, specs) for a TrainState: params + AdamW moments (the moments
+    share the parameter sharding).  ZeRO-2's bucket-sharded moments are
+    topology-specific and handled by its own state_dict."""
+    specs = model_shard_specs(state.model)
+    names = [n for n, _ in state.model.named_parameters()]
+    tree = {"params": dict(state.model.state_dict()),
+            "step": state.step_count}
+    full_specs = {f"params.{k}": v for k, v in specs.items()}
+    opt_sd = state.optimizer.state_dict()
+    if "exp_avgs" in opt_sd and len(opt_sd["exp_avgs"]) == len(names):
+        tree["opt_m"] = dict(zip(names, opt_sd["exp_avgs"]))
+        tree["opt_v"] = dict(zip(names, opt_sd["exp_avg_sqs"]))
+        for n in names:
+            full_specs[f"opt_m.{n}"] = specs[n]
+            full_specs[f"opt_v.{n}"] = specs[n]
+    return tree, full_specs
+
+
+def save_train_state(ckpt_dir: str, state, step: Optional[int] = None
+                     ) -> None:
+    """Save model + optimizer of a TrainState (reference
+    save_checkpoint, serialization.py:75)."""
+    step = step if step is not None else state.step_count
+    tree, full_specs = _train_state_tree_and_specs(state)
+    save_checkpoint(ckpt_dir, tree, step, full_specs)
+
+
+def restore_train_state(ckpt_dir: str, state, step: int) -> None:
+    """Restore in place, resharding to this rank's placement (reference
+    restore_checkpoint, serialization.py:137)."""
+    tree, full_specs = _train_state_tree_and_specs(state)
+    loaded = restore_checkpoint(ckpt_dir, step, tree, full_specs)
+    if isinstance(loaded.get("step"), int):
+        state.optimizer.step_count = loaded["step"]
+        state.step_count = loaded["step"]
