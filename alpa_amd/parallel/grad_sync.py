@@ -36,6 +36,8 @@ class _Bucket:
     params: List[torch.nn.Parameter] = field(default_factory=list)
     numel: int = 0
     flat: Optional[torch.Tensor] = None
+    #: expert-parallel params: grads complete locally, no dp collective
+    no_sync: bool = False
     # per-step state
     pending: int = 0
     work: Optional[object] = None  # dist.Work
@@ -61,19 +63,27 @@ class GradSynchronizer:
 
         params = [p for p in params if p.requires_grad]
         self.params = params
-        # reverse order ≈ backward completion order
+        # reverse order ≈ backward completion order; expert-parallel params
+        # (marked by ExpertParallelMLP) go to dedicated no-sync buckets
         order = list(reversed(params))
         self.buckets: List[_Bucket] = []
         cur = _Bucket()
+        cur_local = _Bucket(no_sync=True)
         for p in order:
+            tgt = cur_local if getattr(p, "_expert_parallel", False) else cur
             esize = p.element_size()
-            if cur.params and (cur.numel + p.numel()) * esize > bucket_bytes:
-                self.buckets.append(cur)
-                cur = _Bucket()
-            cur.params.append(p)
-            cur.numel += p.numel()
+            if tgt.params and (tgt.numel + p.numel()) * esize > bucket_bytes:
+                self.buckets.append(tgt)
+                if tgt is cur:
+                    cur = tgt = _Bucket()
+                else:
+                    cur_local = tgt = _Bucket(no_sync=True)
+            tgt.params.append(p)
+            tgt.numel += p.numel()
         if cur.params:
             self.buckets.append(cur)
+        if cur_local.params:
+            self.buckets.append(cur_local)
 
         # allocate flat buffers + wire .grad views
         self._param_bucket = {}
@@ -118,7 +128,7 @@ class GradSynchronizer:
             self._launch(b)
 
     def _launch(self, b: _Bucket):
-        if self.dp == 1 or not is_distributed():
+        if b.no_sync or self.dp == 1 or not is_distributed():
             return
         group = self.mesh.axis_group(self.axis)
         stream = self.mesh.comm_stream
