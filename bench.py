@@ -47,6 +47,8 @@ def parse_args():
     p.add_argument("--dp", type=int, default=0, help="data-parallel degree "
                    "(0 = all GPUs)")
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
+    p.add_argument("--parallel", choices=["auto", "manual"], default="auto",
+                   help="auto = ILP auto-sharding picks (dp, tp)")
     p.add_argument("--seq", type=int, default=1024)
     return p.parse_args()
 
@@ -72,12 +74,26 @@ def main():
                         seq_len=64, vocab_size=512)
         batch_per_gpu = 4
 
-    dp = args.dp if args.dp > 0 else n // args.tp
-    tp = args.tp
-    assert dp * tp == n, (dp, tp, n)
-
-    method = aa.ShardParallel(num_micro_batches=args.nmb,
-                              logical_mesh_shape=(dp, tp))
+    if args.parallel == "auto" and args.dp == 0 and n > 1:
+        # ILP auto-sharding picks (dp, tp) from the op graph + xGMI costs
+        state_bytes = 12.0 * cfg.num_params()
+        hint = {
+            "family": "gpt", "hidden": cfg.hidden_size,
+            "layers": cfg.num_layers, "vocab": cfg.vocab_size,
+            "tokens": args.batch_per_gpu * n * cfg.seq_len // args.nmb,
+            "memory_budget": (0.9 * 288e9
+                              if state_bytes > 0.5 * 288e9 else None),
+        }
+        method = aa.ShardParallel(num_micro_batches=args.nmb,
+                                  model_hint=hint)
+        mesh_probe = method.resolve_mesh()
+        dp, tp = mesh_probe.shape
+    else:
+        dp = args.dp if args.dp > 0 else n // args.tp
+        tp = args.tp
+        assert dp * tp == n, (dp, tp, n)
+        method = aa.ShardParallel(num_micro_batches=args.nmb,
+                                  logical_mesh_shape=(dp, tp))
 
     def build(mesh=None, axis=1, dtype=torch.float32, device=None):
         torch.manual_seed(1234)
@@ -88,15 +104,16 @@ def main():
         lambda model, b: model.loss(b["ids"], b["labels"]), method=method)
 
     # synthetic data of the benchmark shape, one local batch per dp rank
-    g = torch.Generator().manual_seed(4321 + aa.rank())
+    # (tp ranks of the same dp group must see identical data)
+    mesh = state.mesh
+    dp_idx = mesh.axis_index(0) if mesh is not None and mesh.is_member else 0
+    g = torch.Generator().manual_seed(4321 + max(dp_idx, 0))
     ids = torch.randint(0, cfg.vocab_size, (batch_per_gpu, cfg.seq_len),
                         generator=g)
     labels = torch.randint(0, cfg.vocab_size, (batch_per_gpu, cfg.seq_len),
                            generator=g)
     dev = aa.device()
     batch = {"ids": ids.to(dev), "labels": labels.to(dev)}
-
-    mesh = state.mesh
 
     def sync():
         if mesh is not None:
@@ -148,6 +165,7 @@ def main():
                 "global_batch": global_batch, "seq_len": cfg.seq_len,
                 "num_micro_batches": args.nmb,
                 "parallelism": f"dp{dp}" + (f"tp{tp}" if tp > 1 else ""),
+                "auto_sharded": args.parallel == "auto",
                 "loss": float(loss),
             },
         }
