@@ -160,18 +160,22 @@ class _FlashAttentionQKV(torch.autograd.Function):
         v = qkv5[:, :, :, 2].permute(0, 2, 1, 3)
         do4 = do.view(B, S, h, d).permute(0, 2, 1, 3)
         o4 = o_buf.view(B, S, h, d).permute(0, 2, 1, 3)
+        dqkv = torch.empty_like(qkv)
+        dqkv5 = dqkv.view(B, S, h, 3, d)
         if use_hip(qkv):
-            dq, dk, dv = hip_ops().attn_bwd(do4, q, k, v, o4, lse, causal,
-                                            scale)
+            # fused MFMA bwd writes straight into the packed dqkv views
+            hip_ops().attn_bwd_out(
+                do4, q, k, v, o4, lse,
+                dqkv5[:, :, :, 0].permute(0, 2, 1, 3),
+                dqkv5[:, :, :, 1].permute(0, 2, 1, 3),
+                dqkv5[:, :, :, 2].permute(0, 2, 1, 3), causal, scale)
         else:
             dq, dk, dv = ref.attention_bwd(
                 do4.contiguous(), q.contiguous(), k.contiguous(),
                 v.contiguous(), o4.contiguous(), lse, causal, scale)
-        dqkv = torch.empty_like(qkv)
-        dqkv5 = dqkv.view(B, S, h, 3, d)
-        dqkv5[:, :, :, 0].copy_(dq.permute(0, 2, 1, 3))
-        dqkv5[:, :, :, 1].copy_(dk.permute(0, 2, 1, 3))
-        dqkv5[:, :, :, 2].copy_(dv.permute(0, 2, 1, 3))
+            dqkv5[:, :, :, 0].copy_(dq.permute(0, 2, 1, 3))
+            dqkv5[:, :, :, 1].copy_(dk.permute(0, 2, 1, 3))
+            dqkv5[:, :, :, 2].copy_(dv.permute(0, 2, 1, 3))
         return dqkv, None, None, None
 
 

@@ -289,3 +289,454 @@ extern "C" hipError_t launch_mfma_probe(const void* a, const void* b,
                                                       (const short*)b, d);
   return hipGetLastError();
 }
+
+// ===========================================================================
+// Flash attention BACKWARD (bf16, causal/full), deterministic two-kernel
+// split (no atomics): dq kernel over q-blocks, dk/dv kernel over kv-blocks,
+// each recomputing P from (q, k, lse).  Same MFMA fragment layouts as fwd.
+//
+// delta = rowsum(dO * O) is precomputed by attn_bwd_preprocess.
+// ===========================================================================
+
+// All tensors logically [B, H, S, D] with arbitrary B/H/S strides and a
+// contiguous D (so the packed qkv/dqkv layouts feed the kernels directly).
+struct AttnBwdStrides {
+  int64_t qb, qh, qs;
+  int64_t kb, kh, ks;
+  int64_t vb, vh, vs;
+  int64_t dob, doh, dos;
+  int64_t dqb, dqh, dqs;
+  int64_t dkb, dkh, dks;
+  int64_t dvb, dvh, dvs;
+};
+
+__global__ void attn_bwd_preprocess_kernel(const short* __restrict__ dout,
+                                           const short* __restrict__ o,
+                                           float* __restrict__ delta,
+                                           int64_t rows, int H, int S, int D,
+                                           int64_t dob, int64_t doh,
+                                           int64_t dos, int64_t ob,
+                                           int64_t oh, int64_t os) {
+  // one wave per row
+  int64_t row = (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= rows) return;
+  int lane = threadIdx.x & 63;
+  int64_t sidx = row % S, h = (row / S) % H, b = row / ((int64_t)S * H);
+  const short* dr = dout + b * dob + h * doh + sidx * dos;
+  const short* orow = o + b * ob + h * oh + sidx * os;
+  float s = 0.f;
+  for (int i = lane * 4; i < D; i += 64 * 4) {
+    bf16x4 dv = *reinterpret_cast<const bf16x4*>(dr + i);
+    bf16x4 ov = *reinterpret_cast<const bf16x4*>(orow + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) s += bf2f(dv[j]) * bf2f(ov[j]);
+  }
+  s = wave_reduce_sum(s);
+  if (lane == 0) delta[row] = s;
+}
+
+// ---------------------------------------------------------------- dq kernel
+// Block: 4 waves, 64 q rows; loops kv blocks of 64.
+//   S = Q K^T          (A=Q regs, B=K_lds row-major)
+//   P = exp(S*? - lse) (lse already includes the scale from fwd)
+//   dP = dO V^T        (A=dO regs, B=V_lds row-major)
+//   dS = P*(dP-delta)*scale
+//   dQ += dS K         (A=dS via p_lds, B=Kt_lds transposed)
+template <int Dp>
+__global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dq_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, const short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    short* __restrict__ dq, int H, int S, int Skv, int D, float scale,
+    int causal, AttnBwdStrides st) {
+  constexpr int KD = Dp / 32;          // k-steps over head dim
+  constexpr int NT = 4;                // 64 kv per tile
+  constexpr int DT = Dp / 16;
+
+  const int qb = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int batch = bh / H, head = bh % H;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 15, hi = lane >> 4;
+
+  const short* qp = q + batch * st.qb + head * st.qh;
+  const short* kp = k + batch * st.kb + head * st.kh;
+  const short* vp = v + batch * st.vb + head * st.vh;
+  const short* dop = dout + batch * st.dob + head * st.doh;
+  short* dqp = dq + batch * st.dqb + head * st.dqh;
+  const int q_row0 = qb * 64 + wave * 16;
+
+  __shared__ short k_lds[64][Dp];
+  __shared__ short kt_lds[Dp][64];
+  __shared__ short v_lds[64][Dp];
+  __shared__ short p_lds[4][16][64];
+
+  // Q and dO fragments in registers (A-operand: m=lo, k=hi*8+j)
+  bf16x8 q_frag[KD], do_frag[KD];
+  {
+    int row = min(q_row0 + lo, S - 1);
+#pragma unroll
+    for (int ks = 0; ks < KD; ++ks) {
+      int col = ks * 32 + hi * 8;
+      bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+      q_frag[ks] = z;
+      do_frag[ks] = z;
+      if (col + 8 <= D) {
+        q_frag[ks] = *reinterpret_cast<const bf16x8*>(
+            qp + (int64_t)row * st.qs + col);
+        do_frag[ks] = *reinterpret_cast<const bf16x8*>(
+            dop + (int64_t)row * st.dos + col);
+      }
+    }
+  }
+  // per-row lse/delta (rows hi*4+r)
+  float lse_r[4], delta_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int qi = q_row0 + hi * 4 + r;
+    lse_r[r] = (qi < S) ? lse[(int64_t)bh * S + qi] : 0.f;
+    delta_r[r] = (qi < S) ? delta[(int64_t)bh * S + qi] : 0.f;
+  }
+
+  f32x4 dq_acc[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) dq_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int kv_limit = causal ? min(Skv, qb * 64 + 64) : Skv;
+  for (int kvb = 0; kvb < kv_limit; kvb += 64) {
+    __syncthreads();
+    {  // stage K (row + transposed) and V
+      constexpr int GPR = Dp / 8;
+      for (int t = threadIdx.x; t < 64 * GPR; t += ATTN_THREADS) {
+        int kvr = t / GPR, dg = (t % GPR) * 8;
+        bf16x8 kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        int src = kvb + kvr;
+        if (src < Skv && dg + 8 <= D) {
+          kv8 = *reinterpret_cast<const bf16x8*>(kp + (int64_t)src * st.ks + dg);
+          vv8 = *reinterpret_cast<const bf16x8*>(vp + (int64_t)src * st.vs + dg);
+        }
+        *reinterpret_cast<bf16x8*>(&k_lds[kvr][dg]) = kv8;
+        *reinterpret_cast<bf16x8*>(&v_lds[kvr][dg]) = vv8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) kt_lds[dg + j][kvr] = kv8[j];
+      }
+    }
+    __syncthreads();
+
+    // S and dP tiles
+    f32x4 s_acc[NT], dp_acc[NT];
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt) {
+      f32x4 sa = {0.f, 0.f, 0.f, 0.f}, da = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KD; ++ks) {
+        bf16x8 bk = *reinterpret_cast<const bf16x8*>(
+            &k_lds[nt * 16 + lo][ks * 32 + hi * 8]);
+        bf16x8 bv = *reinterpret_cast<const bf16x8*>(
+            &v_lds[nt * 16 + lo][ks * 32 + hi * 8]);
+        sa = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], bk, sa, 0,
+                                                     0, 0);
+        da = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[ks], bv, da, 0,
+                                                     0, 0);
+      }
+      s_acc[nt] = sa;
+      dp_acc[nt] = da;
+    }
+
+    // dS = P * (dP - delta) * scale, written to p_lds as A-operand
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt) {
+      int kv_idx = kvb + nt * 16 + lo;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int q_idx = qb * 64 + wave * 16 + hi * 4 + r;
+        bool masked = (kv_idx >= Skv) || (causal && kv_idx > q_idx) ||
+                      (q_idx >= S);
+        float p = masked ? 0.f
+                         : __expf(s_acc[nt][r] * scale - lse_r[r]);
+        float ds = p * (dp_acc[nt][r] - delta_r[r]) * scale;
+        p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(ds);
+      }
+    }
+
+    // dQ += dS @ K   (A = dS from p_lds, B = Kt_lds)
+#pragma unroll
+    for (int ks = 0; ks < 64 / 32; ++ks) {
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          &p_lds[wave][lo][ks * 32 + hi * 8]);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &kt_lds[dt * 16 + lo][ks * 32 + hi * 8]);
+        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
+                                                             dq_acc[dt], 0,
+                                                             0, 0);
+      }
+    }
+  }
+
+  // store dQ (C-layout scatter)
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int q_idx = qb * 64 + wave * 16 + hi * 4 + r;
+    if (q_idx >= S) continue;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      int col = dt * 16 + lo;
+      if (col < D)
+        dqp[(int64_t)q_idx * st.dqs + col] = f2bf(dq_acc[dt][r]);
+    }
+  }
+}
+
+// ------------------------------------------------------------- dk/dv kernel
+// Block: 4 waves, 64 kv rows; loops q blocks of 64 (from the diagonal for
+// causal).
+//   S^T = K Q^T        (A=K regs, B=Q_lds row-major)
+//   P^T = exp(S^T*scale - lse[q])
+//   dV += P^T dO       (A=P^T via p_lds, B=dOt_lds)
+//   dP^T = V dO^T      (A=V regs, B=dO_lds row-major)
+//   dS^T = P^T*(dP^T - delta[q])*scale
+//   dK += dS^T Q       (A=dS^T via p_lds, B=Qt_lds)
+template <int Dp>
+__global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dkv_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, const short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    short* __restrict__ dk, short* __restrict__ dv, int H, int S, int Skv,
+    int D, float scale, int causal, AttnBwdStrides st) {
+  constexpr int KD = Dp / 32;
+  constexpr int NT = 4;  // 64 q per tile
+  constexpr int DT = Dp / 16;
+
+  const int kvb_idx = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int batch = bh / H, head = bh % H;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 15, hi = lane >> 4;
+
+  const short* qp = q + batch * st.qb + head * st.qh;
+  const short* kp = k + batch * st.kb + head * st.kh;
+  const short* vp = v + batch * st.vb + head * st.vh;
+  const short* dop = dout + batch * st.dob + head * st.doh;
+  short* dkp = dk + batch * st.dkb + head * st.dkh;
+  short* dvp = dv + batch * st.dvb + head * st.dvh;
+  const int kv_row0 = kvb_idx * 64 + wave * 16;
+
+  __shared__ short q_lds[64][Dp];
+  __shared__ short qt_lds[Dp][64];
+  __shared__ short do_lds[64][Dp];
+  __shared__ short dot_lds[Dp][64];
+  __shared__ short p_lds[4][16][64];
+  __shared__ float lse_lds[64];
+  __shared__ float delta_lds[64];
+
+  // K and V fragments in registers (A-operand)
+  bf16x8 k_frag[KD], v_frag[KD];
+  {
+    int row = min(kv_row0 + lo, Skv - 1);
+#pragma unroll
+    for (int ks = 0; ks < KD; ++ks) {
+      int col = ks * 32 + hi * 8;
+      bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+      k_frag[ks] = z;
+      v_frag[ks] = z;
+      if (col + 8 <= D) {
+        k_frag[ks] = *reinterpret_cast<const bf16x8*>(
+            kp + (int64_t)row * st.ks + col);
+        v_frag[ks] = *reinterpret_cast<const bf16x8*>(
+            vp + (int64_t)row * st.vs + col);
+      }
+    }
+  }
+
+  f32x4 dk_acc[DT], dv_acc[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) {
+    dk_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dv_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+
+  const int q_start = causal ? (kvb_idx * 64 / 64) * 64 : 0;
+  for (int qb = q_start; qb < S; qb += 64) {
+    __syncthreads();
+    {  // stage Q, dO (row + transposed) and lse/delta
+      constexpr int GPR = Dp / 8;
+      for (int t = threadIdx.x; t < 64 * GPR; t += ATTN_THREADS) {
+        int qr = t / GPR, dg = (t % GPR) * 8;
+        bf16x8 q8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        bf16x8 d8 = {0, 0, 0, 0, 0, 0, 0, 0};
+        int src = qb + qr;
+        if (src < S && dg + 8 <= D) {
+          q8 = *reinterpret_cast<const bf16x8*>(qp + (int64_t)src * st.qs + dg);
+          d8 = *reinterpret_cast<const bf16x8*>(dop + (int64_t)src * st.dos + dg);
+        }
+        *reinterpret_cast<bf16x8*>(&q_lds[qr][dg]) = q8;
+        *reinterpret_cast<bf16x8*>(&do_lds[qr][dg]) = d8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          qt_lds[dg + j][qr] = q8[j];
+          dot_lds[dg + j][qr] = d8[j];
+        }
+      }
+      for (int t = threadIdx.x; t < 64; t += ATTN_THREADS) {
+        int src = qb + t;
+        lse_lds[t] = (src < S) ? lse[(int64_t)bh * S + src] : 0.f;
+        delta_lds[t] = (src < S) ? delta[(int64_t)bh * S + src] : 0.f;
+      }
+    }
+    __syncthreads();
+
+    // S^T and dP^T tiles (rows = kv, cols = q)
+    f32x4 st_acc[NT], dpt_acc[NT];
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt) {
+      f32x4 sa = {0.f, 0.f, 0.f, 0.f}, da = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KD; ++ks) {
+        bf16x8 bq = *reinterpret_cast<const bf16x8*>(
+            &q_lds[nt * 16 + lo][ks * 32 + hi * 8]);
+        bf16x8 bd = *reinterpret_cast<const bf16x8*>(
+            &do_lds[nt * 16 + lo][ks * 32 + hi * 8]);
+        sa = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[ks], bq, sa, 0,
+                                                     0, 0);
+        da = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[ks], bd, da, 0,
+                                                     0, 0);
+      }
+      st_acc[nt] = sa;
+      dpt_acc[nt] = da;
+    }
+
+    // P^T -> p_lds for dV; then dS^T -> p_lds for dK (two passes over the
+    // same per-wave buffer, separated by the MFMA consumption)
+    float pt[NT][4], dst[NT][4];
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt) {
+      int q_idx = qb + nt * 16 + lo;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int kv_idx = kvb_idx * 64 + wave * 16 + hi * 4 + r;
+        bool masked = (q_idx >= S) || (kv_idx >= Skv) ||
+                      (causal && kv_idx > q_idx);
+        float p = masked ? 0.f
+                         : __expf(st_acc[nt][r] * scale - lse_lds[nt * 16 + lo]);
+        pt[nt][r] = p;
+        dst[nt][r] = p * (dpt_acc[nt][r] - delta_lds[nt * 16 + lo]) * scale;
+      }
+    }
+
+    // dV += P^T @ dO (A = P^T, B = dOt)
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(pt[nt][r]);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          &p_lds[wave][lo][ks * 32 + hi * 8]);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &dot_lds[dt * 16 + lo][ks * 32 + hi * 8]);
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
+                                                             dv_acc[dt], 0,
+                                                             0, 0);
+      }
+    }
+
+    // dK += dS^T @ Q (A = dS^T, B = Qt)
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(dst[nt][r]);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          &p_lds[wave][lo][ks * 32 + hi * 8]);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &qt_lds[dt * 16 + lo][ks * 32 + hi * 8]);
+        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
+                                                             dk_acc[dt], 0,
+                                                             0, 0);
+      }
+    }
+  }
+
+  // store dK, dV
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int kv_idx = kvb_idx * 64 + wave * 16 + hi * 4 + r;
+    if (kv_idx >= Skv) continue;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+      int col = dt * 16 + lo;
+      if (col < D) {
+        dkp[(int64_t)kv_idx * st.dks + col] = f2bf(dk_acc[dt][r]);
+        dvp[(int64_t)kv_idx * st.dvs + col] = f2bf(dv_acc[dt][r]);
+      }
+    }
+  }
+}
+
+extern "C" {
+
+hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
+                           const void* o, const void* dout,
+                           const float* lse, float* delta_ws, void* dq,
+                           void* dk, void* dv, int64_t B, int64_t H,
+                           int64_t S, int64_t Skv, int64_t D, float scale,
+                           int causal, const int64_t* strides,
+                           hipStream_t stream) {
+  // strides layout: q(3) k(3) v(3) do(3) dq(3) dk(3) dv(3) o(3)
+  AttnBwdStrides st;
+  const int64_t* p = strides;
+  st.qb = p[0]; st.qh = p[1]; st.qs = p[2];
+  st.kb = p[3]; st.kh = p[4]; st.ks = p[5];
+  st.vb = p[6]; st.vh = p[7]; st.vs = p[8];
+  st.dob = p[9]; st.doh = p[10]; st.dos = p[11];
+  st.dqb = p[12]; st.dqh = p[13]; st.dqs = p[14];
+  st.dkb = p[15]; st.dkh = p[16]; st.dks = p[17];
+  st.dvb = p[18]; st.dvh = p[19]; st.dvs = p[20];
+  int64_t rows = B * H * S;
+  {
+    dim3 block(256);
+    dim3 grid((uint32_t)ceil_div(rows, 4));
+    attn_bwd_preprocess_kernel<<<grid, block, 0, stream>>>(
+        (const short*)dout, (const short*)o, delta_ws, rows, (int)H, (int)S,
+        (int)D, st.dob, st.doh, st.dos, p[21], p[22], p[23]);
+  }
+  dim3 block(ATTN_THREADS);
+  dim3 grid_q((uint32_t)ceil_div(S, 64), (uint32_t)(B * H));
+  dim3 grid_kv((uint32_t)ceil_div(Skv, 64), (uint32_t)(B * H));
+#define LAUNCH_BWD(DP)                                                       \
+  do {                                                                       \
+    attn_bwd_dq_kernel<DP><<<grid_q, block, 0, stream>>>(                    \
+        (const short*)q, (const short*)k, (const short*)v,                   \
+        (const short*)dout, lse, delta_ws, (short*)dq, (int)H, (int)S,       \
+        (int)Skv, (int)D, scale, causal, st);                                \
+    attn_bwd_dkv_kernel<DP><<<grid_kv, block, 0, stream>>>(                  \
+        (const short*)q, (const short*)k, (const short*)v,                   \
+        (const short*)dout, lse, delta_ws, (short*)dk, (short*)dv, (int)H,   \
+        (int)S, (int)Skv, (int)D, scale, causal, st);                        \
+  } while (0)
+  if (D <= 64) {
+    LAUNCH_BWD(64);
+  } else if (D <= 96) {
+    LAUNCH_BWD(96);
+  } else if (D <= 128) {
+    LAUNCH_BWD(128);
+  } else {
+    return hipErrorInvalidValue;
+  }
+#undef LAUNCH_BWD
+  return hipGetLastError();
+}
+
+}  // extern "C"

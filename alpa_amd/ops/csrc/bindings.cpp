@@ -43,6 +43,11 @@ hipError_t launch_attn_fwd(const void*, const void*, const void*, void*,
                            float*, int64_t, int64_t, int64_t, int64_t,
                            int64_t, float, int, const int64_t*, hipStream_t);
 hipError_t launch_mfma_probe(const void*, const void*, float*, hipStream_t);
+hipError_t launch_attn_bwd(const void*, const void*, const void*,
+                           const void*, const void*, const float*, float*,
+                           void*, void*, void*, int64_t, int64_t, int64_t,
+                           int64_t, int64_t, float, int, const int64_t*,
+                           hipStream_t);
 }
 
 namespace {
@@ -241,7 +246,7 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
 // S x S score matrix from materializing beyond one [Bq, Skv] block.
 // (A fully hand-written HIP bwd kernel replaces this on the optimization
 // path.)
-std::vector<at::Tensor> attn_bwd(const at::Tensor& dout_, const at::Tensor& q_,
+std::vector<at::Tensor> attn_bwd_blocked(const at::Tensor& dout_, const at::Tensor& q_,
                                  const at::Tensor& k_, const at::Tensor& v_,
                                  const at::Tensor& o_, const at::Tensor& lse,
                                  bool causal, double scale) {
@@ -279,6 +284,52 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout_, const at::Tensor& q_,
   return {dq, dk.to(k.scalar_type()), dv.to(v.scalar_type())};
 }
 
+// Hand-written flash-attention backward (gfx950 MFMA, deterministic
+// two-kernel split).  All tensors are logical [B,H,S,D] views with
+// arbitrary B/H/S strides (contiguous D) — dq/dk/dv may be strided views
+// into a packed dqkv buffer.
+void attn_bwd_out(const at::Tensor& dout, const at::Tensor& q,
+                  const at::Tensor& k, const at::Tensor& v,
+                  const at::Tensor& o, const at::Tensor& lse, at::Tensor dq,
+                  at::Tensor dk, at::Tensor dv, bool causal, double scale) {
+  check_bf16_strided4(q, "q");
+  check_bf16_strided4(dout, "dout");
+  check_bf16_strided4(dq, "dq");
+  int64_t B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  int64_t Skv = k.size(2);
+  auto delta = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  auto lse_c = lse.contiguous();
+  int64_t strides[24] = {
+      q.stride(0), q.stride(1), q.stride(2),
+      k.stride(0), k.stride(1), k.stride(2),
+      v.stride(0), v.stride(1), v.stride(2),
+      dout.stride(0), dout.stride(1), dout.stride(2),
+      dq.stride(0), dq.stride(1), dq.stride(2),
+      dk.stride(0), dk.stride(1), dk.stride(2),
+      dv.stride(0), dv.stride(1), dv.stride(2),
+      o.stride(0), o.stride(1), o.stride(2)};
+  HIP_OK(launch_attn_bwd(q.const_data_ptr(), k.const_data_ptr(),
+                         v.const_data_ptr(), o.const_data_ptr(),
+                         dout.const_data_ptr(),
+                         (const float*)lse_c.const_data_ptr(),
+                         (float*)delta.mutable_data_ptr(),
+                         dq.mutable_data_ptr(), dk.mutable_data_ptr(),
+                         dv.mutable_data_ptr(), B, H, S, Skv, D,
+                         (float)scale, causal ? 1 : 0, strides,
+                         cur_stream()));
+}
+
+std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
+                                 const at::Tensor& k, const at::Tensor& v,
+                                 const at::Tensor& o, const at::Tensor& lse,
+                                 bool causal, double scale) {
+  auto dq = at::empty_like(q.contiguous());
+  auto dk = at::empty_like(k.contiguous());
+  auto dv = at::empty_like(v.contiguous());
+  attn_bwd_out(dout, q, k, v, o, lse, dq, dk, dv, causal, scale);
+  return {dq, dk, dv};
+}
+
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
   check_bf16_contig(a, "a");
   auto d = at::empty({16, 16}, a.options().dtype(at::kFloat));
@@ -299,6 +350,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE bwd (gfx950)");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)");
   m.def("attn_fwd_out", &attn_fwd_out, "flash attention fwd, strided out");
-  m.def("attn_bwd", &attn_bwd, "attention bwd (blocked, deterministic)");
+  m.def("attn_bwd", &attn_bwd, "flash attention bwd (gfx950 MFMA)");
+  m.def("attn_bwd_out", &attn_bwd_out, "flash attention bwd, strided out");
+  m.def("attn_bwd_blocked", &attn_bwd_blocked,
+        "attention bwd (blocked hipBLASLt reference)");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
 }
