@@ -1,0 +1,84 @@
+"""Data loader, timers/tracer, parallel plan save/load (reference:
+tests/runtime/ + alpa/parallel_plan.py)."""
+import json
+import os
+
+import torch
+
+from alpa_amd.data_loader import DataLoader, shard_batch, synthetic_lm_batches
+from alpa_amd.parallel_plan import (ParallelPlan, PipelinePlan, StagePlan,
+                                    method_to_plan, plan_to_method)
+from alpa_amd.parallel_method import PipeshardParallel, ShardParallel
+from alpa_amd.timer import Timers, Tracer
+
+
+def test_shard_batch():
+    b = {"x": torch.arange(8).view(8, 1), "y": [torch.arange(8)]}
+    s = shard_batch(b, 2, 1)
+    assert s["x"].tolist() == [[4], [5], [6], [7]]
+    assert s["y"][0].tolist() == [4, 5, 6, 7]
+
+
+def test_data_loader_serial():
+    batches = list(synthetic_lm_batches(5, 8, 16, 100))
+    dl = DataLoader(batches, mesh=None)
+    out = list(dl)
+    assert len(out) == 5
+    torch.testing.assert_close(out[0]["ids"], batches[0]["ids"])
+
+
+def test_data_loader_prefetch_order():
+    dl = DataLoader([{"i": torch.tensor([k])} for k in range(7)],
+                    prefetch_size=3)
+    assert [int(b["i"]) for b in dl] == list(range(7))
+
+
+def test_timers():
+    t = Timers()
+    with t("a"):
+        pass
+    t("a").start()
+    t("a").stop()
+    assert len(t("a").costs) == 2
+    assert "a:" in t.log()
+
+
+def test_tracer_chrome_dump(tmp_path):
+    tr = Tracer()
+    tr.begin("step")
+    tr.begin("fwd")
+    tr.end()
+    tr.end()
+    p = tmp_path / "trace.json"
+    tr.dump_chrome_trace(str(p))
+    d = json.loads(p.read_text())
+    names = [e["name"] for e in d["traceEvents"]]
+    assert set(names) == {"step", "fwd"}
+    assert all(e["ph"] == "X" for e in d["traceEvents"])
+
+
+def test_parallel_plan_roundtrip(tmp_path):
+    plan = ParallelPlan(
+        world_size=8, num_micro_batches=4,
+        pipeline_plan=PipelinePlan(num_stages=2, schedule="1f1b",
+                                   layer_ranges=[(0, 16), (16, 32)],
+                                   stage_mesh_shape=(2, 2)),
+        stage_plans=[StagePlan((2, 2), {"b0.qkv": "b0_col1"}, 1.5)])
+    p = tmp_path / "plan.json"
+    plan.save(str(p))
+    loaded = ParallelPlan.load(str(p))
+    assert loaded.pipeline_plan.num_stages == 2
+    assert loaded.pipeline_plan.layer_ranges == [(0, 16), (16, 32)]
+    assert loaded.stage_plans[0].strategy_choices == {"b0.qkv": "b0_col1"}
+
+    m = plan_to_method(loaded)
+    assert isinstance(m, PipeshardParallel)
+    assert m.num_stages == 2 and m.stage_mesh_shape == (2, 2)
+
+
+def test_method_to_plan_shard():
+    m = ShardParallel(num_micro_batches=2, logical_mesh_shape=(4, 2))
+    plan = method_to_plan(m, 8)
+    assert plan.stage_plans[0].logical_mesh_shape == (4, 2)
+    m2 = plan_to_method(plan)
+    assert m2.logical_mesh_shape == (4, 2)
