@@ -136,6 +136,46 @@ class PipeshardParallel(ParallelMethod):
         return full_mesh((1, world_size()))
 
 
+@dataclass
+class CreateStateParallel(ParallelMethod):
+    """Build/initialize state directly with the training method's
+    placement (reference CreateStateParallel, parallel_method.py:336 +
+    create_state_parallel.py:73).  In the SPMD-per-rank runtime,
+    TrainState.create already constructs every shard on its target device
+    with the resolved mesh — this method makes that alignment explicit by
+    borrowing the training method's mesh."""
+    train_method: Optional[ParallelMethod] = None
+
+    def resolve_mesh(self) -> DeviceMesh:
+        assert self.train_method is not None
+        return self.train_method.resolve_mesh()
+
+
+@dataclass
+class FollowParallel(ParallelMethod):
+    """Run another function (e.g. an eval/inference step) with the input
+    placement of an already-compiled training state (reference
+    FollowParallel, parallel_method.py:380 / follow_parallel.py:25)."""
+    train_method: Optional[ParallelMethod] = None
+
+    def resolve_mesh(self) -> DeviceMesh:
+        assert self.train_method is not None
+        return self.train_method.resolve_mesh()
+
+
+def parallelize_inference(fn, state):
+    """FollowParallel execution: run fn(model, batch) under no_grad on the
+    state's mesh/placement (degenerate sharding-propagation-only compile,
+    reference follow_parallel.py:25)."""
+    import torch as _torch
+
+    def run(batch):
+        with _torch.no_grad():
+            return fn(state.model, batch)
+
+    return run
+
+
 def get_3d_parallel_method(num_micro_batches: int, data_parallel: int,
                            operator_parallel: int, pipeline_parallel: int
                            ) -> ParallelMethod:
