@@ -53,9 +53,14 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   short* op = o + batch * st.ob + head * st.oh;
   const int q_row0 = qb * ATTN_BLOCK_Q + wave * 16;  // wave's first q row
 
-  __shared__ short k_lds[ATTN_BLOCK_K][Dp];
-  __shared__ short vt_lds[Dp][ATTN_BLOCK_K];
-  __shared__ short p_lds[4][16][ATTN_BLOCK_K];  // per-wave P tile
+  // +8 bf16 (16 B) row padding: unpadded strides are multiples of 128 B,
+  // putting all 16 fragment-read lanes in the same LDS bank (8-16-way
+  // conflict); the pad keeps 16 B alignment while spreading banks
+  // (guide §6 Guideline 4).
+  constexpr int LP = 8;
+  __shared__ short k_lds[ATTN_BLOCK_K][Dp + LP];
+  __shared__ short vt_lds[Dp][ATTN_BLOCK_K + LP];
+  __shared__ short p_lds[4][16][ATTN_BLOCK_K + LP];  // per-wave P tile
 
   // ---- load Q fragments (held in registers for the whole kv loop) ----
   bf16x8 q_frag[KSTEPS_QK];
@@ -367,10 +372,11 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dq_kernel(
   short* dqp = dq + batch * st.dqb + head * st.dqh;
   const int q_row0 = qb * 64 + wave * 16;
 
-  __shared__ short k_lds[64][Dp];
-  __shared__ short kt_lds[Dp][64];
-  __shared__ short v_lds[64][Dp];
-  __shared__ short p_lds[4][16][64];
+  constexpr int LP = 8;  // bank-conflict row padding (see fwd kernel)
+  __shared__ short k_lds[64][Dp + LP];
+  __shared__ short kt_lds[Dp][64 + LP];
+  __shared__ short v_lds[64][Dp + LP];
+  __shared__ short p_lds[4][16][64 + LP];
 
   // Q and dO fragments in registers (A-operand: m=lo, k=hi*8+j)
   bf16x8 q_frag[KD], do_frag[KD];
@@ -526,11 +532,12 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dkv_kernel(
   short* dvp = dv + batch * st.dvb + head * st.dvh;
   const int kv_row0 = kvb_idx * 64 + wave * 16;
 
-  __shared__ short q_lds[64][Dp];
-  __shared__ short qt_lds[Dp][64];
-  __shared__ short do_lds[64][Dp];
-  __shared__ short dot_lds[Dp][64];
-  __shared__ short p_lds[4][16][64];
+  constexpr int LP = 8;  // bank-conflict row padding (see fwd kernel)
+  __shared__ short q_lds[64][Dp + LP];
+  __shared__ short qt_lds[Dp][64 + LP];
+  __shared__ short do_lds[64][Dp + LP];
+  __shared__ short dot_lds[Dp][64 + LP];
+  __shared__ short p_lds[4][16][64 + LP];
   __shared__ float lse_lds[64];
   __shared__ float delta_lds[64];
 

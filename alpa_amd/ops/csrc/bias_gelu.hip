@@ -104,8 +104,7 @@ hipError_t launch_bias_gelu_bwd(const void* dy, const void* x,
   dim3 grid((uint32_t)ceil_div(F, 256), P);
   bias_grad_partial_kernel<<<grid, dim3(256), 0, stream>>>(
       (const short*)dx, db_part, N, (int)F);
-  bias_colsum_kernel<<<dim3((uint32_t)ceil_div(F, 256)), dim3(256), 0,
-                       stream>>>(db_part, db, P, (int)F);
+  // final P-row reduction handled by the caller (at::sum)
   return hipGetLastError();
 }
 

@@ -115,6 +115,8 @@ std::vector<at::Tensor> layer_norm_bwd(const at::Tensor& dy,
       (const float*)rstd.const_data_ptr(), (float*)dw_part.mutable_data_ptr(),
       (float*)db_part.mutable_data_ptr(), (float*)dw.mutable_data_ptr(),
       (float*)db.mutable_data_ptr(), N, H, kStripes, cur_stream()));
+  at::sum_out(dw, dw_part, {0});
+  at::sum_out(db, db_part, {0});
   return {dx, dw, db};
 }
 
@@ -147,6 +149,7 @@ std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& dy,
                               (float*)db_part.mutable_data_ptr(),
                               (float*)db.mutable_data_ptr(), N, F, kStripes,
                               cur_stream()));
+  at::sum_out(db, db_part, {0});
   return {dx, db};
 }
 

@@ -172,15 +172,14 @@ hipError_t launch_layer_norm_bwd_dx(const void* dy, const void* x,
 hipError_t launch_layer_norm_bwd_dwdb(const void* dy, const void* x,
                                       const float* mean, const float* rstd,
                                       float* dw_part, float* db_part,
-                                      float* dw, float* db, int64_t N,
+                                      float* /*dw*/, float* /*db*/, int64_t N,
                                       int64_t H, int P, hipStream_t stream) {
   dim3 grid((uint32_t)ceil_div(H, LNB_COLS), P);
   layer_norm_bwd_dwdb_partial<<<grid, dim3(LNB_COLS), 0, stream>>>(
       (const short*)dy, (const short*)x, mean, rstd, dw_part, db_part,
       (int)N, (int)H);
-  dim3 grid2((uint32_t)ceil_div(H, 256));
-  column_sum_kernel<<<grid2, dim3(256), 0, stream>>>(dw_part, dw, P, (int)H);
-  column_sum_kernel<<<grid2, dim3(256), 0, stream>>>(db_part, db, P, (int)H);
+  // final P-row reduction handled by the caller (at::sum — parallel
+  // reduce kernel beats the serial per-column loop)
   return hipGetLastError();
 }
 
