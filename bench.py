@@ -53,8 +53,21 @@ def parse_args():
     return p.parse_args()
 
 
+def _enable_tunableop():
+    """Load the shipped hipBLASLt TunableOp selections for gfx950 (tuned
+    once on MI355X; ~2% over the default heuristics)."""
+    base = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                        "alpa_amd", "ops", "tuned", "tunableop_gfx950_.csv")
+    if os.path.exists(base.replace("_.csv", "_0.csv")):
+        os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", base)
+
+
 def main():
     args = parse_args()
+    if torch.cuda.is_available():
+        _enable_tunableop()
     aa.init()
     on_gpu = torch.cuda.is_available()
     n = aa.world_size()
