@@ -141,3 +141,15 @@ def test_pipeline2_matches_serial(nmb, schedule):
     for r in results:
         for a, b in zip(r, serial):
             assert abs(a - b) < 2e-4, (r, serial)
+
+
+@pytest.mark.parametrize("nmb", [4, 2])
+def test_pipeline4_matches_serial(nmb):
+    """4 stages over gloo ws=4 — deeper warmup/cooldown interleavings than
+    P=2 (the shape the 8-GPU runs will hit)."""
+    serial = run_serial(nmb)
+    results = run_distributed(_pp_worker, world_size=4,
+                              args=(nmb, "1f1b"), timeout=300)
+    for r in results:
+        for a, b in zip(r, serial):
+            assert abs(a - b) < 2e-4, (r, serial)
