@@ -88,7 +88,9 @@ def main():
         batch_per_gpu = 4
 
     if args.parallel == "auto" and args.dp == 0 and n > 1:
-        # ILP auto-sharding picks (dp, tp) from the op graph + xGMI costs
+        # ILP auto-sharding picks (dp, tp) from the op graph + xGMI costs.
+        # Solve on rank 0 only and broadcast: every rank must construct the
+        # same mesh (group creation is collective).
         state_bytes = 12.0 * cfg.num_params()
         hint = {
             "family": "gpt", "hidden": cfg.hidden_size,
@@ -97,10 +99,18 @@ def main():
             "memory_budget": (0.9 * 288e9
                               if state_bytes > 0.5 * 288e9 else None),
         }
+        from alpa_amd.shard_parallel.mesh_search import choose_mesh_shape
+        probe = aa.ShardParallel(num_micro_batches=args.nmb,
+                                 model_hint=hint)
+        shape = choose_mesh_shape(probe, n) if aa.rank() == 0 else None
+        if aa.world_size() > 1:
+            import torch.distributed as dist
+            lst = [shape]
+            dist.broadcast_object_list(lst, src=0)
+            shape = tuple(lst[0])
+        dp, tp = shape
         method = aa.ShardParallel(num_micro_batches=args.nmb,
-                                  model_hint=hint)
-        mesh_probe = method.resolve_mesh()
-        dp, tp = mesh_probe.shape
+                                  logical_mesh_shape=(dp, tp))
     else:
         dp = args.dp if args.dp > 0 else n // args.tp
         tp = args.tp
