@@ -17,7 +17,8 @@
 
 #define ATTN_BLOCK_Q 64
 #define ATTN_BLOCK_K 64
-#define ATTN_THREADS 256
+#define ATTN_THREADS 1024
+#define ATTN_BWD_THREADS 512
 
 // Strides are in elements; the innermost (D) dim must be contiguous.
 // Strided addressing lets the packed qkv layout [B, S, heads, 3*D] feed the
@@ -29,211 +30,300 @@ struct AttnStrides {
   int64_t ob, oh, os;
 };
 
-template <int Dp>
+// ABL: ablation level for perf diagnosis (0 = full kernel; higher skips
+// later phases; asm keep-alives prevent dead-code elimination of earlier
+// phases — guide methodology rule 17)
+template <int Dp, int ABL = 0>
 __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int H, int S, int Skv, int D, float scale,
     int causal, AttnStrides st) {
-  constexpr int KSTEPS_QK = Dp / 32;   // k-steps over head dim
+  // 8 waves x 16 q-rows = 128 q rows per block; K/V tiles of 64 kv are
+  // double-buffered in LDS with register-prefetched staging (loads for
+  // tile t+1 issue before computing tile t and land after it — the HBM
+  // latency hides under the MFMA/softmax work), one barrier per tile.
+  constexpr int KSTEPS_QK = Dp / 32;
   constexpr int NTILES = ATTN_BLOCK_K / 16;  // 4
   constexpr int DTILES = Dp / 16;
+  constexpr int LP = 8;                      // LDS bank padding (16 B)
+  constexpr int NW = ATTN_THREADS / 64;      // 8 waves
+  constexpr int GPR = Dp / 8;                // bf16x8 groups per kv row
+  constexpr int TOTAL_G = ATTN_BLOCK_K * GPR;
+  constexpr int G_PER_T = (TOTAL_G + ATTN_THREADS - 1) / ATTN_THREADS;
 
-  const int qb = blockIdx.x;           // q block index
-  const int bh = blockIdx.y;           // fused batch*head
+  const int qb = blockIdx.x;
+  const int bh = blockIdx.y;
   const int batch = bh / H, head = bh % H;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int lo = lane & 15;            // fragment row/col low index
-  const int hi = lane >> 4;            // fragment quartet index
+  const int lo = lane & 15;
+  const int hi = lane >> 4;
 
   const short* qp = q + batch * st.qb + head * st.qh;
   const short* kp = k + batch * st.kb + head * st.kh;
   const short* vp = v + batch * st.vb + head * st.vh;
   short* op = o + batch * st.ob + head * st.oh;
-  const int q_row0 = qb * ATTN_BLOCK_Q + wave * 16;  // wave's first q row
+  const int q_row0 = qb * (16 * NW) + wave * 16;
 
-  // +8 bf16 (16 B) row padding: unpadded strides are multiples of 128 B,
-  // putting all 16 fragment-read lanes in the same LDS bank (8-16-way
-  // conflict); the pad keeps 16 B alignment while spreading banks
-  // (guide §6 Guideline 4).
-  constexpr int LP = 8;
-  __shared__ short k_lds[ATTN_BLOCK_K][Dp + LP];
-  __shared__ short vt_lds[Dp][ATTN_BLOCK_K + LP];
-  __shared__ short p_lds[4][16][ATTN_BLOCK_K + LP];  // per-wave P tile
+  __shared__ short k_lds[2][ATTN_BLOCK_K][Dp + LP];
+  __shared__ short vt_lds[2][Dp][ATTN_BLOCK_K + LP];
+  __shared__ short p_lds[NW][16][ATTN_BLOCK_K + LP];
 
-  // ---- load Q fragments (held in registers for the whole kv loop) ----
+  // ---- Q fragments in registers ----
   bf16x8 q_frag[KSTEPS_QK];
   {
-    int m = q_row0 + lo;
-    int row = min(m, S - 1);
+    int row = min(q_row0 + lo, S - 1);
 #pragma unroll
     for (int ks = 0; ks < KSTEPS_QK; ++ks) {
       int col = ks * 32 + hi * 8;
-      if (col + 8 <= D) {
+      bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+      q_frag[ks] = z;
+      if (col + 8 <= D)
         q_frag[ks] =
             *reinterpret_cast<const bf16x8*>(qp + (int64_t)row * st.qs + col);
-      } else {
-        bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
-        q_frag[ks] = z;
-      }
     }
   }
 
-  // ---- online softmax state (per owned row r = hi*4 + reg... here the
-  // wave's 16 rows map: reg r of C holds row hi*4+r; every lane tracks the
-  // 4 rows of its quartet) ----
-  float m_state[4], l_state[4];
-  f32x4 o_acc[DTILES];
-#pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    m_state[r] = -INFINITY;
-    l_state[r] = 0.f;
-  }
+  // Swapped-operand form: the wave computes S^T = K Q^T, so each lane
+  // holds 16 S values of ONE q column (q = q_row0 + lo) — the softmax
+  // row-reduction is 16 in-register ops + 2 shfl steps across the 4
+  // hi-groups (vs 8 serial shfl chains per row in the naive layout).
+  // PV becomes O^T = V^T P^T with both operands read contiguously.
+  float m_state = -INFINITY, l_state = 0.f;
+  f32x4 o_acc[DTILES];  // O^T C-layout: d = hi*4+r (+16*dt), q = lo
 #pragma unroll
   for (int dt = 0; dt < DTILES; ++dt) o_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int kv_limit =
-      causal ? min(Skv, qb * ATTN_BLOCK_Q + ATTN_BLOCK_Q) : Skv;
+      causal ? min(Skv, qb * (16 * NW) + 16 * NW) : Skv;
+  const int n_tiles = (kv_limit + ATTN_BLOCK_K - 1) / ATTN_BLOCK_K;
+  const int my_q = q_row0 + lo;  // this lane's q row
 
-  for (int kvb = 0; kvb < kv_limit; kvb += ATTN_BLOCK_K) {
-    // ---- stage K tile and transposed V tile (all 256 threads) ----
-    __syncthreads();
-    {
-      constexpr int GROUPS_PER_ROW = Dp / 8;
-      constexpr int TOTAL = ATTN_BLOCK_K * GROUPS_PER_ROW;
-      for (int t = threadIdx.x; t < TOTAL; t += ATTN_THREADS) {
-        int kvr = t / GROUPS_PER_ROW;
-        int dg = (t % GROUPS_PER_ROW) * 8;
-        bf16x8 kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        int src = kvb + kvr;
-        if (src < Skv && dg + 8 <= D) {
-          kv8 = *reinterpret_cast<const bf16x8*>(kp + (int64_t)src * st.ks + dg);
-          vv8 = *reinterpret_cast<const bf16x8*>(vp + (int64_t)src * st.vs + dg);
-        }
-        *reinterpret_cast<bf16x8*>(&k_lds[kvr][dg]) = kv8;
+  // staging: load tile -> regs (two kv rows per thread so the V transpose
+  // writes pair as b32)
+  bf16x8 kreg[2 * G_PER_T], vreg[2 * G_PER_T];
+  auto issue_loads = [&](int tile) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) vt_lds[dg + j][kvr] = vv8[j];
+    for (int it = 0; it < G_PER_T; ++it) {
+      int t = threadIdx.x + it * ATTN_THREADS;
+      bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+      kreg[2 * it] = z; kreg[2 * it + 1] = z;
+      vreg[2 * it] = z; vreg[2 * it + 1] = z;
+      if (t < TOTAL_G / 2) {
+        int kvr = (t / GPR) * 2, dg = (t % GPR) * 8;
+#pragma unroll
+        for (int u = 0; u < 2; ++u) {
+          int src = tile * ATTN_BLOCK_K + kvr + u;
+          if (src < Skv && dg + 8 <= D) {
+            kreg[2 * it + u] = *reinterpret_cast<const bf16x8*>(
+                kp + (int64_t)src * st.ks + dg);
+            vreg[2 * it + u] = *reinterpret_cast<const bf16x8*>(
+                vp + (int64_t)src * st.vs + dg);
+          }
+        }
       }
     }
-    __syncthreads();
+  };
+  auto write_tile = [&](int buf) {
+#pragma unroll
+    for (int it = 0; it < G_PER_T; ++it) {
+      int t = threadIdx.x + it * ATTN_THREADS;
+      if (t < TOTAL_G / 2) {
+        int kvr = (t / GPR) * 2, dg = (t % GPR) * 8;
+        *reinterpret_cast<bf16x8*>(&k_lds[buf][kvr][dg]) = kreg[2 * it];
+        *reinterpret_cast<bf16x8*>(&k_lds[buf][kvr + 1][dg]) =
+            kreg[2 * it + 1];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          short2 pr;
+          pr.x = vreg[2 * it][j];
+          pr.y = vreg[2 * it + 1][j];
+          *reinterpret_cast<short2*>(&vt_lds[buf][dg + j][kvr]) = pr;
+        }
+      }
+    }
+  };
 
-    // ---- S = Q K^T over this tile (4 ntiles of 16 kv) ----
+  if (n_tiles > 0) {
+    issue_loads(0);
+    write_tile(0);
+    __syncthreads();
+  }
+
+  for (int ti = 0; ti < n_tiles; ++ti) {
+    const int kvb = ti * ATTN_BLOCK_K;
+    const int cur = ti & 1;
+    if (ti + 1 < n_tiles) issue_loads(ti + 1);  // lands under the compute
+
+    if (ABL >= 4) {  // staging-only
+      float keep = bf2f(k_lds[cur][lane][0]) + bf2f(vt_lds[cur][lane][0]);
+      asm volatile("" ::"v"(keep));
+      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      continue;
+    }
+
+    // ---- S^T = K Q^T (A = K tile from LDS, B = Q fragments) ----
     f32x4 s_acc[NTILES];
 #pragma unroll
     for (int nt = 0; nt < NTILES; ++nt) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < KSTEPS_QK; ++ks) {
-        bf16x8 b =
-            *reinterpret_cast<const bf16x8*>(&k_lds[nt * 16 + lo][ks * 32 + hi * 8]);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], b, acc, 0,
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &k_lds[cur][nt * 16 + lo][ks * 32 + hi * 8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, q_frag[ks], acc, 0,
                                                       0, 0);
       }
       s_acc[nt] = acc;
     }
 
-    // ---- mask + scale; rowwise max ----
-    float rowmax[4];
+    if (ABL >= 3) {  // QK^T only
 #pragma unroll
-    for (int r = 0; r < 4; ++r) rowmax[r] = -INFINITY;
+      for (int nt = 0; nt < NTILES; ++nt)
+        asm volatile("" ::"v"(s_acc[nt][0]), "v"(s_acc[nt][3]));
+      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      continue;
+    }
+    if (ABL == 2) {  // no softmax math: raw S write + one read
+#pragma unroll
+      for (int nt = 0; nt < NTILES; ++nt)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          p_lds[wave][lo][nt * 16 + hi * 4 + r] = f2bf(s_acc[nt][r]);
+      bf16x8 a0 = *reinterpret_cast<const bf16x8*>(&p_lds[wave][lo][hi * 8]);
+      asm volatile("" ::"v"(a0));
+      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      continue;
+    }
+
+    // ---- mask + scale + lane-local max over this lane's 16 kv ----
+    float tile_max = -INFINITY;
 #pragma unroll
     for (int nt = 0; nt < NTILES; ++nt) {
-      int kv_idx = kvb + nt * 16 + lo;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int q_idx = qb * ATTN_BLOCK_Q + wave * 16 + hi * 4 + r;
+        int kv_idx = kvb + nt * 16 + hi * 4 + r;
         float sv = s_acc[nt][r] * scale;
-        bool masked = (kv_idx >= Skv) || (causal && kv_idx > q_idx) ||
-                      (q_idx >= S);
+        bool masked = (kv_idx >= Skv) || (causal && kv_idx > my_q) ||
+                      (my_q >= S);
         sv = masked ? -INFINITY : sv;
         s_acc[nt][r] = sv;
-        rowmax[r] = fmaxf(rowmax[r], sv);
+        tile_max = fmaxf(tile_max, sv);
       }
     }
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        rowmax[r] = fmaxf(rowmax[r], __shfl_xor(rowmax[r], off, 16));
-    }
+    // cross-hi reduce (lanes lo, lo+16, lo+32, lo+48 share q column)
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 16, 64));
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
 
-    // ---- online rescale; P = exp(S - m_new); row sums ----
-    float rowsum[4];
+    // ---- online rescale; P^T = exp(S^T - m); sum ----
+    float m_new = fmaxf(m_state, tile_max);
+    float alpha = (m_state == -INFINITY) ? 0.f : __expf(m_state - m_new);
+    m_state = m_new;
+    l_state *= alpha;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float m_new = fmaxf(m_state[r], rowmax[r]);
-      float alpha = (m_state[r] == -INFINITY) ? 0.f : __expf(m_state[r] - m_new);
-      m_state[r] = m_new;
-      l_state[r] *= alpha;
-#pragma unroll
-      for (int dt = 0; dt < DTILES; ++dt) o_acc[dt][r] *= alpha;
-      rowsum[r] = 0.f;
+    for (int dt = 0; dt < DTILES; ++dt) {
+      o_acc[dt][0] *= alpha; o_acc[dt][1] *= alpha;
+      o_acc[dt][2] *= alpha; o_acc[dt][3] *= alpha;
     }
+    float part = 0.f;
 #pragma unroll
     for (int nt = 0; nt < NTILES; ++nt) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float p = (s_acc[nt][r] == -INFINITY)
-                      ? 0.f
-                      : __expf(s_acc[nt][r] - m_state[r]);
-        s_acc[nt][r] = p;
-        rowsum[r] += p;
-        // write P to the wave's LDS tile for re-fragmentation
-        p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(p);
+        float pval = (s_acc[nt][r] == -INFINITY)
+                         ? 0.f
+                         : __expf(s_acc[nt][r] - m_state);
+        part += pval;
+        // P^T transposed store: p_lds[q][kv] so the PV B-fragment reads
+        // are contiguous
+        p_lds[wave][lo][nt * 16 + hi * 4 + r] = f2bf(pval);
       }
     }
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        rowsum[r] += __shfl_xor(rowsum[r], off, 16);
-      l_state[r] += rowsum[r];
+    part += __shfl_xor(part, 16, 64);
+    part += __shfl_xor(part, 32, 64);
+    l_state += part;
+
+    if (ABL >= 1) {
+      asm volatile("" ::"v"(l_state), "v"(m_state));
+      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      continue;
     }
 
-    // ---- O += P V  (A = P from LDS, B = Vt from LDS) ----
-    // (p_lds write->read is wave-local; compiler inserts the lgkmcnt wait)
+    // ---- O^T += V^T P^T  (A = V^T from vt_lds, B = P^T from p_lds) ----
 #pragma unroll
     for (int ks = 0; ks < ATTN_BLOCK_K / 32; ++ks) {
-      bf16x8 a =
-          *reinterpret_cast<const bf16x8*>(&p_lds[wave][lo][ks * 32 + hi * 8]);
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &p_lds[wave][lo][ks * 32 + hi * 8]);
 #pragma unroll
       for (int dt = 0; dt < DTILES; ++dt) {
-        bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            &vt_lds[dt * 16 + lo][ks * 32 + hi * 8]);
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &vt_lds[cur][dt * 16 + lo][ks * 32 + hi * 8]);
         o_acc[dt] =
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, o_acc[dt], 0, 0, 0);
       }
     }
+
+    if (ti + 1 < n_tiles) {
+      write_tile(1 - cur);  // waits on the prefetched loads here
+      __syncthreads();
+    }
   }
 
-  // ---- epilogue: normalize, store O and lse ----
+  // ---- epilogue: lane holds O^T[d = 16*dt + hi*4 + r][q = lo] ----
+  {
+    int q_idx = q_row0 + lo;
+    if (q_idx < S) {
+      float inv_l = (l_state > 0.f) ? 1.0f / l_state : 0.f;
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    int q_idx = qb * ATTN_BLOCK_Q + wave * 16 + hi * 4 + r;
-    if (q_idx >= S) continue;
-    float inv_l = (l_state[r] > 0.f) ? 1.0f / l_state[r] : 0.f;
+      for (int dt = 0; dt < DTILES; ++dt) {
 #pragma unroll
-    for (int dt = 0; dt < DTILES; ++dt) {
-      int col = dt * 16 + lo;
-      if (col < D)
-        op[(int64_t)q_idx * st.os + col] = f2bf(o_acc[dt][r] * inv_l);
+        for (int r = 0; r < 4; ++r) {
+          int col = dt * 16 + hi * 4 + r;
+          if (col < D)
+            op[(int64_t)q_idx * st.os + col] = f2bf(o_acc[dt][r] * inv_l);
+        }
+      }
+      if (hi == 0 && lse_out != nullptr)
+        lse_out[(int64_t)bh * S + q_idx] =
+            (l_state > 0.f) ? m_state + __logf(l_state) : -INFINITY;
     }
-    if (lo == 0 && lse_out != nullptr)
-      lse_out[(int64_t)bh * S + q_idx] =
-          (l_state[r] > 0.f) ? m_state[r] + __logf(l_state[r]) : -INFINITY;
   }
 }
 
 extern "C" {
+
+hipError_t launch_attn_fwd_ablate(const void* q, const void* k,
+                                  const void* v, void* o, float* lse,
+                                  int64_t B, int64_t H, int64_t S,
+                                  int64_t Skv, int64_t D, float scale,
+                                  int causal, const int64_t* strides,
+                                  int abl, hipStream_t stream) {
+  dim3 grid((uint32_t)ceil_div(S, 16 * (ATTN_THREADS / 64)), (uint32_t)(B * H));
+  dim3 block(ATTN_THREADS);
+  AttnStrides st;
+  st.qb = strides[0]; st.qh = strides[1]; st.qs = strides[2];
+  st.kb = strides[3]; st.kh = strides[4]; st.ks = strides[5];
+  st.vb = strides[6]; st.vh = strides[7]; st.vs = strides[8];
+  st.ob = strides[9]; st.oh = strides[10]; st.os = strides[11];
+  if (D > 96) return hipErrorInvalidValue;
+#define ABL_CASE(A)                                                         \
+  if (abl == A) {                                                           \
+    attn_fwd_kernel<96, A><<<grid, block, 0, stream>>>(                     \
+        (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,  \
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);               \
+  }
+  ABL_CASE(0) ABL_CASE(1) ABL_CASE(2) ABL_CASE(3) ABL_CASE(4)
+#undef ABL_CASE
+  return hipGetLastError();
+}
 
 hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
                            void* o, float* lse, int64_t B, int64_t H,
                            int64_t S, int64_t Skv, int64_t D, float scale,
                            int causal, const int64_t* strides,
                            hipStream_t stream) {
-  dim3 grid((uint32_t)ceil_div(S, ATTN_BLOCK_Q), (uint32_t)(B * H));
+  dim3 grid((uint32_t)ceil_div(S, 16 * (ATTN_THREADS / 64)), (uint32_t)(B * H));
   dim3 block(ATTN_THREADS);
   AttnStrides st;
   st.qb = strides[0];
@@ -348,15 +438,22 @@ __global__ void attn_bwd_preprocess_kernel(const short* __restrict__ dout,
 //   dS = P*(dP-delta)*scale
 //   dQ += dS K         (A=dS via p_lds, B=Kt_lds transposed)
 template <int Dp>
-__global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dq, int H, int S, int Skv, int D, float scale,
     int causal, AttnBwdStrides st) {
-  constexpr int KD = Dp / 32;          // k-steps over head dim
-  constexpr int NT = 4;                // 64 kv per tile
+  // 8 waves x 16 q rows = 128 q rows per block; K/V/K^T staged per kv tile
+  // with register-prefetched loads (issue before compute, write after the
+  // barrier) and paired-row transposes.
+  constexpr int KD = Dp / 32;
+  constexpr int NT = 4;                 // 64 kv per tile
   constexpr int DT = Dp / 16;
+  constexpr int LP = 8;
+  constexpr int NW = ATTN_BWD_THREADS / 64;
+  constexpr int GPR = Dp / 8;
+  constexpr int PAIRS = 32 * GPR;       // (64 rows / 2) * groups
 
   const int qb = blockIdx.x;
   const int bh = blockIdx.y;
@@ -370,15 +467,13 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dq_kernel(
   const short* vp = v + batch * st.vb + head * st.vh;
   const short* dop = dout + batch * st.dob + head * st.doh;
   short* dqp = dq + batch * st.dqb + head * st.dqh;
-  const int q_row0 = qb * 64 + wave * 16;
+  const int q_row0 = qb * (16 * NW) + wave * 16;
 
-  constexpr int LP = 8;  // bank-conflict row padding (see fwd kernel)
   __shared__ short k_lds[64][Dp + LP];
   __shared__ short kt_lds[Dp][64 + LP];
   __shared__ short v_lds[64][Dp + LP];
-  __shared__ short p_lds[4][16][64 + LP];
+  __shared__ short p_lds[NW][16][64 + LP];
 
-  // Q and dO fragments in registers (A-operand: m=lo, k=hi*8+j)
   bf16x8 q_frag[KD], do_frag[KD];
   {
     int row = min(q_row0 + lo, S - 1);
@@ -396,7 +491,6 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dq_kernel(
       }
     }
   }
-  // per-row lse/delta (rows hi*4+r)
   float lse_r[4], delta_r[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
@@ -409,29 +503,57 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dq_kernel(
 #pragma unroll
   for (int dt = 0; dt < DT; ++dt) dq_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int kv_limit = causal ? min(Skv, qb * 64 + 64) : Skv;
-  for (int kvb = 0; kvb < kv_limit; kvb += 64) {
-    __syncthreads();
-    {  // stage K (row + transposed) and V
-      constexpr int GPR = Dp / 8;
-      for (int t = threadIdx.x; t < 64 * GPR; t += ATTN_THREADS) {
-        int kvr = t / GPR, dg = (t % GPR) * 8;
-        bf16x8 kv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        bf16x8 vv8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        int src = kvb + kvr;
-        if (src < Skv && dg + 8 <= D) {
-          kv8 = *reinterpret_cast<const bf16x8*>(kp + (int64_t)src * st.ks + dg);
-          vv8 = *reinterpret_cast<const bf16x8*>(vp + (int64_t)src * st.vs + dg);
-        }
-        *reinterpret_cast<bf16x8*>(&k_lds[kvr][dg]) = kv8;
-        *reinterpret_cast<bf16x8*>(&v_lds[kvr][dg]) = vv8;
+  const int kv_limit = causal ? min(Skv, qb * (16 * NW) + 16 * NW) : Skv;
+  const int n_tiles = (kv_limit + 63) / 64;
+
+  bf16x8 kreg[2], vreg[2];
+  auto issue_loads = [&](int tile) {
+    int t = threadIdx.x;
+    bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+    kreg[0] = kreg[1] = vreg[0] = vreg[1] = z;
+    if (t < PAIRS) {
+      int kvr = (t / GPR) * 2, dg = (t % GPR) * 8;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) kt_lds[dg + j][kvr] = kv8[j];
+      for (int u = 0; u < 2; ++u) {
+        int src = tile * 64 + kvr + u;
+        if (src < Skv && dg + 8 <= D) {
+          kreg[u] = *reinterpret_cast<const bf16x8*>(
+              kp + (int64_t)src * st.ks + dg);
+          vreg[u] = *reinterpret_cast<const bf16x8*>(
+              vp + (int64_t)src * st.vs + dg);
+        }
       }
     }
-    __syncthreads();
+  };
+  auto write_tile = [&]() {
+    int t = threadIdx.x;
+    if (t < PAIRS) {
+      int kvr = (t / GPR) * 2, dg = (t % GPR) * 8;
+      *reinterpret_cast<bf16x8*>(&k_lds[kvr][dg]) = kreg[0];
+      *reinterpret_cast<bf16x8*>(&k_lds[kvr + 1][dg]) = kreg[1];
+      *reinterpret_cast<bf16x8*>(&v_lds[kvr][dg]) = vreg[0];
+      *reinterpret_cast<bf16x8*>(&v_lds[kvr + 1][dg]) = vreg[1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        short2 pr;
+        pr.x = kreg[0][j];
+        pr.y = kreg[1][j];
+        *reinterpret_cast<short2*>(&kt_lds[dg + j][kvr]) = pr;
+      }
+    }
+  };
 
-    // S and dP tiles
+  if (n_tiles > 0) {
+    issue_loads(0);
+    write_tile();
+    __syncthreads();
+  }
+
+  for (int ti = 0; ti < n_tiles; ++ti) {
+    const int kvb = ti * 64;
+    if (ti + 1 < n_tiles) issue_loads(ti + 1);
+
+    // S and dP tiles (C: row = q (hi*4+r), col = kv (nt*16+lo))
     f32x4 s_acc[NT], dp_acc[NT];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
@@ -451,42 +573,45 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dq_kernel(
       dp_acc[nt] = da;
     }
 
-    // dS = P * (dP - delta) * scale, written to p_lds as A-operand
+    // dS = P * (dP - delta) * scale -> p_lds as the dS A-operand
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
       int kv_idx = kvb + nt * 16 + lo;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int q_idx = qb * 64 + wave * 16 + hi * 4 + r;
+        int q_idx = q_row0 + hi * 4 + r;
         bool masked = (kv_idx >= Skv) || (causal && kv_idx > q_idx) ||
                       (q_idx >= S);
-        float p = masked ? 0.f
-                         : __expf(s_acc[nt][r] * scale - lse_r[r]);
-        float ds = p * (dp_acc[nt][r] - delta_r[r]) * scale;
+        float pv = masked ? 0.f : __expf(s_acc[nt][r] * scale - lse_r[r]);
+        float ds = pv * (dp_acc[nt][r] - delta_r[r]) * scale;
         p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(ds);
       }
     }
 
-    // dQ += dS @ K   (A = dS from p_lds, B = Kt_lds)
+    // dQ += dS @ K  (A = dS from p_lds, B = Kt)
 #pragma unroll
-    for (int ks = 0; ks < 64 / 32; ++ks) {
+    for (int ks = 0; ks < 2; ++ks) {
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
           &p_lds[wave][lo][ks * 32 + hi * 8]);
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
         bf16x8 b = *reinterpret_cast<const bf16x8*>(
             &kt_lds[dt * 16 + lo][ks * 32 + hi * 8]);
-        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
-                                                             dq_acc[dt], 0,
-                                                             0, 0);
+        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, b, dq_acc[dt], 0, 0, 0);
       }
+    }
+
+    __syncthreads();
+    if (ti + 1 < n_tiles) {
+      write_tile();
+      __syncthreads();
     }
   }
 
-  // store dQ (C-layout scatter)
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    int q_idx = qb * 64 + wave * 16 + hi * 4 + r;
+    int q_idx = q_row0 + hi * 4 + r;
     if (q_idx >= S) continue;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
@@ -507,15 +632,23 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dq_kernel(
 //   dS^T = P^T*(dP^T - delta[q])*scale
 //   dK += dS^T Q       (A=dS^T via p_lds, B=Qt_lds)
 template <int Dp>
-__global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dk, short* __restrict__ dv, int H, int S, int Skv,
     int D, float scale, int causal, AttnBwdStrides st) {
+  // 8 waves x 16 kv rows = 128 kv rows per block; Q/dO tiles (row-major +
+  // transposed) staged per q tile with register-prefetched loads and
+  // paired-row transposes — the staging cost is the dominant term here and
+  // is amortized over 2x the kv rows vs the 4-wave version.
   constexpr int KD = Dp / 32;
-  constexpr int NT = 4;  // 64 q per tile
+  constexpr int NT = 4;   // 64 q per tile
   constexpr int DT = Dp / 16;
+  constexpr int LP = 8;
+  constexpr int NW = ATTN_BWD_THREADS / 64;
+  constexpr int GPR = Dp / 8;
+  constexpr int PAIRS = 32 * GPR;
 
   const int kvb_idx = blockIdx.x;
   const int bh = blockIdx.y;
@@ -530,18 +663,16 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dkv_kernel(
   const short* dop = dout + batch * st.dob + head * st.doh;
   short* dkp = dk + batch * st.dkb + head * st.dkh;
   short* dvp = dv + batch * st.dvb + head * st.dvh;
-  const int kv_row0 = kvb_idx * 64 + wave * 16;
+  const int kv_row0 = kvb_idx * (16 * NW) + wave * 16;
 
-  constexpr int LP = 8;  // bank-conflict row padding (see fwd kernel)
   __shared__ short q_lds[64][Dp + LP];
   __shared__ short qt_lds[Dp][64 + LP];
   __shared__ short do_lds[64][Dp + LP];
   __shared__ short dot_lds[Dp][64 + LP];
-  __shared__ short p_lds[4][16][64 + LP];
+  __shared__ short p_lds[NW][16][64 + LP];
   __shared__ float lse_lds[64];
   __shared__ float delta_lds[64];
 
-  // K and V fragments in registers (A-operand)
   bf16x8 k_frag[KD], v_frag[KD];
   {
     int row = min(kv_row0 + lo, Skv - 1);
@@ -567,37 +698,74 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dkv_kernel(
     dv_acc[dt] = f32x4{0.f, 0.f, 0.f, 0.f};
   }
 
-  const int q_start = causal ? (kvb_idx * 64 / 64) * 64 : 0;
-  for (int qb = q_start; qb < S; qb += 64) {
-    __syncthreads();
-    {  // stage Q, dO (row + transposed) and lse/delta
-      constexpr int GPR = Dp / 8;
-      for (int t = threadIdx.x; t < 64 * GPR; t += ATTN_THREADS) {
-        int qr = t / GPR, dg = (t % GPR) * 8;
-        bf16x8 q8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        bf16x8 d8 = {0, 0, 0, 0, 0, 0, 0, 0};
-        int src = qb + qr;
-        if (src < S && dg + 8 <= D) {
-          q8 = *reinterpret_cast<const bf16x8*>(qp + (int64_t)src * st.qs + dg);
-          d8 = *reinterpret_cast<const bf16x8*>(dop + (int64_t)src * st.dos + dg);
-        }
-        *reinterpret_cast<bf16x8*>(&q_lds[qr][dg]) = q8;
-        *reinterpret_cast<bf16x8*>(&do_lds[qr][dg]) = d8;
+  const int q_start = causal ? kvb_idx * (16 * NW) / 64 * 64 : 0;
+  const int n_tiles = (S - q_start + 63) / 64;
+
+  bf16x8 qreg[2], doreg[2];
+  float lse_reg, delta_reg;
+  auto issue_loads = [&](int tile) {
+    int t = threadIdx.x;
+    bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+    qreg[0] = qreg[1] = doreg[0] = doreg[1] = z;
+    int qb0 = q_start + tile * 64;
+    if (t < PAIRS) {
+      int qr = (t / GPR) * 2, dg = (t % GPR) * 8;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          qt_lds[dg + j][qr] = q8[j];
-          dot_lds[dg + j][qr] = d8[j];
+      for (int u = 0; u < 2; ++u) {
+        int src = qb0 + qr + u;
+        if (src < S && dg + 8 <= D) {
+          qreg[u] = *reinterpret_cast<const bf16x8*>(
+              qp + (int64_t)src * st.qs + dg);
+          doreg[u] = *reinterpret_cast<const bf16x8*>(
+              dop + (int64_t)src * st.dos + dg);
         }
-      }
-      for (int t = threadIdx.x; t < 64; t += ATTN_THREADS) {
-        int src = qb + t;
-        lse_lds[t] = (src < S) ? lse[(int64_t)bh * S + src] : 0.f;
-        delta_lds[t] = (src < S) ? delta[(int64_t)bh * S + src] : 0.f;
       }
     }
-    __syncthreads();
+    lse_reg = delta_reg = 0.f;
+    if (t < 64) {
+      int src = qb0 + t;
+      if (src < S) {
+        lse_reg = lse[(int64_t)bh * S + src];
+        delta_reg = delta[(int64_t)bh * S + src];
+      }
+    }
+  };
+  auto write_tile = [&]() {
+    int t = threadIdx.x;
+    if (t < PAIRS) {
+      int qr = (t / GPR) * 2, dg = (t % GPR) * 8;
+      *reinterpret_cast<bf16x8*>(&q_lds[qr][dg]) = qreg[0];
+      *reinterpret_cast<bf16x8*>(&q_lds[qr + 1][dg]) = qreg[1];
+      *reinterpret_cast<bf16x8*>(&do_lds[qr][dg]) = doreg[0];
+      *reinterpret_cast<bf16x8*>(&do_lds[qr + 1][dg]) = doreg[1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        short2 pq, pd;
+        pq.x = qreg[0][j];
+        pq.y = qreg[1][j];
+        pd.x = doreg[0][j];
+        pd.y = doreg[1][j];
+        *reinterpret_cast<short2*>(&qt_lds[dg + j][qr]) = pq;
+        *reinterpret_cast<short2*>(&dot_lds[dg + j][qr]) = pd;
+      }
+    }
+    if (t < 64) {
+      lse_lds[t] = lse_reg;
+      delta_lds[t] = delta_reg;
+    }
+  };
 
-    // S^T and dP^T tiles (rows = kv, cols = q)
+  if (n_tiles > 0) {
+    issue_loads(0);
+    write_tile();
+    __syncthreads();
+  }
+
+  for (int ti = 0; ti < n_tiles; ++ti) {
+    const int qb0 = q_start + ti * 64;
+    if (ti + 1 < n_tiles) issue_loads(ti + 1);
+
+    // S^T and dP^T tiles (C: row = kv (hi*4+r), col = q (nt*16+lo))
     f32x4 st_acc[NT], dpt_acc[NT];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
@@ -617,25 +785,25 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dkv_kernel(
       dpt_acc[nt] = da;
     }
 
-    // P^T -> p_lds for dV; then dS^T -> p_lds for dK (two passes over the
-    // same per-wave buffer, separated by the MFMA consumption)
     float pt[NT][4], dst[NT][4];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
-      int q_idx = qb + nt * 16 + lo;
+      int q_idx = qb0 + nt * 16 + lo;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int kv_idx = kvb_idx * 64 + wave * 16 + hi * 4 + r;
+        int kv_idx = kv_row0 + hi * 4 + r;
         bool masked = (q_idx >= S) || (kv_idx >= Skv) ||
                       (causal && kv_idx > q_idx);
-        float p = masked ? 0.f
-                         : __expf(st_acc[nt][r] * scale - lse_lds[nt * 16 + lo]);
-        pt[nt][r] = p;
-        dst[nt][r] = p * (dpt_acc[nt][r] - delta_lds[nt * 16 + lo]) * scale;
+        float pv = masked ? 0.f
+                          : __expf(st_acc[nt][r] * scale -
+                                   lse_lds[nt * 16 + lo]);
+        pt[nt][r] = pv;
+        dst[nt][r] =
+            pv * (dpt_acc[nt][r] - delta_lds[nt * 16 + lo]) * scale;
       }
     }
 
-    // dV += P^T @ dO (A = P^T, B = dOt)
+    // dV += P^T @ dO (A = P^T via p_lds, B = dOt)
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt)
 #pragma unroll
@@ -649,13 +817,12 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dkv_kernel(
       for (int dt = 0; dt < DT; ++dt) {
         bf16x8 b = *reinterpret_cast<const bf16x8*>(
             &dot_lds[dt * 16 + lo][ks * 32 + hi * 8]);
-        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
-                                                             dv_acc[dt], 0,
-                                                             0, 0);
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, b, dv_acc[dt], 0, 0, 0);
       }
     }
 
-    // dK += dS^T @ Q (A = dS^T, B = Qt)
+    // dK += dS^T @ Q (A = dS^T via p_lds, B = Qt)
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt)
 #pragma unroll
@@ -669,17 +836,21 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_bwd_dkv_kernel(
       for (int dt = 0; dt < DT; ++dt) {
         bf16x8 b = *reinterpret_cast<const bf16x8*>(
             &qt_lds[dt * 16 + lo][ks * 32 + hi * 8]);
-        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b,
-                                                             dk_acc[dt], 0,
-                                                             0, 0);
+        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, b, dk_acc[dt], 0, 0, 0);
       }
+    }
+
+    __syncthreads();
+    if (ti + 1 < n_tiles) {
+      write_tile();
+      __syncthreads();
     }
   }
 
-  // store dK, dV
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    int kv_idx = kvb_idx * 64 + wave * 16 + hi * 4 + r;
+    int kv_idx = kv_row0 + hi * 4 + r;
     if (kv_idx >= Skv) continue;
 #pragma unroll
     for (int dt = 0; dt < DT; ++dt) {
@@ -719,9 +890,11 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
         (const short*)dout, (const short*)o, delta_ws, rows, (int)H, (int)S,
         (int)D, st.dob, st.doh, st.dos, p[21], p[22], p[23]);
   }
-  dim3 block(ATTN_THREADS);
-  dim3 grid_q((uint32_t)ceil_div(S, 64), (uint32_t)(B * H));
-  dim3 grid_kv((uint32_t)ceil_div(Skv, 64), (uint32_t)(B * H));
+  dim3 block(ATTN_BWD_THREADS);
+  dim3 grid_q((uint32_t)ceil_div(S, 16 * (ATTN_BWD_THREADS / 64)),
+              (uint32_t)(B * H));
+  dim3 grid_kv((uint32_t)ceil_div(Skv, 16 * (ATTN_BWD_THREADS / 64)),
+               (uint32_t)(B * H));
 #define LAUNCH_BWD(DP)                                                       \
   do {                                                                       \
     attn_bwd_dq_kernel<DP><<<grid_q, block, 0, stream>>>(                    \

@@ -43,6 +43,10 @@ hipError_t launch_attn_fwd(const void*, const void*, const void*, void*,
                            float*, int64_t, int64_t, int64_t, int64_t,
                            int64_t, float, int, const int64_t*, hipStream_t);
 hipError_t launch_mfma_probe(const void*, const void*, float*, hipStream_t);
+hipError_t launch_attn_fwd_ablate(const void*, const void*, const void*,
+                                  void*, float*, int64_t, int64_t, int64_t,
+                                  int64_t, int64_t, float, int,
+                                  const int64_t*, int, hipStream_t);
 hipError_t launch_attn_bwd(const void*, const void*, const void*,
                            const void*, const void*, const float*, float*,
                            void*, void*, void*, int64_t, int64_t, int64_t,
@@ -234,6 +238,24 @@ at::Tensor attn_fwd_out(const at::Tensor& q, const at::Tensor& k,
   return o;
 }
 
+at::Tensor attn_fwd_ablate(const at::Tensor& q, const at::Tensor& k,
+                           const at::Tensor& v, bool causal, double scale,
+                           int64_t abl) {
+  int64_t B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  int64_t strides[12] = {q.stride(0), q.stride(1), q.stride(2),
+                         k.stride(0), k.stride(1), k.stride(2),
+                         v.stride(0), v.stride(1), v.stride(2),
+                         o.stride(0), o.stride(1), o.stride(2)};
+  HIP_OK(launch_attn_fwd_ablate(q.const_data_ptr(), k.const_data_ptr(),
+                                v.const_data_ptr(), o.mutable_data_ptr(),
+                                (float*)lse.mutable_data_ptr(), B, H, S,
+                                k.size(2), D, (float)scale, causal ? 1 : 0,
+                                strides, (int)abl, cur_stream()));
+  return o;
+}
+
 std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                  const at::Tensor& v, bool causal,
                                  double scale) {
@@ -353,6 +375,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE bwd (gfx950)");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)");
   m.def("attn_fwd_out", &attn_fwd_out, "flash attention fwd, strided out");
+  m.def("attn_fwd_ablate", &attn_fwd_ablate, "ablation variants (perf)");
   m.def("attn_bwd", &attn_bwd, "flash attention bwd (gfx950 MFMA)");
   m.def("attn_bwd_out", &attn_bwd_out, "flash attention bwd, strided out");
   m.def("attn_bwd_blocked", &attn_bwd_blocked,

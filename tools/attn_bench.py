@@ -29,6 +29,7 @@ def timeit(fn, iters):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--iters", type=int, default=20)
+    p.add_argument("--ablate", action="store_true")
     p.add_argument("--B", type=int, default=8)
     p.add_argument("--H", type=int, default=32)
     p.add_argument("--S", type=int, default=1024)
@@ -45,6 +46,14 @@ def main():
     fwd_flops = 4 * B * H * S * S * D * 0.5  # causal
     t = timeit(lambda: ext.attn_fwd(q, k, v, True, scale), args.iters)
     print(f"fwd:  {t*1e6:8.1f} us  {fwd_flops/t/1e12:7.1f} TF")
+
+    if args.ablate:
+        names = {0: "full", 1: "noPV", 2: "noSM(QK+pwrite)",
+                 3: "QKonly", 4: "staging"}
+        for abl in (0, 1, 2, 3, 4):
+            t = timeit(lambda: ext.attn_fwd_ablate(q, k, v, True, scale,
+                                                   abl), args.iters)
+            print(f"  abl{abl} {names[abl]:16s} {t*1e6:8.1f} us")
 
     o, lse = ext.attn_fwd(q, k, v, True, scale)
     do = torch.randn_like(o)
