@@ -1,0 +1,185 @@
+"""Cross-mesh resharding: tile algebra + send/recv plan + execution.
+
+Capability analog of the reference's ``pipeline_parallel/
+cross_mesh_resharding.py`` (1903 LoC) and ``resharding_tensor.py``:
+`VirtualDistributedArray:25` (sharding spec -> per-device tiles without
+materialization), `ReshardingTaskSpec:674` with
+`dst_tile_to_src_tiles_map:718` (tile intersection algebra), and the
+send/recv strategy (`SymbolicReshardingTask:184` compiles per-worker
+send/recv tile lists).
+
+Used when a tensor sharded over one mesh (src placement) must move to a
+different sharding on another (or the same) mesh — heterogeneous pipeline
+stage boundaries, plan changes on restore, mesh resizing.  Execution is
+RCCL/gloo p2p: every (src_rank, dst_rank) pair exchanges exactly the
+overlap of their tiles; deterministic task order (sorted by rank pair)
+keeps the exchange deadlock-free (reference sorts strategy order,
+SURVEY.md §5.2).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..mesh import is_distributed, rank
+
+Index = Tuple[Tuple[int, int], ...]  # per-dim (start, stop)
+
+
+@dataclass(frozen=True)
+class Placement:
+    """A tensor distributed over `ranks`: dim_partitions[d] = how many ways
+    dim d is split; rank_grid maps partition coordinates -> global rank.
+
+    This is the executable core of the reference's sharding-spec ->
+    tile mapping (VirtualDistributedArray, resharding_tensor.py:25).
+    """
+    global_shape: Tuple[int, ...]
+    dim_partitions: Tuple[int, ...]
+    ranks: Tuple[int, ...]  # row-major over partition coords; replicas allowed
+
+    def __post_init__(self):
+        n_tiles = int(np.prod(self.dim_partitions))
+        assert len(self.ranks) % n_tiles == 0, \
+            (len(self.ranks), self.dim_partitions)
+
+    @property
+    def n_replicas(self) -> int:
+        return len(self.ranks) // int(np.prod(self.dim_partitions))
+
+    def tile_index(self, coord: Sequence[int]) -> Index:
+        out = []
+        for d, c in enumerate(coord):
+            n = self.dim_partitions[d]
+            size = self.global_shape[d]
+            assert size % n == 0, (size, n)
+            per = size // n
+            out.append((c * per, (c + 1) * per))
+        return tuple(out)
+
+    def tiles(self) -> List[Tuple[Index, List[int]]]:
+        """[(index, [owner ranks (replicas)])] in row-major coord order."""
+        coords = np.indices(self.dim_partitions).reshape(
+            len(self.dim_partitions), -1).T
+        n_tiles = len(coords)
+        out = []
+        for i, c in enumerate(coords):
+            owners = [self.ranks[r * n_tiles + i]
+                      for r in range(self.n_replicas)]
+            out.append((self.tile_index(c), owners))
+        return out
+
+    def rank_tile(self, r: int) -> Optional[Index]:
+        for idx, owners in self.tiles():
+            if r in owners:
+                return idx
+        return None
+
+
+def _intersect(a: Index, b: Index) -> Optional[Index]:
+    out = []
+    for (a0, a1), (b0, b1) in zip(a, b):
+        lo, hi = max(a0, b0), min(a1, b1)
+        if lo >= hi:
+            return None
+        out.append((lo, hi))
+    return tuple(out)
+
+
+@dataclass(frozen=True)
+class TileTransfer:
+    src_rank: int
+    dst_rank: int
+    region: Index        # global coordinates of the overlap
+    src_offset: Index    # region relative to the src tile
+    dst_offset: Index    # region relative to the dst tile
+
+
+@dataclass
+class ReshardingTaskSpec:
+    """All transfers needed to convert src placement -> dst placement
+    (reference ReshardingTaskSpec + dst_tile_to_src_tiles_map)."""
+    src: Placement
+    dst: Placement
+    transfers: List[TileTransfer]
+
+    @staticmethod
+    def build(src: Placement, dst: Placement) -> "ReshardingTaskSpec":
+        assert src.global_shape == dst.global_shape
+        transfers = []
+        src_tiles = src.tiles()
+        for dst_idx, dst_owners in dst.tiles():
+            for d_own in dst_owners:
+                for src_idx, src_owners in src_tiles:
+                    inter = _intersect(dst_idx, src_idx)
+                    if inter is None:
+                        continue
+                    # pick the source replica deterministically (balance by
+                    # dst rank, reference load-balancing is fancier)
+                    s_own = src_owners[d_own % len(src_owners)]
+                    transfers.append(TileTransfer(
+                        src_rank=s_own, dst_rank=d_own, region=inter,
+                        src_offset=tuple((lo - s0, hi - s0) for (lo, hi),
+                                         (s0, _) in zip(inter, src_idx)),
+                        dst_offset=tuple((lo - d0, hi - d0) for (lo, hi),
+                                         (d0, _) in zip(inter, dst_idx))))
+        # deterministic global order => deadlock-free paired exchange
+        transfers.sort(key=lambda t: (t.src_rank, t.dst_rank, t.region))
+        return ReshardingTaskSpec(src, dst, transfers)
+
+    def total_bytes(self, elem_size: int = 2) -> int:
+        n = 0
+        for t in self.transfers:
+            if t.src_rank != t.dst_rank:
+                n += int(np.prod([hi - lo for lo, hi in t.region])) * elem_size
+        return n
+
+
+def _slice(t: torch.Tensor, idx: Index) -> torch.Tensor:
+    return t[tuple(slice(lo, hi) for lo, hi in idx)]
+
+
+def execute_resharding(spec: ReshardingTaskSpec, local_src: Optional[torch.Tensor],
+                       dst_buf: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
+    """Run this rank's part of the exchange.
+
+    local_src: this rank's src tile (or None if it owns none);
+    dst_buf: preallocated dst tile (or None).  Local overlaps are copies;
+    remote ones are batched isend/irecv (one batch per peer pair —
+    both sides post together, so the exchange cannot deadlock).
+    """
+    me = rank()
+    # local copies first
+    for t in spec.transfers:
+        if t.src_rank == me and t.dst_rank == me:
+            _slice(dst_buf, t.dst_offset).copy_(_slice(local_src,
+                                                       t.src_offset))
+    if not is_distributed():
+        return dst_buf
+    ops = []
+    staged = []
+    for t in spec.transfers:
+        if t.src_rank == t.dst_rank:
+            continue
+        if t.src_rank == me:
+            payload = _slice(local_src, t.src_offset).contiguous()
+            staged.append(payload)
+            ops.append(dist.P2POp(dist.isend, payload, t.dst_rank))
+        elif t.dst_rank == me:
+            shape = tuple(hi - lo for lo, hi in t.region)
+            buf = torch.empty(shape, dtype=dst_buf.dtype,
+                              device=dst_buf.device)
+            staged.append((buf, t))
+            ops.append(dist.P2POp(dist.irecv, buf, t.src_rank))
+    if ops:
+        for w in dist.batch_isend_irecv(ops):
+            w.wait()
+    for item in staged:
+        if isinstance(item, tuple):
+            buf, t = item
+            _slice(dst_buf, t.dst_offset).copy_(buf)
+    return dst_buf

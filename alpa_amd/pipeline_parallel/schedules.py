@@ -61,6 +61,31 @@ def inference_schedule(num_stages: int, num_microbatches: int
             for s in range(num_stages)]
 
 
+def overlap_friendly_1f1b_schedule(num_stages: int, num_microbatches: int,
+                                   extra_warmup: int = 1
+                                   ) -> List[List[Tuple[str, int]]]:
+    """1F1B with extra forward warmup so cross-stage sends can overlap
+    compute (reference OverlapFriendlyPipeDreamSchedule:452): stage s runs
+    min(P - s - 1 + extra_warmup, M) warmup forwards — deeper in-flight
+    buffering trades a little activation memory for comm/compute overlap
+    headroom on the xGMI links."""
+    P, M = num_stages, num_microbatches
+    out = []
+    for s in range(P):
+        warmup = min(P - s - 1 + extra_warmup, M)
+        steady = M - warmup
+        instrs: List[Tuple[str, int]] = []
+        for i in range(warmup):
+            instrs.append((FWD, i))
+        for i in range(steady):
+            instrs.append((FWD, warmup + i))
+            instrs.append((BWD, i))
+        for i in range(steady, M):
+            instrs.append((BWD, i))
+        out.append(instrs)
+    return out
+
+
 def make_schedule(name: str, num_stages: int, num_microbatches: int):
     if name == "gpipe":
         return gpipe_schedule(num_stages, num_microbatches)
@@ -68,6 +93,8 @@ def make_schedule(name: str, num_stages: int, num_microbatches: int):
         return one_f_one_b_schedule(num_stages, num_microbatches)
     if name == "inference":
         return inference_schedule(num_stages, num_microbatches)
+    if name == "1f1b_overlap_friendly":
+        return overlap_friendly_1f1b_schedule(num_stages, num_microbatches)
     raise ValueError(f"unknown schedule {name!r}")
 
 

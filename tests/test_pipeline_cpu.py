@@ -153,3 +153,13 @@ def test_pipeline4_matches_serial(nmb):
     for r in results:
         for a, b in zip(r, serial):
             assert abs(a - b) < 2e-4, (r, serial)
+
+
+def test_overlap_friendly_schedule_invariants():
+    sched = schedules.overlap_friendly_1f1b_schedule(4, 8)
+    for s, instrs in enumerate(sched):
+        fwd = [mb for op, mb in instrs if op == schedules.FWD]
+        bwd = [mb for op, mb in instrs if op == schedules.BWD]
+        assert fwd == list(range(8)) and bwd == list(range(8))
+        # one extra in-flight forward vs plain 1F1B
+        assert schedules.peak_live_activations(instrs) <= min(4 - s + 1, 8)
