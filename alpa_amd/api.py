@@ -74,17 +74,85 @@ class TrainState:
         model = model_fn(mesh=mesh, axis=method.tp_axis, dtype=dtype,
                          device=device())
         state = cls(model, None, method, mesh)
-        state._install_grad_sync()
-        if method.zero_stage >= 2:
+        if method.zero_stage >= 3:
+            state._install_zero3(lr, betas, weight_decay)
+        elif method.zero_stage == 2:
+            state._install_grad_sync()
             from .parallel.zero import ZeroOptimizer
             state.optimizer = ZeroOptimizer(state.grad_sync, lr=lr,
                                             betas=betas,
                                             weight_decay=weight_decay)
         else:
+            state._install_grad_sync()
             state.optimizer = optimizer_cls(model.parameters(), lr=lr,
                                             betas=betas,
                                             weight_decay=weight_decay)
         return state
+
+    def _install_zero3(self, lr, betas, weight_decay):
+        """ZeRO-3: block params sharded + JIT-gathered (parallel/zero3.py);
+        leftover params (e.g. positional embeddings) use plain DP sync."""
+        from .parallel.zero3 import Zero3Manager, Zero3Optimizer
+        model, m = self.model, self.method
+        if hasattr(model, "zero3_blocks"):
+            blocks = model.zero3_blocks()
+        else:
+            blocks = []
+            for child in model.children():
+                if isinstance(child, torch.nn.ModuleList):
+                    blocks.extend(child)
+                elif any(p.requires_grad for p in child.parameters()):
+                    blocks.append(child)
+        # drop blocks whose params are already covered (tied weights)
+        seen = set()
+        uniq = []
+        for b in blocks:
+            ps = [p for p in b.parameters() if p.requires_grad and
+                  id(p) not in seen]
+            if ps:
+                uniq.append(b)
+                seen.update(id(p) for p in ps)
+        manager = Zero3Manager(uniq, self.mesh, axis=m.dp_axis)
+        rest = [p for p in model.parameters()
+                if p.requires_grad and id(p) not in manager._by_param]
+        self.grad_sync = GradSynchronizer(rest, self.mesh, axis=m.dp_axis)
+        z3 = Zero3Optimizer(manager, lr=lr, betas=betas,
+                            weight_decay=weight_decay)
+        rest_opt = AdamW(rest, lr=lr, betas=betas,
+                         weight_decay=weight_decay) if rest else None
+        state_self = self
+
+        class _Composite:
+
+            def __init__(self):
+                self.step_count = 0
+
+            def step(self, grads=None, grad_scale: float = 1.0):
+                z3.step(grad_scale=grad_scale)
+                if rest_opt is not None:
+                    rest_opt.step(grad_scale=grad_scale)
+                # shard grads accumulated across microbatches: clear for
+                # the next step
+                manager.zero_grads()
+                self.step_count += 1
+
+            def zero_grad(self):
+                manager.zero_grads()
+
+            def state_dict(self):
+                sd = {"zero3": z3.state_dict(), "step": self.step_count}
+                if rest_opt is not None:
+                    sd["rest"] = rest_opt.state_dict()
+                return sd
+
+            def load_state_dict(self, sd):
+                z3.load_state_dict(sd["zero3"])
+                if rest_opt is not None and "rest" in sd:
+                    rest_opt.load_state_dict(sd["rest"])
+                self.step_count = sd.get("step", 0)
+
+        self.zero3_manager = manager
+        self.optimizer = _Composite()
 
     def _install_grad_sync(self):
         m = self.method
