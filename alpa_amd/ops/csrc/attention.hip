@@ -164,7 +164,6 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     }
 
     // ---- S^T = K Q^T (A = K tile from LDS, B = Q fragments) ----
-    __builtin_amdgcn_s_setprio(1);
     f32x4 s_acc[NTILES];
 #pragma unroll
     for (int nt = 0; nt < NTILES; ++nt) {
@@ -179,7 +178,6 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
       s_acc[nt] = acc;
     }
 
-    __builtin_amdgcn_s_setprio(0);
     if (ABL >= 3) {  // QK^T only
 #pragma unroll
       for (int nt = 0; nt < NTILES; ++nt)
@@ -253,7 +251,6 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     }
 
     // ---- O^T += V^T P^T  (A = V^T from vt_lds, B = P^T from p_lds) ----
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < ATTN_BLOCK_K / 32; ++ks) {
       bf16x8 b = *reinterpret_cast<const bf16x8*>(
@@ -267,7 +264,6 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
       }
     }
 
-    __builtin_amdgcn_s_setprio(0);
     if (ti + 1 < n_tiles) {
       write_tile(1 - cur);  // waits on the prefetched loads here
       __syncthreads();
@@ -558,7 +554,6 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     if (ti + 1 < n_tiles) issue_loads(ti + 1);
 
     // S and dP tiles (C: row = q (hi*4+r), col = kv (nt*16+lo))
-    __builtin_amdgcn_s_setprio(1);
     f32x4 s_acc[NT], dp_acc[NT];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
@@ -578,7 +573,6 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
       dp_acc[nt] = da;
     }
 
-    __builtin_amdgcn_s_setprio(0);
     // dS = P * (dP - delta) * scale -> p_lds as the dS A-operand
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
@@ -595,7 +589,6 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     }
 
     // dQ += dS @ K  (A = dS from p_lds, B = Kt)
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
@@ -609,7 +602,6 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
       }
     }
 
-    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
     if (ti + 1 < n_tiles) {
       write_tile();
@@ -774,7 +766,6 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     if (ti + 1 < n_tiles) issue_loads(ti + 1);
 
     // S^T and dP^T tiles (C: row = kv (hi*4+r), col = q (nt*16+lo))
-    __builtin_amdgcn_s_setprio(1);
     f32x4 st_acc[NT], dpt_acc[NT];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
@@ -794,7 +785,6 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       dpt_acc[nt] = da;
     }
 
-    __builtin_amdgcn_s_setprio(0);
     float pt[NT][4], dst[NT][4];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
@@ -814,7 +804,6 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     }
 
     // dV += P^T @ dO (A = P^T via p_lds, B = dOt)
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt)
 #pragma unroll
@@ -833,9 +822,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       }
     }
 
-    __builtin_amdgcn_s_setprio(0);
     // dK += dS^T @ Q (A = dS^T via p_lds, B = Qt)
-    __builtin_amdgcn_s_setprio(1);
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt)
 #pragma unroll
@@ -854,7 +841,6 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       }
     }
 
-    __builtin_amdgcn_s_setprio(0);
     __syncthreads();
     if (ti + 1 < n_tiles) {
       write_tile();
