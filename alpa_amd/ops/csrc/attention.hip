@@ -156,6 +156,12 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     const int cur = ti & 1;
     if (ti + 1 < n_tiles) issue_loads(ti + 1);  // lands under the compute
 
+    // causal: tiles entirely above this wave's q rows contribute nothing
+    if (causal && kvb > q_row0 + 15) {
+      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      continue;
+    }
+
     if (ABL >= 4) {  // staging-only
       float keep = bf2f(k_lds[cur][lane][0]) + bf2f(vt_lds[cur][lane][0]);
       asm volatile("" ::"v"(keep));
@@ -438,7 +444,7 @@ __global__ void attn_bwd_preprocess_kernel(const short* __restrict__ dout,
 //   dS = P*(dP-delta)*scale
 //   dQ += dS K         (A=dS via p_lds, B=Kt_lds transposed)
 template <int Dp>
-__global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(ATTN_BWD_THREADS, 4) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -553,6 +559,13 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     const int kvb = ti * 64;
     if (ti + 1 < n_tiles) issue_loads(ti + 1);
 
+    // causal: kv tiles entirely above this wave's q rows are masked out
+    if (causal && kvb > q_row0 + 15) {
+      __syncthreads();
+      if (ti + 1 < n_tiles) { write_tile(); __syncthreads(); }
+      continue;
+    }
+
     // S and dP tiles (C: row = q (hi*4+r), col = kv (nt*16+lo))
     f32x4 s_acc[NT], dp_acc[NT];
 #pragma unroll
@@ -631,8 +644,8 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
 //   dP^T = V dO^T      (A=V regs, B=dO_lds row-major)
 //   dS^T = P^T*(dP^T - delta[q])*scale
 //   dK += dS^T Q       (A=dS^T via p_lds, B=Qt_lds)
-template <int Dp>
-__global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
+template <int Dp, int ABL = 0>
+__global__ __launch_bounds__(ATTN_BWD_THREADS, 4) void attn_bwd_dkv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -765,6 +778,21 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     const int qb0 = q_start + ti * 64;
     if (ti + 1 < n_tiles) issue_loads(ti + 1);
 
+    // causal: q tiles entirely before this wave's kv rows are masked out
+    if (causal && qb0 + 63 < kv_row0) {
+      __syncthreads();
+      if (ti + 1 < n_tiles) { write_tile(); __syncthreads(); }
+      continue;
+    }
+
+    if (ABL >= 3) {  // staging only
+      float keep = bf2f(q_lds[lane & 63][0]) + bf2f(dot_lds[0][lane & 63]) +
+                   bf2f(qt_lds[0][lane & 63]) + bf2f(do_lds[lane & 63][0]);
+      asm volatile("" ::"v"(keep));
+      __syncthreads();
+      if (ti + 1 < n_tiles) { write_tile(); __syncthreads(); }
+      continue;
+    }
     // S^T and dP^T tiles (C: row = kv (hi*4+r), col = q (nt*16+lo))
     f32x4 st_acc[NT], dpt_acc[NT];
 #pragma unroll
@@ -785,7 +813,10 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       dpt_acc[nt] = da;
     }
 
-    float pt[NT][4], dst[NT][4];
+    // P^T written to p_lds immediately; dS^T retained PACKED as bf16
+    // (8 VGPRs, not 16 floats) — keeps the kernel under the 128-VGPR
+    // occupancy cliff (166 VGPR cost 1 wave/SIMD before this)
+    short dst_pk[NT][4];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
       int q_idx = qb0 + nt * 16 + lo;
@@ -797,18 +828,21 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
         float pv = masked ? 0.f
                           : __expf(st_acc[nt][r] * scale -
                                    lse_lds[nt * 16 + lo]);
-        pt[nt][r] = pv;
-        dst[nt][r] =
-            pv * (dpt_acc[nt][r] - delta_lds[nt * 16 + lo]) * scale;
+        p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(pv);
+        dst_pk[nt][r] =
+            f2bf(pv * (dpt_acc[nt][r] - delta_lds[nt * 16 + lo]) * scale);
       }
     }
 
+    if (ABL >= 2) {  // S^T/dP^T + softmax math only
+#pragma unroll
+      for (int nt = 0; nt < NT; ++nt)
+        asm volatile("" ::"v"((int)dst_pk[nt][0]), "v"((int)dst_pk[nt][3]));
+      __syncthreads();
+      if (ti + 1 < n_tiles) { write_tile(); __syncthreads(); }
+      continue;
+    }
     // dV += P^T @ dO (A = P^T via p_lds, B = dOt)
-#pragma unroll
-    for (int nt = 0; nt < NT; ++nt)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(pt[nt][r]);
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
@@ -822,12 +856,17 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       }
     }
 
+    if (ABL >= 1) {  // skip the dK group
+      __syncthreads();
+      if (ti + 1 < n_tiles) { write_tile(); __syncthreads(); }
+      continue;
+    }
     // dK += dS^T @ Q (A = dS^T via p_lds, B = Qt)
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt)
 #pragma unroll
       for (int r = 0; r < 4; ++r)
-        p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(dst[nt][r]);
+        p_lds[wave][hi * 4 + r][nt * 16 + lo] = dst_pk[nt][r];
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
@@ -864,6 +903,38 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
 }
 
 extern "C" {
+
+hipError_t launch_attn_bwd_dkv_ablate(const void* q, const void* k,
+                                      const void* v, const void* dout,
+                                      const float* lse, const float* delta,
+                                      void* dk, void* dv, int64_t B,
+                                      int64_t H, int64_t S, int64_t Skv,
+                                      int64_t D, float scale, int causal,
+                                      const int64_t* strides, int abl,
+                                      hipStream_t stream) {
+  AttnBwdStrides st;
+  const int64_t* pp = strides;
+  st.qb = pp[0]; st.qh = pp[1]; st.qs = pp[2];
+  st.kb = pp[3]; st.kh = pp[4]; st.ks = pp[5];
+  st.vb = pp[6]; st.vh = pp[7]; st.vs = pp[8];
+  st.dob = pp[9]; st.doh = pp[10]; st.dos = pp[11];
+  st.dqb = pp[12]; st.dqh = pp[13]; st.dqs = pp[14];
+  st.dkb = pp[15]; st.dkh = pp[16]; st.dks = pp[17];
+  st.dvb = pp[18]; st.dvh = pp[19]; st.dvs = pp[20];
+  dim3 block(ATTN_BWD_THREADS);
+  dim3 grid_kv((uint32_t)ceil_div(Skv, 16 * (ATTN_BWD_THREADS / 64)),
+               (uint32_t)(B * H));
+  if (D > 96) return hipErrorInvalidValue;
+#define DKV_CASE(A)                                                        \
+  if (abl == A)                                                            \
+    attn_bwd_dkv_kernel<96, A><<<grid_kv, block, 0, stream>>>(             \
+        (const short*)q, (const short*)k, (const short*)v,                 \
+        (const short*)dout, lse, delta, (short*)dk, (short*)dv, (int)H,    \
+        (int)S, (int)Skv, (int)D, scale, causal, st);
+  DKV_CASE(0) DKV_CASE(1) DKV_CASE(2) DKV_CASE(3)
+#undef DKV_CASE
+  return hipGetLastError();
+}
 
 hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
                            const void* o, const void* dout,

@@ -47,6 +47,12 @@ hipError_t launch_attn_fwd_ablate(const void*, const void*, const void*,
                                   void*, float*, int64_t, int64_t, int64_t,
                                   int64_t, int64_t, float, int,
                                   const int64_t*, int, hipStream_t);
+hipError_t launch_attn_bwd_dkv_ablate(const void*, const void*,
+                                      const void*, const void*,
+                                      const float*, const float*, void*,
+                                      void*, int64_t, int64_t, int64_t,
+                                      int64_t, int64_t, float, int,
+                                      const int64_t*, int, hipStream_t);
 hipError_t launch_attn_bwd(const void*, const void*, const void*,
                            const void*, const void*, const float*, float*,
                            void*, void*, void*, int64_t, int64_t, int64_t,
@@ -355,6 +361,28 @@ std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
   return {dq, dk, dv};
 }
 
+void attn_bwd_dkv_ablate(const at::Tensor& dout, const at::Tensor& q,
+                         const at::Tensor& k, const at::Tensor& v,
+                         const at::Tensor& lse, const at::Tensor& delta,
+                         at::Tensor dk, at::Tensor dv, bool causal,
+                         double scale, int64_t abl) {
+  int64_t B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  int64_t strides[24] = {
+      q.stride(0), q.stride(1), q.stride(2),
+      k.stride(0), k.stride(1), k.stride(2),
+      v.stride(0), v.stride(1), v.stride(2),
+      dout.stride(0), dout.stride(1), dout.stride(2),
+      0, 0, 0,
+      dk.stride(0), dk.stride(1), dk.stride(2),
+      dv.stride(0), dv.stride(1), dv.stride(2), 0, 0, 0};
+  HIP_OK(launch_attn_bwd_dkv_ablate(
+      q.const_data_ptr(), k.const_data_ptr(), v.const_data_ptr(),
+      dout.const_data_ptr(), (const float*)lse.const_data_ptr(),
+      (const float*)delta.const_data_ptr(), dk.mutable_data_ptr(),
+      dv.mutable_data_ptr(), B, H, S, k.size(2), D, (float)scale,
+      causal ? 1 : 0, strides, (int)abl, cur_stream()));
+}
+
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
   check_bf16_contig(a, "a");
   auto d = at::empty({16, 16}, a.options().dtype(at::kFloat));
@@ -376,6 +404,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)");
   m.def("attn_fwd_out", &attn_fwd_out, "flash attention fwd, strided out");
   m.def("attn_fwd_ablate", &attn_fwd_ablate, "ablation variants (perf)");
+  m.def("attn_bwd_dkv_ablate", &attn_bwd_dkv_ablate, "dkv ablation (perf)");
   m.def("attn_bwd", &attn_bwd, "flash attention bwd (gfx950 MFMA)");
   m.def("attn_bwd_out", &attn_bwd_out, "flash attention bwd, strided out");
   m.def("attn_bwd_blocked", &attn_bwd_blocked,

@@ -1,0 +1,50 @@
+"""Serve an OPT model with auto-sharded TP + HTTP frontend (the reference's
+examples/llm_serving workflow; random-init weights — no network access).
+
+  torchrun --standalone --nproc-per-node N examples/serve_opt.py \
+      --model 1.3B --port 8265
+Rank 0 serves HTTP; all ranks participate in TP generation.
+"""
+import argparse
+import os
+import sys
+
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                ".."))
+import alpa_amd as aa
+from alpa_amd.models.opt import OPTModel, opt_config
+from alpa_amd.serve import Controller, run_controller
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default="125M")
+    p.add_argument("--port", type=int, default=8265)
+    p.add_argument("--demo", action="store_true",
+                   help="run one generation and exit (no HTTP server)")
+    args = p.parse_args()
+
+    aa.init()
+    mesh = aa.full_mesh((1, aa.world_size()))
+    dtype = torch.bfloat16 if torch.cuda.is_available() else torch.float32
+    cfg = opt_config(args.model, max_seq_len=512)
+    model = OPTModel(cfg, mesh, 1, dtype, aa.device(), init_seed=0)
+
+    if args.demo or aa.world_size() > 1:
+        ids = torch.randint(0, cfg.vocab_size, (1, 8), device=aa.device())
+        out = model.generate(ids, max_new_tokens=8)
+        if aa.rank() == 0:
+            print("generated:", out.tolist())
+        if args.demo:
+            aa.shutdown()
+            return
+    c = Controller()
+    c.register_model(f"opt-{args.model}",
+                     lambda ids, mt: model.generate(ids.to(aa.device()), mt))
+    run_controller(c, port=args.port)
+
+
+if __name__ == "__main__":
+    main()

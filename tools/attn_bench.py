@@ -63,6 +63,17 @@ def main():
                args.iters)
     print(f"bwd:  {t*1e6:8.1f} us  {bwd_flops/t/1e12:7.1f} TF")
 
+    if args.ablate:
+        delta = (do.float() * o.float()).sum(-1)
+        dk = torch.empty_like(k)
+        dv = torch.empty_like(v)
+        names = {0: "dkv full", 1: "no dK", 2: "S/dP+math", 3: "staging"}
+        for abl in (0, 1, 2, 3):
+            t = timeit(lambda: ext.attn_bwd_dkv_ablate(
+                do, q, k, v, lse3, delta, dk, dv, True, scale, abl),
+                args.iters)
+            print(f"  dkv abl{abl} {names[abl]:12s} {t*1e6:8.1f} us")
+
 
 if __name__ == "__main__":
     main()
