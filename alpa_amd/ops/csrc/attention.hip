@@ -444,7 +444,7 @@ __global__ void attn_bwd_preprocess_kernel(const short* __restrict__ dout,
 //   dS = P*(dP-delta)*scale
 //   dQ += dS K         (A=dS via p_lds, B=Kt_lds transposed)
 template <int Dp>
-__global__ __launch_bounds__(ATTN_BWD_THREADS, 4) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -645,7 +645,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS, 4) void attn_bwd_dq_kernel(
 //   dS^T = P^T*(dP^T - delta[q])*scale
 //   dK += dS^T Q       (A=dS^T via p_lds, B=Qt_lds)
 template <int Dp, int ABL = 0>
-__global__ __launch_bounds__(ATTN_BWD_THREADS, 4) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
