@@ -50,6 +50,8 @@ def parse_args():
     p.add_argument("--parallel", choices=["auto", "manual"], default="auto",
                    help="auto = ILP auto-sharding picks (dp, tp)")
     p.add_argument("--seq", type=int, default=1024)
+    p.add_argument("--no-hipgraph", action="store_true",
+                   help="disable hipGraph step capture (single-GPU only)")
     return p.parse_args()
 
 
@@ -145,11 +147,32 @@ def main():
             torch.cuda.synchronize()
 
     for _ in range(args.warmup):
-        step_fn(state, batch)
+        loss = step_fn(state, batch)
+
+    # hipGraph capture of the whole training step (single GPU: no RCCL ops
+    # in the graph).  The BASELINE north star's "HIP streams and graphs"
+    # execution mode: after capture, one replay = fwd+bwd x nmb + fused
+    # AdamW with zero host-side dispatch.
+    graph = None
+    if on_gpu and aa.world_size() == 1 and not args.no_hipgraph:
+        try:
+            torch.cuda.synchronize()
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                loss = step_fn(state, batch)
+            graph.replay()  # one warm replay
+            torch.cuda.synchronize()
+        except Exception as e:
+            print(f"# hipGraph capture unavailable ({e}); eager steps")
+            graph = None
+
     sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        loss = step_fn(state, batch)
+        if graph is not None:
+            graph.replay()
+        else:
+            loss = step_fn(state, batch)
     sync()
     elapsed = time.perf_counter() - t0
     # MAX over ranks
@@ -189,6 +212,7 @@ def main():
                 "num_micro_batches": args.nmb,
                 "parallelism": f"dp{dp}" + (f"tp{tp}" if tp > 1 else ""),
                 "auto_sharded": args.parallel == "auto",
+                "hip_graph": graph is not None,
                 "loss": float(loss),
             },
         }
