@@ -158,6 +158,17 @@ class Block(nn.Module):
         x = x + self.mlp(self.ln2(x))
         return x
 
+    def forward_fused(self, res, delta):
+        """Residual-threaded form: both residual adds fuse into the
+        LayerNorm kernels (ops.add_layer_norm).  Returns the next
+        (residual, delta) pair; the stream value is res + delta."""
+        h, y1 = ops.add_layer_norm(res, delta, self.ln1.weight,
+                                   self.ln1.bias, self.ln1.eps)
+        a = self.attn(y1)
+        h2, y2 = ops.add_layer_norm(h, a, self.ln2.weight, self.ln2.bias,
+                                    self.ln2.eps)
+        return h2, self.mlp(y2)
+
 
 class GPTModel(nn.Module):
     """Full GPT LM. `mesh`/`axis` give the tensor-parallel axis (axis 1 of a
@@ -193,10 +204,12 @@ class GPTModel(nn.Module):
     def forward(self, ids: torch.Tensor) -> torch.Tensor:
         """ids [B, S] -> logits [B, S, V/tp] (vocab-sharded)."""
         B, S = ids.shape
-        x = self.wte(ids) + self.wpe[:S]
+        res = self.wte(ids) + self.wpe[:S]
+        delta = None
         for blk in self.blocks:
-            x = blk(x)
-        x = self.ln_f(x)
+            res, delta = blk.forward_fused(res, delta)
+        _, x = ops.add_layer_norm(res, delta, self.ln_f.weight,
+                                  self.ln_f.bias, self.ln_f.eps)
         return self.lm_head(x)
 
     def loss(self, ids: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
@@ -261,11 +274,14 @@ class GPTStage(nn.Module):
         if self.is_first:
             S = ids.shape[1]
             x = self.wte(ids) + self.wpe[:S]
+        res, delta = x, None
         for blk in self.blocks:
-            x = blk(x)
+            res, delta = blk.forward_fused(res, delta)
         if not self.is_last:
-            return x
-        x = self.ln_f(x)
+            # materialize the stream value at the stage boundary
+            return res + delta if delta is not None else res
+        _, x = ops.add_layer_norm(res, delta, self.ln_f.weight,
+                                  self.ln_f.bias, self.ln_f.eps)
         logits = self.lm_head(x)
         N = logits.shape[0] * logits.shape[1]
         logits = logits.reshape(N, -1)

@@ -22,8 +22,9 @@ from . import reference as ref
 from ._backend import hip_ops, hip_ops_available, use_hip
 
 __all__ = [
-    "layer_norm", "bias_gelu", "flash_attention", "flash_attention_qkv",
-    "softmax_cross_entropy", "fused_adamw", "hip_ops_available",
+    "layer_norm", "add_layer_norm", "bias_gelu", "flash_attention",
+    "flash_attention_qkv", "softmax_cross_entropy", "fused_adamw",
+    "hip_ops_available",
 ]
 
 
@@ -52,6 +53,48 @@ class _LayerNorm(torch.autograd.Function):
 def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
                eps: float = 1e-5) -> torch.Tensor:
     return _LayerNorm.apply(x, weight, bias, eps)
+
+
+class _AddLayerNorm(torch.autograd.Function):
+    """Fused h = a + delta; y = LN(h).  Returns (h, y) — h feeds the
+    residual stream, y the block input.  Saves the separate residual-add
+    HBM pass (XLA fusion analog)."""
+
+    @staticmethod
+    def forward(ctx, a, delta, weight, bias, eps):
+        if use_hip(a):
+            h, y, mean, rstd = hip_ops().add_layer_norm_fwd(
+                a, delta, weight, bias, eps)
+        else:
+            h = a + delta if delta is not None else a
+            y, mean, rstd = ref.layer_norm_fwd(h, weight, bias, eps)
+        ctx.save_for_backward(h, weight, mean, rstd)
+        ctx.has_delta = delta is not None
+        return h, y
+
+    @staticmethod
+    def backward(ctx, dh, dy):
+        h, weight, mean, rstd = ctx.saved_tensors
+        dy = dy.contiguous()
+        dhc = dh.contiguous() if dh is not None else None
+        if use_hip(h):
+            dx, dw, db = hip_ops().add_layer_norm_bwd(dy, dhc, h, weight,
+                                                      mean, rstd)
+        else:
+            dx, dw, db = ref.layer_norm_bwd(dy, h, weight, mean, rstd)
+            if dhc is not None:
+                dx = dx + dhc
+        g2 = dx if ctx.has_delta else None
+        return dx, g2, dw.to(weight.dtype), db.to(weight.dtype), None
+
+
+def add_layer_norm(a: torch.Tensor, delta, weight: torch.Tensor,
+                   bias: torch.Tensor, eps: float = 1e-5):
+    """(h, y) = (a [+ delta], LN(a [+ delta])).  delta may be None."""
+    if delta is None:
+        # no add to fuse: plain LN, h aliases a
+        return a, _LayerNorm.apply(a, weight, bias, eps)
+    return _AddLayerNorm.apply(a, delta, weight, bias, eps)
 
 
 class _BiasGelu(torch.autograd.Function):

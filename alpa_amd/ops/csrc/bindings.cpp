@@ -28,6 +28,14 @@ hipError_t launch_layer_norm_bwd_dwdb(const void*, const void*, const float*,
                                       const float*, float*, float*, float*,
                                       float*, int64_t, int64_t, int,
                                       hipStream_t);
+hipError_t launch_add_layer_norm_fwd(const void*, const void*, const void*,
+                                     const void*, void*, void*, float*,
+                                     float*, int64_t, int64_t, float,
+                                     hipStream_t);
+hipError_t launch_add_layer_norm_bwd_dx(const void*, const void*,
+                                        const void*, const void*,
+                                        const float*, const float*, void*,
+                                        int64_t, int64_t, hipStream_t);
 hipError_t launch_bias_gelu_fwd(const void*, const void*, void*, int64_t,
                                 int64_t, hipStream_t);
 hipError_t launch_bias_gelu_bwd(const void*, const void*, const void*, void*,
@@ -125,6 +133,62 @@ std::vector<at::Tensor> layer_norm_bwd(const at::Tensor& dy,
       (const float*)rstd.const_data_ptr(), (float*)dw_part.mutable_data_ptr(),
       (float*)db_part.mutable_data_ptr(), (float*)dw.mutable_data_ptr(),
       (float*)db.mutable_data_ptr(), N, H, kStripes, cur_stream()));
+  at::sum_out(dw, dw_part, {0});
+  at::sum_out(db, db_part, {0});
+  return {dx, dw, db};
+}
+
+// --------------------------- Add + LayerNorm -----------------------------
+
+std::vector<at::Tensor> add_layer_norm_fwd(const at::Tensor& a,
+                                           const c10::optional<at::Tensor>& b,
+                                           const at::Tensor& w,
+                                           const at::Tensor& bias,
+                                           double eps) {
+  check_bf16_contig(a, "a");
+  int64_t H = a.size(-1);
+  int64_t N = a.numel() / H;
+  auto h = at::empty_like(a);
+  auto y = at::empty_like(a);
+  auto f32 = a.options().dtype(at::kFloat);
+  auto mean = at::empty({N}, f32);
+  auto rstd = at::empty({N}, f32);
+  const void* bp = b.has_value() ? b->const_data_ptr() : nullptr;
+  HIP_OK(launch_add_layer_norm_fwd(
+      a.const_data_ptr(), bp, w.const_data_ptr(), bias.const_data_ptr(),
+      h.mutable_data_ptr(), y.mutable_data_ptr(),
+      (float*)mean.mutable_data_ptr(), (float*)rstd.mutable_data_ptr(), N,
+      H, (float)eps, cur_stream()));
+  return {h, y, mean, rstd};
+}
+
+std::vector<at::Tensor> add_layer_norm_bwd(
+    const at::Tensor& dy, const c10::optional<at::Tensor>& dh,
+    const at::Tensor& h, const at::Tensor& w, const at::Tensor& mean,
+    const at::Tensor& rstd) {
+  check_bf16_contig(dy, "dy");
+  int64_t H = h.size(-1);
+  int64_t N = h.numel() / H;
+  auto dx = at::empty_like(h);
+  auto f32 = h.options().dtype(at::kFloat);
+  auto dw = at::empty({H}, f32);
+  auto db = at::empty({H}, f32);
+  const int kStripes = stripes_for(N, H);
+  auto dw_part = at::empty({kStripes, H}, f32);
+  auto db_part = at::empty({kStripes, H}, f32);
+  const void* dhp = dh.has_value() ? dh->const_data_ptr() : nullptr;
+  HIP_OK(launch_add_layer_norm_bwd_dx(
+      dy.const_data_ptr(), dhp, h.const_data_ptr(), w.const_data_ptr(),
+      (const float*)mean.const_data_ptr(),
+      (const float*)rstd.const_data_ptr(), dx.mutable_data_ptr(), N, H,
+      cur_stream()));
+  HIP_OK(launch_layer_norm_bwd_dwdb(
+      dy.const_data_ptr(), h.const_data_ptr(),
+      (const float*)mean.const_data_ptr(),
+      (const float*)rstd.const_data_ptr(),
+      (float*)dw_part.mutable_data_ptr(), (float*)db_part.mutable_data_ptr(),
+      (float*)dw.mutable_data_ptr(), (float*)db.mutable_data_ptr(), N, H,
+      kStripes, cur_stream()));
   at::sum_out(dw, dw_part, {0});
   at::sum_out(db, db_part, {0});
   return {dx, dw, db};
@@ -396,6 +460,10 @@ at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layer_norm_fwd", &layer_norm_fwd, "LayerNorm forward (gfx950)");
   m.def("layer_norm_bwd", &layer_norm_bwd, "LayerNorm backward (gfx950)");
+  m.def("add_layer_norm_fwd", &add_layer_norm_fwd,
+        "fused residual-add + LayerNorm fwd (gfx950)");
+  m.def("add_layer_norm_bwd", &add_layer_norm_bwd,
+        "fused residual-add + LayerNorm bwd (gfx950)");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "bias+GeLU forward (gfx950)");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "bias+GeLU backward (gfx950)");
   m.def("adamw_step_raw", &adamw_step_raw, "multi-tensor AdamW (gfx950)");

@@ -184,3 +184,146 @@ hipError_t launch_layer_norm_bwd_dwdb(const void* dy, const void* x,
 }
 
 }  // extern "C"
+
+// ------------------------------------------------------------------
+// Fused residual-add + LayerNorm: h = a + b; y = LN(h) * w + bias.
+// Saves the separate elementwise-add pass over the residual stream
+// (the reference gets this from XLA fusion).  b may be null (plain LN
+// with h written = a, used at the embedding boundary).
+__global__ void add_layer_norm_fwd_kernel(const short* __restrict__ a,
+                                          const short* __restrict__ b,
+                                          const short* __restrict__ w,
+                                          const short* __restrict__ bias,
+                                          short* __restrict__ h,
+                                          short* __restrict__ y,
+                                          float* __restrict__ mean_out,
+                                          float* __restrict__ rstd_out,
+                                          int H, float eps) {
+  const int row = blockIdx.x;
+  const short* ar = a + (int64_t)row * H;
+  const short* br = (b == nullptr) ? nullptr : b + (int64_t)row * H;
+  short* hr = h + (int64_t)row * H;
+  short* yr = y + (int64_t)row * H;
+  __shared__ float red[LN_BLOCK / 64];
+
+  float s = 0.f, ss = 0.f;
+  for (int i = threadIdx.x * 8; i < H; i += LN_BLOCK * 8) {
+    bf16x8 av = *reinterpret_cast<const bf16x8*>(ar + i);
+    bf16x8 hv;
+    if (br != nullptr) {
+      bf16x8 bv = *reinterpret_cast<const bf16x8*>(br + i);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(av[j]) + bf2f(bv[j]);
+        hv[j] = f2bf(f);
+        s += f;
+        ss += f * f;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = bf2f(av[j]);
+        hv[j] = av[j];
+        s += f;
+        ss += f * f;
+      }
+    }
+    *reinterpret_cast<bf16x8*>(hr + i) = hv;
+  }
+  s = block_reduce_sum(s, red);
+  ss = block_reduce_sum(ss, red);
+  const float mean = s / H;
+  const float rstd = rsqrtf(ss / H - mean * mean + eps);
+  if (threadIdx.x == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  for (int i = threadIdx.x * 8; i < H; i += LN_BLOCK * 8) {
+    bf16x8 hv = *reinterpret_cast<const bf16x8*>(hr + i);
+    bf16x8 wv = *reinterpret_cast<const bf16x8*>(w + i);
+    bf16x8 bv2 = *reinterpret_cast<const bf16x8*>(bias + i);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float xhat = (bf2f(hv[j]) - mean) * rstd;
+      o[j] = f2bf(xhat * bf2f(wv[j]) + bf2f(bv2[j]));
+    }
+    *reinterpret_cast<bf16x8*>(yr + i) = o;
+  }
+}
+
+// backward dx with an extra residual-gradient term:
+// g = LN_bwd_dx(dy) + dh   (dh may be null)
+__global__ void add_layer_norm_bwd_dx_kernel(
+    const short* __restrict__ dy, const short* __restrict__ dh,
+    const short* __restrict__ x, const short* __restrict__ w,
+    const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
+    short* __restrict__ dx, int H) {
+  const int row = blockIdx.x;
+  const short* dyr = dy + (int64_t)row * H;
+  const short* dhr = (dh == nullptr) ? nullptr : dh + (int64_t)row * H;
+  const short* xr = x + (int64_t)row * H;
+  short* dxr = dx + (int64_t)row * H;
+  const float mean = mean_in[row], rstd = rstd_in[row];
+  __shared__ float red[LN_BLOCK / 64];
+
+  float c1 = 0.f, c2 = 0.f;
+  for (int i = threadIdx.x * 8; i < H; i += LN_BLOCK * 8) {
+    bf16x8 dv = *reinterpret_cast<const bf16x8*>(dyr + i);
+    bf16x8 xv = *reinterpret_cast<const bf16x8*>(xr + i);
+    bf16x8 wv = *reinterpret_cast<const bf16x8*>(w + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float wdy = bf2f(dv[j]) * bf2f(wv[j]);
+      float xhat = (bf2f(xv[j]) - mean) * rstd;
+      c1 += wdy;
+      c2 += wdy * xhat;
+    }
+  }
+  c1 = block_reduce_sum(c1, red) / H;
+  c2 = block_reduce_sum(c2, red) / H;
+
+  for (int i = threadIdx.x * 8; i < H; i += LN_BLOCK * 8) {
+    bf16x8 dv = *reinterpret_cast<const bf16x8*>(dyr + i);
+    bf16x8 xv = *reinterpret_cast<const bf16x8*>(xr + i);
+    bf16x8 wv = *reinterpret_cast<const bf16x8*>(w + i);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float wdy = bf2f(dv[j]) * bf2f(wv[j]);
+      float xhat = (bf2f(xv[j]) - mean) * rstd;
+      float g = (wdy - c1 - xhat * c2) * rstd;
+      if (dhr != nullptr) g += bf2f(dhr[i + j]);
+      o[j] = f2bf(g);
+    }
+    *reinterpret_cast<bf16x8*>(dxr + i) = o;
+  }
+}
+
+extern "C" {
+
+hipError_t launch_add_layer_norm_fwd(const void* a, const void* b,
+                                     const void* w, const void* bias,
+                                     void* h, void* y, float* mean,
+                                     float* rstd, int64_t N, int64_t H,
+                                     float eps, hipStream_t stream) {
+  add_layer_norm_fwd_kernel<<<dim3((uint32_t)N), dim3(LN_BLOCK), 0,
+                              stream>>>(
+      (const short*)a, (const short*)b, (const short*)w, (const short*)bias,
+      (short*)h, (short*)y, mean, rstd, (int)H, eps);
+  return hipGetLastError();
+}
+
+hipError_t launch_add_layer_norm_bwd_dx(const void* dy, const void* dh,
+                                        const void* x, const void* w,
+                                        const float* mean, const float* rstd,
+                                        void* dx, int64_t N, int64_t H,
+                                        hipStream_t stream) {
+  add_layer_norm_bwd_dx_kernel<<<dim3((uint32_t)N), dim3(LN_BLOCK), 0,
+                                 stream>>>(
+      (const short*)dy, (const short*)dh, (const short*)x, (const short*)w,
+      mean, rstd, (short*)dx, (int)H);
+  return hipGetLastError();
+}
+
+}  // extern "C"

@@ -246,3 +246,30 @@ def test_moe_gpt_step_on_gpu():
     losses = [float(step(state, (ids, labels))) for _ in range(6)]
     assert all(l == l for l in losses), f"NaN in {losses}"
     assert losses[-1] < losses[0], losses
+
+
+def test_add_layer_norm_fused(ext):
+    """Fused residual-add + LN vs reference."""
+    torch.manual_seed(12)
+    N, H = 512, 2560
+    a = torch.randn(N, H, device="cuda", dtype=torch.bfloat16)
+    b = torch.randn(N, H, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    bias = torch.randn(H, device="cuda", dtype=torch.bfloat16)
+    h, y, mean, rstd = ext.add_layer_norm_fwd(a, b, w, bias, 1e-5)
+    h_ref = (a.float() + b.float())
+    torch.testing.assert_close(h.float(), h_ref, rtol=2e-2, atol=2e-2)
+    y_ref, mean_ref, rstd_ref = ref.layer_norm_fwd(h.float(), w.float(),
+                                                   bias.float(), 1e-5)
+    torch.testing.assert_close(y.float(), y_ref, rtol=2e-2, atol=2e-2)
+
+    dy = torch.randn_like(y)
+    dh = torch.randn_like(y)
+    dx, dw, db = ext.add_layer_norm_bwd(dy, dh, h, w, mean, rstd)
+    dx_ref, dw_ref, db_ref = ref.layer_norm_bwd(dy.float(), h.float(),
+                                                w.float(), mean_ref,
+                                                rstd_ref)
+    torch.testing.assert_close(dx.float(), dx_ref + dh.float(), rtol=3e-2,
+                               atol=3e-2)
+    torch.testing.assert_close(dw, dw_ref, rtol=2e-2, atol=2e-1)
+    torch.testing.assert_close(db, db_ref, rtol=2e-2, atol=2e-1)
