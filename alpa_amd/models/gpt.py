@@ -262,12 +262,15 @@ class GPTStage(nn.Module):
         if is_last:
             self.ln_f = LayerNorm(cfg.hidden_size, cfg.layernorm_eps, dtype,
                                   device)
-            self.lm_head = ColumnParallelLinear(cfg.hidden_size,
-                                                cfg.vocab_size, mesh, axis,
-                                                bias=False, dtype=dtype,
-                                                device=device,
-                                                init_seed=init_seed,
-                                                init_tag="lm_head")
+            # tied embeddings: the LM head draws the SAME init as wte
+            # (tag + std); the cross-stage grad all-reduce keeps the two
+            # copies synchronized (reference __builtin$CrossMeshAllReduce,
+            # SURVEY.md §2.3 N15)
+            self.lm_head = ColumnParallelLinear(
+                cfg.hidden_size, cfg.vocab_size, mesh, axis, bias=False,
+                dtype=dtype, device=device, init_seed=init_seed,
+                init_tag="wte" if cfg.tie_embeddings else "lm_head",
+                init_std=0.02 if cfg.tie_embeddings else None)
 
     def forward(self, x, microbatch):
         ids, labels = microbatch["ids"], microbatch.get("labels")
@@ -307,5 +310,9 @@ def gpt_pipeline_spec(cfg: GPTConfig):
 
     # per-block cost uniform; embedding/LM head are pinned to the
     # first/last stage by construction
-    return PipelineModelSpec(num_layers=cfg.num_layers,
+    spec = PipelineModelSpec(num_layers=cfg.num_layers,
                              build_stage=build_stage, act_shape=act_shape)
+    if cfg.tie_embeddings:
+        # stage -1 = last; resolved by the pipeline compiler
+        spec.tied_groups = [{0: "wte.weight", -1: "lm_head.weight"}]
+    return spec

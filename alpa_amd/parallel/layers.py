@@ -114,7 +114,7 @@ class ColumnParallelLinear(nn.Module):
                  mesh: Optional[DeviceMesh] = None, axis: int = 1,
                  bias: bool = True, gelu: bool = False,
                  dtype=torch.float32, device=None, init_seed=None,
-                 init_tag: str = None):
+                 init_tag: str = None, init_std: float = None):
         super().__init__()
         self.mesh, self.axis = mesh, axis
         tp = mesh.axis_size(axis) if mesh is not None else 1
@@ -128,8 +128,9 @@ class ColumnParallelLinear(nn.Module):
                         device=device))
         idx = mesh.axis_index(axis) if (mesh is not None and mesh.is_member) else 0
         seed = tag_seed(init_seed, init_tag) if init_tag is not None else None
+        std = init_std if init_std is not None else 1.0 / math.sqrt(in_features)
         _sharded_normal_(self.weight, (out_features, in_features), 0,
-                         max(idx, 0), tp, 1.0 / math.sqrt(in_features), seed)
+                         max(idx, 0), tp, std, seed)
         if bias:
             self.bias = nn.Parameter(
                 torch.zeros(self.out_per_rank, dtype=dtype, device=device))

@@ -75,6 +75,40 @@ def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
                                     mesh=my_mesh, axis=method.tp_axis,
                                     dtype=dtype, device=device())
 
+    from ..mesh import _get_group
+    tied_comms = []
+
+    def _lookup(module, path):
+        obj = module
+        for attr in path.split("."):
+            obj = getattr(obj, attr)
+        return obj
+
+    for tg in (spec.tied_groups or []):
+        pairs = [((P + s) % P, path) for s, path in tg.items()]
+        uniq_stages = sorted({s for s, _ in pairs})
+        if len(uniq_stages) < 2:
+            # all copies live on one stage: alias the parameters directly
+            if my_stage == uniq_stages[0]:
+                first = _lookup(stage_module, pairs[0][1])
+                for _, path in pairs[1:]:
+                    parent_path, _, leaf = path.rpartition(".")
+                    parent = _lookup(stage_module, parent_path) \
+                        if parent_path else stage_module
+                    setattr(parent, leaf, first)
+            continue
+        coords = [(i, j) for i in range(stage_shape[0])
+                  for j in range(stage_shape[1])]
+        for (ci, cj) in coords:
+            ranks = tuple(int(stage_meshes[si].grid[ci, cj])
+                          for si in uniq_stages)
+            g = _get_group(ranks)
+            if my_rank in ranks and my_stage in uniq_stages:
+                for s, path in pairs:
+                    if s == my_stage:
+                        tied_comms.append(
+                            (g, _lookup(stage_module, path)))
+
     gs = GradSynchronizer(list(stage_module.parameters()), my_mesh,
                           axis=method.dp_axis)
     opt = AdamW(stage_module.parameters(), lr=lr, betas=betas,
@@ -88,4 +122,9 @@ def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
     engine._act_shape_fn = spec.act_shape
     # rank that holds the authoritative loss (first rank of last stage)
     engine.loss_src_rank = int(stage_meshes[P - 1].ranks[0])
+
+    # tied-weight cross-stage allreduce groups (reference N15): one group
+    # per (tied set x mesh coordinate); every rank creates every group
+    # (collective), members keep (group, local param) pairs
+    engine.tied_comms = tied_comms
     return stage_module, engine, opt, gs, stage_meshes

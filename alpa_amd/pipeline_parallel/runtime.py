@@ -53,6 +53,7 @@ class PipelineEngine:
         self.device = next(stage_module.parameters()).device
         self._act_shape_fn = None  # lazy: microbatch -> act shape
         self.loss_src_rank = 0
+        self.tied_comms = []  # [(process_group, local param)]
 
     # ------------------------- p2p primitives -------------------------
     def _p2p(self, ops: List[dist.P2POp]):
@@ -185,9 +186,19 @@ class PipelineEngine:
 
         if self.grad_sync is not None:
             self.grad_sync.finish()
+        self._sync_tied_grads()
         if losses:
             return torch.stack(losses).mean()
         return torch.zeros((), device=self.device)
+
+    def _sync_tied_grads(self):
+        """Cross-stage all-reduce of tied-weight grads (reference
+        cross-mesh allreduce for shared embeddings)."""
+        if not self.tied_comms or not is_distributed():
+            return
+        for g, p in self.tied_comms:
+            if p.grad is not None:
+                dist.all_reduce(p.grad, group=g)
 
     def _gpipe_step(self, microbatches) -> torch.Tensor:
         M = self.M
@@ -211,6 +222,7 @@ class PipelineEngine:
         pending.clear()
         if self.grad_sync is not None:
             self.grad_sync.finish()
+        self._sync_tied_grads()
         if losses:
             return torch.stack(losses).mean()
         return torch.zeros((), device=self.device)

@@ -33,3 +33,8 @@ class PipelineModelSpec:
     build_stage: Callable[..., torch.nn.Module]
     act_shape: Callable[[Any], Tuple[int, ...]]
     layer_costs: Optional[Sequence[float]] = None
+    #: tied parameters across stages: each dict maps stage index (-1 =
+    #: last) -> param path in that stage's module; grads are all-reduced
+    #: across the same-coordinate ranks of those stages each step
+    #: (reference cross-mesh allreduce for tied embeddings, N15)
+    tied_groups: Optional[list] = None

@@ -163,3 +163,45 @@ def test_overlap_friendly_schedule_invariants():
         assert fwd == list(range(8)) and bwd == list(range(8))
         # one extra in-flight forward vs plain 1F1B
         assert schedules.peak_live_activations(instrs) <= min(4 - s + 1, 8)
+
+
+CFG_TIED = GPTConfig(hidden_size=64, num_layers=4, num_heads=4, seq_len=32,
+                     vocab_size=96, tie_embeddings=True)
+
+
+def _tied_stage_builder(layer_range, is_first, is_last, mesh, axis, dtype,
+                        device):
+    from alpa_amd.models.gpt import GPTStage
+    return GPTStage(CFG_TIED, layer_range, is_first, is_last, mesh, axis,
+                    dtype, device, init_seed=11)
+
+
+def _pp_tied_worker(rank, world_size):
+    method = aa.PipeshardParallel(num_micro_batches=2,
+                                  num_stages=world_size,
+                                  stage_mesh_shape=(1, 1))
+    spec = gpt_pipeline_spec(CFG_TIED)
+    spec.build_stage = _tied_stage_builder
+    state = aa.TrainState.create(spec, method, lr=1e-3)
+    step = aa.parallelize(lambda m, b: None, method=method)
+    return [float(step(state, make_batch(i))) for i in range(STEPS)]
+
+
+def test_tied_embeddings_pipeline_matches_serial():
+    """Tied wte/lm_head across pipeline stages: the cross-stage grad
+    all-reduce (reference N15) must reproduce the serial shared-parameter
+    training trajectory."""
+    method = aa.ShardParallel(num_micro_batches=2,
+                              logical_mesh_shape=(1, 1))
+    state = aa.TrainState.create(
+        lambda mesh=None, axis=1, dtype=torch.float32, device=None:
+        GPTModel(CFG_TIED, mesh, axis, dtype, device, init_seed=11),
+        method, lr=1e-3)
+    step = aa.parallelize(
+        lambda m, b: m.loss(b["ids"], b["labels"]), method=method)
+    serial = [float(step(state, make_batch(i))) for i in range(STEPS)]
+
+    results = run_distributed(_pp_tied_worker, world_size=2, timeout=300)
+    for r in results:
+        for a, b in zip(r, serial):
+            assert abs(a - b) < 3e-4, (r, serial)
