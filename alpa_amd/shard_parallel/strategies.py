@@ -12,11 +12,40 @@ from __future__ import annotations
 
 from typing import List, Optional
 
+import os
+
 from .ir import MeshModel, Node, Spec, Strategy, REPLICATED
 
 # effective per-device matmul throughput used to convert FLOPs -> seconds;
-# calibrated from measured hipBLASLt bf16 rates on MI355X (profiles/)
+# fallback when no profiled database exists (measured hipBLASLt bf16 rate
+# on MI355X, profiles/)
 MATMUL_TFLOPS = 1.2e15
+
+_CALIBRATED: float = None
+
+
+def effective_matmul_flops() -> float:
+    """Measured matmul rate from the profiled cost DB when present
+    (tools/calibrate.py -> prof_database.pkl), else the fallback constant —
+    the "cost model recalibrated from measured counters" loop."""
+    global _CALIBRATED
+    if _CALIBRATED is None:
+        _CALIBRATED = MATMUL_TFLOPS
+        from ..global_env import global_config
+        path = global_config.prof_database_path
+        if os.path.exists(path):
+            try:
+                from ..mesh_profiling import ProfilingResultDatabase
+                db = ProfilingResultDatabase()
+                db.load(path)
+                for (key, shape), r in db.data.items():
+                    if "matmul_bf16" in r.op_curves:
+                        c = r.op_curves["matmul_bf16"]
+                        _CALIBRATED = c.sizes[-1] / c.times[-1]
+                        break
+            except Exception:
+                pass
+    return _CALIBRATED
 # grad multiplier: fwd + bwd matmuls (dX, dW)
 TRAIN_FACTOR = 3.0
 # Gradient all-reduce is overlapped with backward compute by the bucketed
@@ -65,7 +94,7 @@ def matmul_strategies(mesh: MeshModel, tokens: int, k: int, n: int,
                 name=f"b{b_ax}_col{w_ax}",
                 in_specs=[(b_ax, None)],
                 out_spec=(b_ax, w_ax),
-                compute_cost=flops / d / MATMUL_TFLOPS,
+                compute_cost=flops / d / effective_matmul_flops(),
                 comm_cost=gsync + dx_ar,
                 memory=w_state / mesh.axis_size(w_ax) + y_bytes / d))
         # row-parallel: X feature-split on ax, W k-split on ax, out
@@ -86,7 +115,7 @@ def matmul_strategies(mesh: MeshModel, tokens: int, k: int, n: int,
                 name=f"b{b_ax}_row{r_ax}",
                 in_specs=[(b_ax, r_ax)],
                 out_spec=(b_ax, None),
-                compute_cost=flops / d / MATMUL_TFLOPS,
+                compute_cost=flops / d / effective_matmul_flops(),
                 comm_cost=cost_ar + gsync,
                 memory=w_state / mesh.axis_size(r_ax) +
                 y_bytes / dev(b_ax, None)))

@@ -106,3 +106,29 @@ def test_solver_consistency_objective():
             st = n.strategies[res.choices[i]]
             node_cost += st.compute_cost + st.comm_cost
     assert node_cost <= res.objective + 1e-9
+
+
+def test_calibrated_matmul_rate(tmp_path, monkeypatch):
+    """The ILP's compute-cost rate comes from the profiled DB when present
+    (the measured-counter recalibration loop)."""
+    from alpa_amd.global_env import global_config
+    from alpa_amd.mesh_profiling import (CostCurve, MeshProfilingResult,
+                                         ProfilingResultDatabase)
+    from alpa_amd.shard_parallel import strategies
+
+    db = ProfilingResultDatabase()
+    r = MeshProfilingResult((1, 1))
+    c = CostCurve()
+    c.add(1e12, 1e12 / 0.9e15)  # 0.9 PF measured
+    r.op_curves["matmul_bf16"] = c
+    db.update_one_mesh("test", (1, 1), r)
+    path = tmp_path / "db.pkl"
+    db.save(str(path))
+
+    monkeypatch.setattr(global_config, "prof_database_path", str(path))
+    strategies._CALIBRATED = None
+    try:
+        rate = strategies.effective_matmul_flops()
+        assert abs(rate - 0.9e15) / 0.9e15 < 1e-6
+    finally:
+        strategies._CALIBRATED = None
