@@ -31,9 +31,12 @@ class _AllToAll(torch.autograd.Function):
         ctx.mesh, ctx.axis = mesh, axis
         if mesh is None or mesh.axis_size(axis) == 1 or not is_distributed():
             return x.clone()
-        out = torch.empty_like(x)
-        dist.all_to_all_single(out, x.contiguous(),
-                               group=mesh.axis_group(axis))
+        # NB: the output buffer must be truly contiguous — empty_like on a
+        # non-contiguous input (e.g. a permute's grad) inherits its strides
+        # and the collective would scatter bytes into the wrong layout
+        xin = x.contiguous()
+        out = torch.empty_like(xin)
+        dist.all_to_all_single(out, xin, group=mesh.axis_group(axis))
         return out
 
     @staticmethod
@@ -41,9 +44,9 @@ class _AllToAll(torch.autograd.Function):
         mesh, axis = ctx.mesh, ctx.axis
         if mesh is None or mesh.axis_size(axis) == 1 or not is_distributed():
             return g, None, None
-        out = torch.empty_like(g)
-        dist.all_to_all_single(out, g.contiguous(),
-                               group=mesh.axis_group(axis))
+        gin = g.contiguous()
+        out = torch.empty_like(gin)
+        dist.all_to_all_single(out, gin, group=mesh.axis_group(axis))
         return out, None, None
 
 
