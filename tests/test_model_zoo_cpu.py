@@ -80,3 +80,14 @@ def test_follow_parallel_inference():
     assert not loss.requires_grad
     fm = FollowParallel(train_method=method)
     assert fm.resolve_mesh().shape == (1, 1)
+
+
+def test_conformer_trains():
+    from alpa_amd.models.conformer import ConformerConfig, ConformerEncoder
+    cfg = ConformerConfig(hidden_size=64, num_layers=2, num_heads=4,
+                          conv_kernel=7)
+    m = ConformerEncoder(cfg, input_dim=40)
+    x = torch.randn(2, 32, 40)
+    y = m(x)
+    assert y.shape == (2, 32, 64)
+    y.square().mean().backward()
