@@ -32,6 +32,7 @@ class _LayerNorm(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, weight, bias, eps):
+        x = x.contiguous()
         if use_hip(x):
             y, mean, rstd = hip_ops().layer_norm_fwd(x, weight, bias, eps)
         else:
@@ -62,6 +63,8 @@ class _AddLayerNorm(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, a, delta, weight, bias, eps):
+        a = a.contiguous()
+        delta = delta.contiguous() if delta is not None else None
         if use_hip(a):
             h, y, mean, rstd = hip_ops().add_layer_norm_fwd(
                 a, delta, weight, bias, eps)
@@ -101,6 +104,7 @@ class _BiasGelu(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, bias):
+        x = x.contiguous()
         ctx.save_for_backward(x, bias)
         if use_hip(x):
             return hip_ops().bias_gelu_fwd(x, bias)
@@ -234,6 +238,7 @@ class _SoftmaxCrossEntropy(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, logits, targets):
+        logits = logits.contiguous()
         if use_hip(logits):
             loss, lse = hip_ops().cross_entropy_fwd(logits, targets)
         else:
