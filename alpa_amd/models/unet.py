@@ -44,12 +44,14 @@ class ResBlock(nn.Module):
 class SelfAttention2d(nn.Module):
     """Spatial self-attention via the flash kernel (HW x HW)."""
 
-    def __init__(self, ch, heads=4):
+    def __init__(self, ch, heads=None):
         super().__init__()
         self.norm = nn.GroupNorm(8, ch)
         self.qkv = nn.Linear(ch, 3 * ch)
         self.proj = nn.Linear(ch, ch)
-        self.heads = heads
+        # flash kernel wants head_dim multiple of 16
+        self.heads = heads if heads is not None else max(1, ch // 32)
+        assert (ch // self.heads) % 16 == 0, (ch, self.heads)
 
     def forward(self, x):
         B, C, H, W = x.shape
