@@ -961,6 +961,18 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
         (const short*)dout, (const short*)o, delta_ws, rows, (int)H, (int)S,
         (int)D, st.dob, st.doh, st.dos, p[21], p[22], p[23]);
   }
+  // dq and dkv are independent (disjoint outputs, shared read-only
+  // inputs) and each runs at ~3 waves/SIMD — launching them on two
+  // streams lets the CU scheduler co-resident them and fill the SIMDs.
+  static hipStream_t side_stream = nullptr;
+  static hipEvent_t ev_fork = nullptr, ev_join = nullptr;
+  if (side_stream == nullptr) {
+    hipStreamCreateWithFlags(&side_stream, hipStreamNonBlocking);
+    hipEventCreateWithFlags(&ev_fork, hipEventDisableTiming);
+    hipEventCreateWithFlags(&ev_join, hipEventDisableTiming);
+  }
+  hipEventRecord(ev_fork, stream);
+  hipStreamWaitEvent(side_stream, ev_fork, 0);
   dim3 block(ATTN_BWD_THREADS);
   dim3 grid_q((uint32_t)ceil_div(S, 16 * (ATTN_BWD_THREADS / 64)),
               (uint32_t)(B * H));
@@ -968,7 +980,7 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
                (uint32_t)(B * H));
 #define LAUNCH_BWD(DP)                                                       \
   do {                                                                       \
-    attn_bwd_dq_kernel<DP><<<grid_q, block, 0, stream>>>(                    \
+    attn_bwd_dq_kernel<DP><<<grid_q, block, 0, side_stream>>>(               \
         (const short*)q, (const short*)k, (const short*)v,                   \
         (const short*)dout, lse, delta_ws, (short*)dq, (int)H, (int)S,       \
         (int)Skv, (int)D, scale, causal, st);                                \
@@ -987,6 +999,8 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
     return hipErrorInvalidValue;
   }
 #undef LAUNCH_BWD
+  hipEventRecord(ev_join, side_stream);
+  hipStreamWaitEvent(stream, ev_join, 0);
   return hipGetLastError();
 }
 
