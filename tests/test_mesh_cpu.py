@@ -1,0 +1,50 @@
+"""DeviceMesh axis-group collectives on a 2x2 mesh over gloo ws=4
+(reference: collective plumbing tests without GPUs)."""
+import torch
+
+from dist_utils import run_distributed
+
+import alpa_amd as aa
+
+
+def _mesh22_worker(rank, world_size):
+    mesh = aa.full_mesh((2, 2))
+    # axis 1 (rows): ranks {0,1} and {2,3}; axis 0 (cols): {0,2}, {1,3}
+    t = torch.tensor([float(rank)])
+    row = t.clone()
+    mesh.all_reduce(row, axis=1)
+    col = t.clone()
+    mesh.all_reduce(col, axis=0)
+    full = t.clone()
+    mesh.all_reduce(full)
+    # all-gather along axis 1
+    out = torch.zeros(2)
+    mesh.all_gather(out, t, axis=1)
+    return {"row": row, "col": col, "full": full, "gather": out,
+            "coord": torch.tensor(mesh.coord)}
+
+
+def test_mesh_2x2_axis_collectives():
+    res = run_distributed(_mesh22_worker, world_size=4)
+    # row sums: {0+1, 2+3}; col sums: {0+2, 1+3}
+    expect_row = [1.0, 1.0, 5.0, 5.0]
+    expect_col = [2.0, 4.0, 2.0, 4.0]
+    for r in range(4):
+        assert float(res[r]["row"]) == expect_row[r]
+        assert float(res[r]["col"]) == expect_col[r]
+        assert float(res[r]["full"]) == 6.0
+    assert res[0]["gather"].tolist() == [0.0, 1.0]
+    assert res[3]["gather"].tolist() == [2.0, 3.0]
+    assert res[2]["coord"].tolist() == [1, 0]
+
+
+def _bcast_worker(rank, world_size):
+    mesh = aa.full_mesh((1, world_size))
+    t = torch.tensor([float(rank) + 1])
+    mesh.broadcast(t, src_coord=1, axis=1)
+    return t
+
+
+def test_mesh_broadcast():
+    res = run_distributed(_bcast_worker, world_size=2)
+    assert all(float(r) == 2.0 for r in res)
