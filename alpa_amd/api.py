@@ -223,7 +223,7 @@ class ParallelizedFunc:
         m = self.method
         engine = state.engine
         loss = engine.train_step(micro)
-        dp = engine.mesh.axis_size(m.dp_axis)
+        dp = getattr(engine, "loss_dp", engine.mesh.axis_size(m.dp_axis))
         scale = 1.0 / (m.num_micro_batches * (dp if dp > 1 else 1))
         state.optimizer.step(grad_scale=scale)
         state.step_count += 1

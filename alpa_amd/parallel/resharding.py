@@ -143,43 +143,53 @@ def _slice(t: torch.Tensor, idx: Index) -> torch.Tensor:
     return t[tuple(slice(lo, hi) for lo, hi in idx)]
 
 
-def execute_resharding(spec: ReshardingTaskSpec, local_src: Optional[torch.Tensor],
-                       dst_buf: Optional[torch.Tensor]) -> Optional[torch.Tensor]:
-    """Run this rank's part of the exchange.
-
-    local_src: this rank's src tile (or None if it owns none);
-    dst_buf: preallocated dst tile (or None).  Local overlaps are copies;
-    remote ones are batched isend/irecv (one batch per peer pair —
-    both sides post together, so the exchange cannot deadlock).
-    """
+def prepare_resharding(spec: ReshardingTaskSpec,
+                       local_src: Optional[torch.Tensor],
+                       dst_buf: Optional[torch.Tensor]):
+    """This rank's part of the exchange, split for composability: performs
+    the local copies now and returns (p2p_ops, fixups); the caller batches
+    the ops (possibly merged with another exchange) in one
+    batch_isend_irecv and then applies the fixups."""
     me = rank()
-    # local copies first
     for t in spec.transfers:
         if t.src_rank == me and t.dst_rank == me:
             _slice(dst_buf, t.dst_offset).copy_(_slice(local_src,
                                                        t.src_offset))
+    ops: List[dist.P2POp] = []
+    fixups = []
     if not is_distributed():
-        return dst_buf
-    ops = []
-    staged = []
+        return ops, fixups
     for t in spec.transfers:
         if t.src_rank == t.dst_rank:
             continue
         if t.src_rank == me:
             payload = _slice(local_src, t.src_offset).contiguous()
-            staged.append(payload)
+            fixups.append((payload, None))  # keep alive until waited
             ops.append(dist.P2POp(dist.isend, payload, t.dst_rank))
         elif t.dst_rank == me:
             shape = tuple(hi - lo for lo, hi in t.region)
             buf = torch.empty(shape, dtype=dst_buf.dtype,
                               device=dst_buf.device)
-            staged.append((buf, t))
+            fixups.append((buf, (dst_buf, t)))
             ops.append(dist.P2POp(dist.irecv, buf, t.src_rank))
+    return ops, fixups
+
+
+def apply_fixups(fixups):
+    for buf, tgt in fixups:
+        if tgt is not None:
+            dst_buf, t = tgt
+            _slice(dst_buf, t.dst_offset).copy_(buf)
+
+
+def execute_resharding(spec: ReshardingTaskSpec,
+                       local_src: Optional[torch.Tensor],
+                       dst_buf: Optional[torch.Tensor]
+                       ) -> Optional[torch.Tensor]:
+    """Run this rank's part of the exchange (see prepare_resharding)."""
+    ops, fixups = prepare_resharding(spec, local_src, dst_buf)
     if ops:
         for w in dist.batch_isend_irecv(ops):
             w.wait()
-    for item in staged:
-        if isinstance(item, tuple):
-            buf, t = item
-            _slice(dst_buf, t.dst_offset).copy_(buf)
+    apply_fixups(fixups)
     return dst_buf

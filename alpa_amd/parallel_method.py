@@ -126,6 +126,11 @@ class PipeshardParallel(ParallelMethod):
     num_stages: Optional[int] = None
     #: per-stage (dp, tp) logical shape; None -> auto
     stage_mesh_shape: Optional[Tuple[int, int]] = None
+    #: HETEROGENEOUS stage meshes: one (dp, tp) per stage, sizes may
+    #: differ (the reference's auto-search emits exactly this, e.g.
+    #: submeshes (1,2),(1,2),(1,4), suite_auto_gpt.py:63); activations
+    #: cross mismatched boundaries through the tile-resharding exchange
+    stage_mesh_shapes: Optional[list] = None
     layer_option: str = "auto"  # "auto" | "manual"
     stage_option: str = "uniform"  # "uniform" | "auto" | "manual"
     schedule: str = "1f1b"  # "1f1b" | "gpipe" | "inference"

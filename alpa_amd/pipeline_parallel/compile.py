@@ -22,8 +22,13 @@ from .spec import PipelineModelSpec
 
 
 def resolve_stage_layout(method, n: Optional[int] = None):
-    """Choose (num_stages, per-stage (dp, tp))."""
+    """Choose (num_stages, per-stage shapes).  Returns (P, shapes) where
+    shapes is a list of one (dp, tp) per stage (heterogeneous allowed)."""
     n = n or world_size()
+    if method.stage_mesh_shapes is not None:
+        shapes = [tuple(sh) for sh in method.stage_mesh_shapes]
+        assert sum(a * b for a, b in shapes) == n, (shapes, n)
+        return len(shapes), shapes
     P = method.num_stages
     if P is None:
         from .stage_construction import auto_num_stages
@@ -32,7 +37,7 @@ def resolve_stage_layout(method, n: Optional[int] = None):
     per = n // P
     shape = method.stage_mesh_shape or (per, 1)
     assert shape[0] * shape[1] == per, (shape, per)
-    return P, shape
+    return P, [shape] * P
 
 
 def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
@@ -41,26 +46,31 @@ def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
     from ..optim import AdamW
 
     n = world_size()
-    P, stage_shape = resolve_stage_layout(method, n)
-    per = n // P
+    P, stage_shapes = resolve_stage_layout(method, n)
+    sizes = [a * b for a, b in stage_shapes]
+    starts = [sum(sizes[:s]) for s in range(P)]
+    hetero = len(set(stage_shapes)) > 1
 
     # contiguous rank blocks per stage; all ranks create all stage meshes
     # (group creation is collective)
     stage_meshes = []
     for s in range(P):
-        ranks = tuple(range(s * per, (s + 1) * per))
-        stage_meshes.append(get_device_mesh(ranks, stage_shape))
+        ranks = tuple(range(starts[s], starts[s] + sizes[s]))
+        stage_meshes.append(get_device_mesh(ranks, stage_shapes[s]))
 
     my_rank = rank()
-    my_stage = my_rank // per
+    my_stage = next(s for s in range(P)
+                    if starts[s] <= my_rank < starts[s] + sizes[s])
     my_mesh = stage_meshes[my_stage]
     coord = my_mesh.coord
+    stage_shape = stage_shapes[my_stage]
 
     prev_peer = next_peer = None
-    if my_stage > 0:
-        prev_peer = int(stage_meshes[my_stage - 1].grid[coord])
-    if my_stage < P - 1:
-        next_peer = int(stage_meshes[my_stage + 1].grid[coord])
+    if not hetero:
+        if my_stage > 0:
+            prev_peer = int(stage_meshes[my_stage - 1].grid[coord])
+        if my_stage < P - 1:
+            next_peer = int(stage_meshes[my_stage + 1].grid[coord])
 
     # layer clustering (auto DP over costs; reference layer_construction.py:342)
     costs = spec.layer_costs or uniform_layer_costs(spec.num_layers)
@@ -120,8 +130,18 @@ def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
         act_shape=None,  # resolved lazily from the first microbatch
         act_dtype=dtype, schedule=method.schedule, grad_sync=gs)
     engine._act_shape_fn = spec.act_shape
+    if hetero:
+        # heterogeneous boundaries: activations cross via the tile
+        # resharding exchange (batch-dim placements per stage); the
+        # engine slices GLOBAL microbatches by its stage's dp
+        engine.hetero = True
+        engine.stage_meshes = stage_meshes
+        engine.stage_shapes = stage_shapes
     # rank that holds the authoritative loss (first rank of last stage)
     engine.loss_src_rank = int(stage_meshes[P - 1].ranks[0])
+    # grad scale divides by the LAST stage's dp (the loss-definition dp);
+    # equals every stage's dp in the uniform case
+    engine.loss_dp = stage_shapes[-1][0]
 
     # tied-weight cross-stage allreduce groups (reference N15): one group
     # per (tied set x mesh coordinate); every rank creates every group

@@ -54,6 +54,12 @@ class PipelineEngine:
         self._act_shape_fn = None  # lazy: microbatch -> act shape
         self.loss_src_rank = 0
         self.tied_comms = []  # [(process_group, local param)]
+        # heterogeneous stage meshes (reference auto-search submeshes):
+        # activations cross mismatched boundaries via tile resharding
+        self.hetero = False
+        self.stage_meshes = None
+        self.stage_shapes = None
+        self._specs = None  # (prev_fwd, prev_bwd, next_fwd, next_bwd)
 
     # ------------------------- p2p primitives -------------------------
     def _p2p(self, ops: List[dist.P2POp]):
@@ -63,25 +69,95 @@ class PipelineEngine:
         for w in works:
             w.wait()
 
+    # heterogeneous boundaries: tile-resharding exchange specs
+    def _act_placement(self, stage: int):
+        from ..parallel.resharding import Placement
+        dp, tp = self.stage_shapes[stage]
+        grid = self.stage_meshes[stage].grid
+        ranks = tuple(int(r) for r in grid.flatten(order="F"))
+        parts = (dp,) + (1,) * (len(self.act_shape) - 1)
+        return Placement(tuple(self.act_shape), parts, ranks)
+
+    def _build_specs(self):
+        from ..parallel.resharding import ReshardingTaskSpec
+        pf = pb = nf = nb = None
+        if self.s > 0:
+            a, b = self._act_placement(self.s - 1), self._act_placement(self.s)
+            pf = ReshardingTaskSpec.build(a, b)
+            pb = ReshardingTaskSpec.build(b, a)
+        if self.s < self.P - 1:
+            a, b = self._act_placement(self.s), self._act_placement(self.s + 1)
+            nf = ReshardingTaskSpec.build(a, b)
+            nb = ReshardingTaskSpec.build(b, a)
+        self._specs = (pf, pb, nf, nb)
+
+    def _local_tile_shape(self) -> tuple:
+        dp = self.stage_shapes[self.s][0]
+        return (self.act_shape[0] // dp,) + tuple(self.act_shape[1:])
+
+    def _hetero_exchange(self, send=None, recv_prev=False, recv_next=False,
+                         send_prev=False, send_next=False):
+        """One batched exchange combining up to one send and one recv with
+        the adjacent stages (both sides post their batch at the matching
+        schedule slot — deadlock-free)."""
+        from ..parallel.resharding import apply_fixups, prepare_resharding
+        if self._specs is None:
+            self._build_specs()
+        pf, pb, nf, nb = self._specs
+        ops, fixups = [], []
+        buf = None
+        if send_next:  # forward activation to next stage
+            o, f = prepare_resharding(nf, send.contiguous(), None)
+            ops += o; fixups += f
+        if send_prev:  # gradient to previous stage
+            o, f = prepare_resharding(pb, send.contiguous(), None)
+            ops += o; fixups += f
+        if recv_prev:  # activation from previous stage
+            buf = torch.empty(self._local_tile_shape(), dtype=self.act_dtype,
+                              device=self.device)
+            o, f = prepare_resharding(pf, None, buf)
+            ops += o; fixups += f
+        if recv_next:  # gradient from next stage
+            buf = torch.empty(self._local_tile_shape(), dtype=self.act_dtype,
+                              device=self.device)
+            o, f = prepare_resharding(nb, None, buf)
+            ops += o; fixups += f
+        self._p2p(ops)
+        apply_fixups(fixups)
+        return buf
+
     def _recv_forward(self) -> torch.Tensor:
+        if self.hetero:
+            return self._hetero_exchange(recv_prev=True)
         buf = torch.empty(self.act_shape, dtype=self.act_dtype,
                           device=self.device)
         self._p2p([dist.P2POp(dist.irecv, buf, self.prev_peer)])
         return buf
 
     def _send_forward(self, y: torch.Tensor):
+        if self.hetero:
+            self._hetero_exchange(send=y, send_next=True)
+            return
         self._p2p([dist.P2POp(dist.isend, y.contiguous(), self.next_peer)])
 
     def _recv_backward(self) -> torch.Tensor:
+        if self.hetero:
+            return self._hetero_exchange(recv_next=True)
         buf = torch.empty(self.act_shape, dtype=self.act_dtype,
                           device=self.device)
         self._p2p([dist.P2POp(dist.irecv, buf, self.next_peer)])
         return buf
 
     def _send_backward(self, g: torch.Tensor):
+        if self.hetero:
+            self._hetero_exchange(send=g, send_prev=True)
+            return
         self._p2p([dist.P2POp(dist.isend, g.contiguous(), self.prev_peer)])
 
     def _send_forward_recv_backward(self, y: torch.Tensor) -> torch.Tensor:
+        if self.hetero:
+            return self._hetero_exchange(send=y, send_next=True,
+                                         recv_next=True)
         buf = torch.empty(self.act_shape, dtype=self.act_dtype,
                           device=self.device)
         self._p2p([dist.P2POp(dist.isend, y.contiguous(), self.next_peer),
@@ -89,6 +165,9 @@ class PipelineEngine:
         return buf
 
     def _send_backward_recv_forward(self, g: torch.Tensor) -> torch.Tensor:
+        if self.hetero:
+            return self._hetero_exchange(send=g, send_prev=True,
+                                         recv_prev=True)
         buf = torch.empty(self.act_shape, dtype=self.act_dtype,
                           device=self.device)
         self._p2p([dist.P2POp(dist.isend, g.contiguous(), self.prev_peer),
@@ -101,7 +180,12 @@ class PipelineEngine:
         """Returns (input, output): output is loss on the last stage."""
         if not self.is_first:
             x = x.requires_grad_(True)
-        out = self.stage(x, microbatches[mb_idx])
+        mb = microbatches[mb_idx]
+        if self.hetero:
+            from ..data_loader import shard_batch
+            dp = self.stage_shapes[self.s][0]
+            mb = shard_batch(mb, dp, max(self.mesh.axis_index(0), 0))
+        out = self.stage(x, mb)
         return x, out
 
     def _backward_step(self, x, out, out_grad,
@@ -188,8 +272,17 @@ class PipelineEngine:
             self.grad_sync.finish()
         self._sync_tied_grads()
         if losses:
-            return torch.stack(losses).mean()
+            return self._mean_loss(losses)
         return torch.zeros((), device=self.device)
+
+    def _mean_loss(self, losses):
+        loss = torch.stack(losses).mean()
+        dp = self.mesh.axis_size(0)
+        if self.is_last and dp > 1:
+            loss = loss.clone()
+            self.mesh.all_reduce(loss, axis=0)
+            loss = loss / dp
+        return loss
 
     def _sync_tied_grads(self):
         """Cross-stage all-reduce of tied-weight grads (reference
@@ -224,7 +317,7 @@ class PipelineEngine:
             self.grad_sync.finish()
         self._sync_tied_grads()
         if losses:
-            return torch.stack(losses).mean()
+            return self._mean_loss(losses)
         return torch.zeros((), device=self.device)
 
     # GPipe backward consumes grads in reverse mb order — receive order from

@@ -205,3 +205,27 @@ def test_tied_embeddings_pipeline_matches_serial():
     for r in results:
         for a, b in zip(r, serial):
             assert abs(a - b) < 3e-4, (r, serial)
+
+
+def _hetero_worker(rank, world_size):
+    """2 stages with DIFFERENT submeshes: stage0 (1,1), stage1 (2,1) —
+    the reference auto-search's heterogeneous submesh shape
+    (suite_auto_gpt.py:63)."""
+    method = aa.PipeshardParallel(num_micro_batches=2,
+                                  stage_mesh_shapes=[(1, 1), (2, 1)])
+    spec = gpt_pipeline_spec(CFG)
+    spec.build_stage = _stage_builder
+    state = aa.TrainState.create(spec, method, lr=1e-3)
+    step = aa.parallelize(lambda m, b: None, method=method)
+    # heterogeneous engines take GLOBAL batches and slice per-stage
+    return [float(step(state, make_batch(i))) for i in range(STEPS)]
+
+
+def test_heterogeneous_stage_meshes_match_serial():
+    """Activations reshard through the tile-exchange at the (1,1)->(2,1)
+    boundary; losses must match serial."""
+    serial = run_serial(2)
+    results = run_distributed(_hetero_worker, world_size=3, timeout=300)
+    for r in results:
+        for a, b in zip(r, serial):
+            assert abs(a - b) < 3e-4, (r, serial)
