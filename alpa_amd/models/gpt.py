@@ -255,10 +255,14 @@ class GPTStage(nn.Module):
                                               init_seed=init_seed,
                                               init_tag="wte")
             self.wpe = nn.Parameter(_wpe_init(cfg, dtype, device, init_seed))
-        self.blocks = nn.ModuleList([
-            Block(cfg, mesh, axis, dtype, device, i, init_seed)
-            for i in range(layer_range[0], layer_range[1])
-        ])
+        # blocks registered under their GLOBAL layer indices so stage
+        # checkpoints interoperate with the serial layout (restore under a
+        # different parallelization — the reference's headline
+        # checkpoint feature)
+        self.blocks = nn.Module()
+        for i in range(layer_range[0], layer_range[1]):
+            self.blocks.add_module(str(i), Block(cfg, mesh, axis, dtype,
+                                                 device, i, init_seed))
         if is_last:
             self.ln_f = LayerNorm(cfg.hidden_size, cfg.layernorm_eps, dtype,
                                   device)
@@ -278,7 +282,7 @@ class GPTStage(nn.Module):
             S = ids.shape[1]
             x = self.wte(ids) + self.wpe[:S]
         res, delta = x, None
-        for blk in self.blocks:
+        for blk in self.blocks.children():
             res, delta = blk.forward_fused(res, delta)
         if not self.is_last:
             # materialize the stream value at the stage boundary

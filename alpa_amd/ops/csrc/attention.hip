@@ -297,6 +297,220 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   }
 }
 
+// ===========================================================================
+// 32x32x16-MFMA forward variant (EXPERIMENTAL, measured SLOWER: 570us vs
+// 272us at the bench shape).  Wave owns 32 q rows; 16 waves per block.
+// Numerically verified (maxdiff 2e-3 vs v1).  Why it loses: 1024-thread
+// blocks require 4 waves/SIMD residency, capping VGPR at 128, while the
+// f32x16 accumulators want ~150 — the compiler spills 272 B/lane to
+// scratch.  The path forward (round 2) is keeping P in registers via
+// permlane exchanges to shed the accumulator+LDS round-trip pressure.
+// Kept as a measured data point + layout reference (mfma_probe32).
+// ===========================================================================
+#define ATTN32_THREADS 1024
+
+template <int Dp>
+__global__ __launch_bounds__(ATTN32_THREADS) void attn_fwd_kernel32(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, short* __restrict__ o,
+    float* __restrict__ lse_out, int H, int S, int Skv, int D, float scale,
+    int causal, AttnStrides st) {
+  constexpr int KS_QK = Dp / 16;       // k-steps over head dim (K=16)
+  constexpr int MT = 2;                // 2 kv m-tiles of 32 per 64-kv tile
+  constexpr int DT = Dp / 32;          // d tiles of 32 (O^T rows)
+  constexpr int LP = 8;
+  constexpr int NW = ATTN32_THREADS / 64;  // 16 waves
+  constexpr int GPR = Dp / 8;
+  constexpr int TOTAL_G = 64 * GPR;
+
+  const int qb = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int batch = bh / H, head = bh % H;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int lo = lane & 31;            // q column / m row (32-wide)
+  const int hi = lane >> 5;            // 0/1
+
+  const short* qp = q + batch * st.qb + head * st.qh;
+  const short* kp = k + batch * st.kb + head * st.kh;
+  const short* vp = v + batch * st.vb + head * st.vh;
+  short* op = o + batch * st.ob + head * st.oh;
+  const int q_row0 = qb * (32 * NW) + wave * 32;
+  const int my_q = q_row0 + lo;
+
+  __shared__ short k_lds[2][64][Dp + LP];
+  __shared__ short vt_lds[2][Dp][64 + LP];
+  __shared__ short p_lds[NW][32][64 + LP];
+
+  // Q fragments (B-operand): q_frag[ks][j] = Q[my_q][ks*16 + hi*8 + j]
+  bf16x8 q_frag[KS_QK];
+  {
+    int row = min(my_q, S - 1);
+#pragma unroll
+    for (int ks = 0; ks < KS_QK; ++ks) {
+      int col = ks * 16 + hi * 8;
+      bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+      q_frag[ks] = z;
+      if (col + 8 <= D)
+        q_frag[ks] =
+            *reinterpret_cast<const bf16x8*>(qp + (int64_t)row * st.qs + col);
+    }
+  }
+
+  float m_state = -INFINITY, l_state = 0.f;
+  f32x16 o_acc[DT];
+#pragma unroll
+  for (int dt = 0; dt < DT; ++dt) o_acc[dt] = f32x16{};
+
+  const int kv_limit = causal ? min(Skv, qb * (32 * NW) + 32 * NW) : Skv;
+  const int n_tiles = (kv_limit + 63) / 64;
+
+  bf16x8 kreg[2], vreg[2];
+  auto issue_loads = [&](int tile) {
+    int t = threadIdx.x;
+    bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+    kreg[0] = kreg[1] = vreg[0] = vreg[1] = z;
+    if (t < TOTAL_G / 2) {
+      int kvr = (t / GPR) * 2, dg = (t % GPR) * 8;
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        int src = tile * 64 + kvr + u;
+        if (src < Skv && dg + 8 <= D) {
+          kreg[u] = *reinterpret_cast<const bf16x8*>(
+              kp + (int64_t)src * st.ks + dg);
+          vreg[u] = *reinterpret_cast<const bf16x8*>(
+              vp + (int64_t)src * st.vs + dg);
+        }
+      }
+    }
+  };
+  auto write_tile = [&](int buf) {
+    int t = threadIdx.x;
+    if (t < TOTAL_G / 2) {
+      int kvr = (t / GPR) * 2, dg = (t % GPR) * 8;
+      *reinterpret_cast<bf16x8*>(&k_lds[buf][kvr][dg]) = kreg[0];
+      *reinterpret_cast<bf16x8*>(&k_lds[buf][kvr + 1][dg]) = kreg[1];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        short2 pr;
+        pr.x = vreg[0][j];
+        pr.y = vreg[1][j];
+        *reinterpret_cast<short2*>(&vt_lds[buf][dg + j][kvr]) = pr;
+      }
+    }
+  };
+
+  if (n_tiles > 0) {
+    issue_loads(0);
+    write_tile(0);
+    __syncthreads();
+  }
+
+  for (int ti = 0; ti < n_tiles; ++ti) {
+    const int kvb = ti * 64;
+    const int cur = ti & 1;
+    if (ti + 1 < n_tiles) issue_loads(ti + 1);
+
+    if (causal && kvb > q_row0 + 31) {
+      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      continue;
+    }
+
+    // ---- S^T = K Q^T: A = K (m=kv), B = q_frag ----
+    f32x16 s_acc[MT];
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+      f32x16 acc = {};
+#pragma unroll
+      for (int ks = 0; ks < KS_QK; ++ks) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &k_lds[cur][mt * 32 + lo][ks * 16 + hi * 8]);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, q_frag[ks], acc, 0,
+                                                      0, 0);
+      }
+      s_acc[mt] = acc;
+    }
+
+    // ---- mask + scale + lane-local max (32 values for q = my_q) ----
+    float tile_max = -INFINITY;
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int kv_idx = kvb + mt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        float sv = s_acc[mt][r] * scale;
+        bool masked = (kv_idx >= Skv) || (causal && kv_idx > my_q) ||
+                      (my_q >= S);
+        sv = masked ? -INFINITY : sv;
+        s_acc[mt][r] = sv;
+        tile_max = fmaxf(tile_max, sv);
+      }
+    }
+    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+
+    float m_new = fmaxf(m_state, tile_max);
+    float alpha = (m_state == -INFINITY) ? 0.f : __expf(m_state - m_new);
+    m_state = m_new;
+    l_state *= alpha;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) o_acc[dt][r] *= alpha;
+
+    float part = 0.f;
+#pragma unroll
+    for (int mt = 0; mt < MT; ++mt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float pval = (s_acc[mt][r] == -INFINITY)
+                         ? 0.f
+                         : __expf(s_acc[mt][r] - m_state);
+        part += pval;
+        p_lds[wave][lo][mt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi] =
+            f2bf(pval);
+      }
+    }
+    part += __shfl_xor(part, 32, 64);
+    l_state += part;
+
+    // ---- O^T += V^T P^T: A = V^T (m=d), B = P^T (n=q) ----
+#pragma unroll
+    for (int ks = 0; ks < 64 / 16; ++ks) {
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          &p_lds[wave][lo][ks * 16 + hi * 8]);
+#pragma unroll
+      for (int dt = 0; dt < DT; ++dt) {
+        bf16x8 a = *reinterpret_cast<const bf16x8*>(
+            &vt_lds[cur][dt * 32 + lo][ks * 16 + hi * 8]);
+        o_acc[dt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, o_acc[dt],
+                                                            0, 0, 0);
+      }
+    }
+
+    if (ti + 1 < n_tiles) {
+      write_tile(1 - cur);
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: lane holds O^T[d = dt*32 + (r&3)+8*(r>>2)+4*hi][my_q] --
+  if (my_q < S) {
+    float inv_l = (l_state > 0.f) ? 1.0f / l_state : 0.f;
+#pragma unroll
+    for (int dt = 0; dt < DT; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int col = dt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+        if (col < D)
+          op[(int64_t)my_q * st.os + col] = f2bf(o_acc[dt][r] * inv_l);
+      }
+    }
+    if (hi == 0 && lse_out != nullptr)
+      lse_out[(int64_t)bh * S + my_q] =
+          (l_state > 0.f) ? m_state + __logf(l_state) : -INFINITY;
+  }
+}
+
 extern "C" {
 
 hipError_t launch_attn_fwd_ablate(const void* q, const void* k,
@@ -321,6 +535,37 @@ hipError_t launch_attn_fwd_ablate(const void* q, const void* k,
   }
   ABL_CASE(0) ABL_CASE(1) ABL_CASE(2) ABL_CASE(3) ABL_CASE(4)
 #undef ABL_CASE
+  return hipGetLastError();
+}
+
+hipError_t launch_attn_fwd_v2(const void* q, const void* k, const void* v,
+                              void* o, float* lse, int64_t B, int64_t H,
+                              int64_t S, int64_t Skv, int64_t D, float scale,
+                              int causal, const int64_t* strides,
+                              hipStream_t stream) {
+  dim3 grid((uint32_t)ceil_div(S, 32 * (ATTN32_THREADS / 64)),
+            (uint32_t)(B * H));
+  dim3 block(ATTN32_THREADS);
+  AttnStrides st;
+  st.qb = strides[0]; st.qh = strides[1]; st.qs = strides[2];
+  st.kb = strides[3]; st.kh = strides[4]; st.ks = strides[5];
+  st.vb = strides[6]; st.vh = strides[7]; st.vs = strides[8];
+  st.ob = strides[9]; st.oh = strides[10]; st.os = strides[11];
+  if (D <= 64) {
+    attn_fwd_kernel32<64><<<grid, block, 0, stream>>>(
+        (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+  } else if (D <= 96) {
+    attn_fwd_kernel32<96><<<grid, block, 0, stream>>>(
+        (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+  } else if (D <= 128) {
+    attn_fwd_kernel32<128><<<grid, block, 0, stream>>>(
+        (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+  } else {
+    return hipErrorInvalidValue;
+  }
   return hipGetLastError();
 }
 
@@ -388,6 +633,36 @@ extern "C" hipError_t launch_mfma_probe(const void* a, const void* b,
                                         float* d, hipStream_t stream) {
   mfma_probe_kernel<<<dim3(1), dim3(64), 0, stream>>>((const short*)a,
                                                       (const short*)b, d);
+  return hipGetLastError();
+}
+
+// 32x32x16 layout probe: D(32x32) = A(32x16) @ B(16x32).
+// Assumed layouts: A: m=lane&31, k=(lane>>5)*8+j; B: n=lane&31, same k;
+// C/D (documented, guide §3): col=lane&31, row=(reg&3)+8*(reg>>2)+4*(lane>>5)
+__global__ void mfma_probe32_kernel(const short* __restrict__ a,
+                                    const short* __restrict__ b,
+                                    float* __restrict__ d) {
+  int lane = threadIdx.x & 63;
+  int lo = lane & 31, hi = lane >> 5;
+  bf16x8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = a[lo * 16 + hi * 8 + j];
+    bf[j] = b[(hi * 8 + j) * 32 + lo];
+  }
+  f32x16 acc = {};
+  acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(af, bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    int row = (r & 3) + 8 * (r >> 2) + 4 * hi;
+    d[row * 32 + lo] = acc[r];
+  }
+}
+
+extern "C" hipError_t launch_mfma_probe32(const void* a, const void* b,
+                                          float* d, hipStream_t stream) {
+  mfma_probe32_kernel<<<dim3(1), dim3(64), 0, stream>>>((const short*)a,
+                                                        (const short*)b, d);
   return hipGetLastError();
 }
 

@@ -50,7 +50,13 @@ hipError_t launch_ce_bwd(const float*, const void*, const int64_t*,
 hipError_t launch_attn_fwd(const void*, const void*, const void*, void*,
                            float*, int64_t, int64_t, int64_t, int64_t,
                            int64_t, float, int, const int64_t*, hipStream_t);
+hipError_t launch_attn_fwd_v2(const void*, const void*, const void*, void*,
+                              float*, int64_t, int64_t, int64_t, int64_t,
+                              int64_t, float, int, const int64_t*,
+                              hipStream_t);
 hipError_t launch_mfma_probe(const void*, const void*, float*, hipStream_t);
+hipError_t launch_mfma_probe32(const void*, const void*, float*,
+                               hipStream_t);
 hipError_t launch_attn_fwd_ablate(const void*, const void*, const void*,
                                   void*, float*, int64_t, int64_t, int64_t,
                                   int64_t, int64_t, float, int,
@@ -326,6 +332,25 @@ at::Tensor attn_fwd_ablate(const at::Tensor& q, const at::Tensor& k,
   return o;
 }
 
+std::vector<at::Tensor> attn_fwd_v2(const at::Tensor& q,
+                                    const at::Tensor& k,
+                                    const at::Tensor& v, bool causal,
+                                    double scale) {
+  int64_t B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  auto o = at::empty_like(q.contiguous());
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  int64_t strides[12] = {q.stride(0), q.stride(1), q.stride(2),
+                         k.stride(0), k.stride(1), k.stride(2),
+                         v.stride(0), v.stride(1), v.stride(2),
+                         o.stride(0), o.stride(1), o.stride(2)};
+  HIP_OK(launch_attn_fwd_v2(q.const_data_ptr(), k.const_data_ptr(),
+                            v.const_data_ptr(), o.mutable_data_ptr(),
+                            (float*)lse.mutable_data_ptr(), B, H, S,
+                            k.size(2), D, (float)scale, causal ? 1 : 0,
+                            strides, cur_stream()));
+  return {o, lse};
+}
+
 std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                  const at::Tensor& v, bool causal,
                                  double scale) {
@@ -455,6 +480,14 @@ at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
   return d;
 }
 
+at::Tensor mfma_probe32(const at::Tensor& a, const at::Tensor& b) {
+  check_bf16_contig(a, "a");
+  auto d = at::empty({32, 32}, a.options().dtype(at::kFloat));
+  HIP_OK(launch_mfma_probe32(a.const_data_ptr(), b.const_data_ptr(),
+                             (float*)d.mutable_data_ptr(), cur_stream()));
+  return d;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -471,6 +504,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE bwd (gfx950)");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)");
   m.def("attn_fwd_out", &attn_fwd_out, "flash attention fwd, strided out");
+  m.def("attn_fwd_v2", &attn_fwd_v2, "32x32-MFMA fwd experiment");
   m.def("attn_fwd_ablate", &attn_fwd_ablate, "ablation variants (perf)");
   m.def("attn_bwd_dkv_ablate", &attn_bwd_dkv_ablate, "dkv ablation (perf)");
   m.def("attn_bwd", &attn_bwd, "flash attention bwd (gfx950 MFMA)");
@@ -478,4 +512,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_blocked", &attn_bwd_blocked,
         "attention bwd (blocked hipBLASLt reference)");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");
+  m.def("mfma_probe32", &mfma_probe32, "MFMA 32x32x16 layout probe");
 }

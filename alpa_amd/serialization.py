@@ -167,13 +167,20 @@ def restore_checkpoint(ckpt_dir: str, step: int, target_tree: Any,
 
     def walk(target, packed, prefix=""):
         if isinstance(target, dict):
-            return {k: walk(v, packed[k], f"{prefix}{k}.")
+            return {k: walk(v, (packed or {}).get(k), f"{prefix}{k}.")
                     for k, v in target.items()}
         path = prefix[:-1]
         if torch.is_tensor(target):
-            assert isinstance(packed, dict) and "__tensor_dir__" in packed, \
-                f"{path}: checkpoint has no tensor dir"
-            tdir = os.path.join(ckpt_dir, packed["__tensor_dir__"])
+            if isinstance(packed, dict) and "__tensor_dir__" in packed:
+                dirname = packed["__tensor_dir__"]
+            else:
+                # the tree writer (rank 0) may not own this leaf (e.g. a
+                # later pipeline stage wrote its shards); fall back to the
+                # deterministic directory naming
+                dirname = _tensor_dir_name(path)
+            tdir = os.path.join(ckpt_dir, dirname)
+            assert os.path.isdir(tdir), \
+                f"{path}: no checkpoint data at {tdir}"
             spec = shard_specs.get(path) or ShardSpec.full(target.shape)
             want = spec.index
             np_dtype = np.float32 if target.dtype in (torch.bfloat16,
@@ -194,7 +201,8 @@ def restore_checkpoint(ckpt_dir: str, step: int, target_tree: Any,
 # ----------------------------------------------------------------------
 
 
-def model_shard_specs(model: torch.nn.Module) -> Dict[str, ShardSpec]:
+def model_shard_specs(model: torch.nn.Module,
+                      owner_mesh=None) -> Dict[str, ShardSpec]:
     """Derive per-parameter ShardSpecs from the parallel layer modules
     (Column/Row/VocabParallel know their global shape + slice; everything
     else is replicated, written by the global writer rank of its dp
@@ -240,11 +248,15 @@ def model_shard_specs(model: torch.nn.Module) -> Dict[str, ShardSpec]:
             specs[prefix + "weight"] = ShardSpec(
                 (mod.num_embeddings, emb),
                 (slice(idx * v, (idx + 1) * v), slice(0, emb)), dp_writer)
-    # default for unsharded params: replicated; written once per dp group
+    # default for unsharded params: replicated across the OWNING mesh
+    # (the stage mesh for pipeline states) — its first rank writes
     for name, p in model.named_parameters():
         if name not in specs:
-            writer = rank() == 0 or not is_distributed()
-            # in a TP group, tp rank 0 of dp rank 0 writes
+            if owner_mesh is not None:
+                writer = owner_mesh.is_member and \
+                    rank() == owner_mesh.ranks[0]
+            else:
+                writer = rank() == 0 or not is_distributed()
             specs[name] = ShardSpec.full(p.shape, writer)
     return specs
 
@@ -253,7 +265,7 @@ def _train_state_tree_and_specs(state):
     """(tree, specs) for a TrainState: params + AdamW moments (the moments
     share the parameter sharding).  ZeRO-2's bucket-sharded moments are
     topology-specific and handled by its own state_dict."""
-    specs = model_shard_specs(state.model)
+    specs = model_shard_specs(state.model, getattr(state, "mesh", None))
     names = [n for n, _ in state.model.named_parameters()]
     tree = {"params": dict(state.model.state_dict()),
             "step": state.step_count}

@@ -112,3 +112,46 @@ def test_checkpoint_format_layout(tmp_path):
     assert meta["shard_names"] == ["shard_0.0"]
     arr = np.load(tdir / "shard_0.0.npy")
     assert arr.shape == (CFG.vocab_size, CFG.hidden_size)
+
+
+def _pipeline_save_worker(rank, world_size, path):
+    from alpa_amd.models.gpt import gpt_pipeline_spec
+    method = aa.PipeshardParallel(num_micro_batches=1,
+                                  num_stages=world_size,
+                                  stage_mesh_shape=(1, 1))
+    spec = gpt_pipeline_spec(CFG)
+
+    def build_stage(layer_range, is_first, is_last, mesh, axis, dtype,
+                    device):
+        from alpa_amd.models.gpt import GPTStage
+        return GPTStage(CFG, layer_range, is_first, is_last, mesh, axis,
+                        dtype, device, init_seed=5)
+
+    spec.build_stage = build_stage
+    state = aa.TrainState.create(spec, method)
+    save_train_state(str(path), state, step=0)
+    return True
+
+
+def test_pipeline_save_serial_restore(tmp_path):
+    """A checkpoint written under 2-stage pipeline parallelism restores
+    into the serial layout (stage blocks carry GLOBAL layer names) — the
+    reference's restore-under-different-parallelization feature."""
+    run_distributed(_pipeline_save_worker, world_size=2,
+                    args=(str(tmp_path),))
+    state, _ = make_state((1, 1))
+    ref_params = {n: p.detach().clone()
+                  for n, p in state.model.named_parameters()}
+    with torch.no_grad():
+        for p in state.model.parameters():
+            p.mul_(0.0)
+    # model-only restore (optimizer moments saved per stage param set)
+    from alpa_amd.serialization import (model_shard_specs,
+                                        restore_checkpoint)
+    specs = {f"params.{k}": v
+             for k, v in model_shard_specs(state.model).items()}
+    tree = {"params": dict(state.model.state_dict())}
+    restore_checkpoint(str(tmp_path), 0, tree, specs)
+    for n, p in state.model.named_parameters():
+        torch.testing.assert_close(p.detach(), ref_params[n], rtol=1e-6,
+                                   atol=1e-6, msg=lambda m: f"{n}: {m}")

@@ -337,3 +337,13 @@ def test_zero_paths_single_gpu():
         for _ in range(4):
             last = float(step(state, (x, y)))
         assert last < first, (type(method).__name__, first, last)
+
+
+def test_mfma32_layout_probe(ext):
+    """Verify the 32x32x16 fragment layouts with asymmetric operands."""
+    torch.manual_seed(13)
+    a = torch.randn(32, 16, device="cuda").to(torch.bfloat16)
+    b = torch.randn(16, 32, device="cuda").to(torch.bfloat16)
+    d = ext.mfma_probe32(a.contiguous(), b.contiguous())
+    d_ref = a.float() @ b.float()
+    torch.testing.assert_close(d, d_ref, rtol=2e-2, atol=2e-2)
