@@ -209,6 +209,7 @@ def model_shard_specs(model: torch.nn.Module,
     group)."""
     from .parallel.layers import (ColumnParallelLinear, RowParallelLinear,
                                   VocabParallelEmbedding)
+    from .parallel.expert import ExpertParallelMLP
     specs: Dict[str, ShardSpec] = {}
     for mod_name, mod in model.named_modules():
         prefix = f"{mod_name}." if mod_name else ""
@@ -242,6 +243,13 @@ def model_shard_specs(model: torch.nn.Module,
                 specs[prefix + "bias"] = ShardSpec(
                     (mod.out_features,), (slice(0, mod.out_features),),
                     dp_writer and idx == 0)
+        elif isinstance(mod, ExpertParallelMLP) and mod.ep > 1:
+            # expert weights: dim-0 shards of the global [E, ...] stack
+            for wname, w in (("w1", mod.w1), ("w2", mod.w2)):
+                gshape = (mod.E,) + tuple(w.shape[1:])
+                idx = (slice(mod.e_start, mod.e_start + mod.e_local),) + \
+                    tuple(slice(0, s) for s in w.shape[1:])
+                specs[prefix + wname] = ShardSpec(gshape, idx, dp_writer)
         elif isinstance(mod, VocabParallelEmbedding) and tp > 1:
             v = mod.vocab_per_rank
             emb = mod.weight.shape[1]
