@@ -299,8 +299,13 @@ class GPTStage(nn.Module):
         return per_tok.mean()
 
 
-def gpt_pipeline_spec(cfg: GPTConfig):
-    """PipelineModelSpec for the GPT family (used by PipeshardParallel)."""
+def gpt_pipeline_spec(cfg: GPTConfig, microbatch_tokens: int = 0):
+    """PipelineModelSpec for the GPT family (used by PipeshardParallel).
+
+    Pass ``microbatch_tokens`` (= microbatch size x seq_len) to enable the
+    profile-guided auto stage search (`stage_option="auto"` with
+    num_stages=None): it fills per-layer flops / boundary bytes / param
+    bytes for stage_construction.profiled_stage_search."""
     from ..pipeline_parallel.spec import PipelineModelSpec
 
     def build_stage(layer_range, is_first, is_last, mesh, axis, dtype,
@@ -319,4 +324,13 @@ def gpt_pipeline_spec(cfg: GPTConfig):
     if cfg.tie_embeddings:
         # stage -1 = last; resolved by the pipeline compiler
         spec.tied_groups = [{0: "wte.weight", -1: "lm_head.weight"}]
+    if microbatch_tokens:
+        H, S = cfg.hidden_size, cfg.seq_len
+        T = microbatch_tokens
+        # fwd+bwd GEMM flops per block (attention QKV/proj + 2 FFN mats +
+        # the S^2 score/context GEMMs), factor 3 for fwd + dX + dW
+        per_layer = 3.0 * 2.0 * T * (12 * H * H + 2 * S * H)
+        spec.layer_flops = [per_layer] * cfg.num_layers
+        spec.boundary_act_bytes = 2.0 * T * H
+        spec.layer_param_bytes = [2.0 * 12 * H * H] * cfg.num_layers
     return spec

@@ -21,9 +21,16 @@ from .runtime import PipelineEngine
 from .spec import PipelineModelSpec
 
 
-def resolve_stage_layout(method, n: Optional[int] = None):
+def resolve_stage_layout(method, n: Optional[int] = None,
+                         spec: Optional[PipelineModelSpec] = None):
     """Choose (num_stages, per-stage shapes).  Returns (P, shapes) where
-    shapes is a list of one (dp, tp) per stage (heterogeneous allowed)."""
+    shapes is a list of one (dp, tp) per stage (heterogeneous allowed).
+
+    With ``stage_option="auto"`` and a spec carrying per-layer flops, the
+    profile-guided search (stage_construction.profiled_stage_search)
+    picks (P, submesh) from the measured cost DB — the closed
+    profiling loop (reference stage_construction training_dp fed by
+    HloCostModelProfileWorker)."""
     n = n or world_size()
     if method.stage_mesh_shapes is not None:
         shapes = [tuple(sh) for sh in method.stage_mesh_shapes]
@@ -31,6 +38,14 @@ def resolve_stage_layout(method, n: Optional[int] = None):
         return len(shapes), shapes
     P = method.num_stages
     if P is None:
+        if (method.stage_option == "auto" and spec is not None
+                and spec.layer_flops is not None
+                and method.stage_mesh_shape is None):
+            from .stage_construction import profiled_stage_search
+            P, shapes, _ranges, _cost = profiled_stage_search(
+                n, method.num_micro_batches, spec.layer_flops,
+                spec.boundary_act_bytes, spec.layer_param_bytes)
+            return P, shapes
         from .stage_construction import auto_num_stages
         P = auto_num_stages(n, method.num_micro_batches)
     assert n % P == 0, f"world {n} not divisible by {P} stages"
@@ -46,7 +61,7 @@ def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
     from ..optim import AdamW
 
     n = world_size()
-    P, stage_shapes = resolve_stage_layout(method, n)
+    P, stage_shapes = resolve_stage_layout(method, n, spec)
     sizes = [a * b for a, b in stage_shapes]
     starts = [sum(sizes[:s]) for s in range(P)]
     hetero = len(set(stage_shapes)) > 1

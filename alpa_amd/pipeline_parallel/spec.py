@@ -38,3 +38,9 @@ class PipelineModelSpec:
     #: across the same-coordinate ranks of those stages each step
     #: (reference cross-mesh allreduce for tied embeddings, N15)
     tied_groups: Optional[list] = None
+    #: optional per-MICROBATCH numbers for the profile-guided stage
+    #: search (stage_construction.profiled_stage_search): fwd+bwd flops
+    #: per layer, boundary activation bytes, bf16 param bytes per layer
+    layer_flops: Optional[Sequence[float]] = None
+    boundary_act_bytes: float = 0.0
+    layer_param_bytes: Optional[Sequence[float]] = None

@@ -71,3 +71,118 @@ def auto_num_stages(num_devices: int, num_microbatches: int,
                     num_layers: int = 32) -> int:
     return choose_stages(num_devices, num_microbatches,
                          num_layers=num_layers)[0]
+
+
+def _factorizations(per: int) -> List[Tuple[int, int]]:
+    return [(dp, per // dp) for dp in divisors(per)]
+
+
+def _default_db():
+    """The profiled cost DB (tools/calibrate.py -> prof_database.pkl) or
+    an analytic fallback (insert_dummy_mesh_result)."""
+    import os
+    from ..mesh_profiling import ProfilingResultDatabase
+    db = ProfilingResultDatabase()
+    path = global_config.prof_database_path
+    if os.path.exists(path):
+        try:
+            db.load(path)
+        except Exception:
+            pass
+    return db
+
+
+def _matmul_curve(db, cluster_key: str):
+    for (key, _shape), r in db.data.items():
+        if key == cluster_key and "matmul_bf16" in r.op_curves:
+            return r.op_curves["matmul_bf16"]
+    for (_key, _shape), r in db.data.items():
+        if "matmul_bf16" in r.op_curves:
+            return r.op_curves["matmul_bf16"]
+    return None
+
+
+def profiled_stage_search(num_devices: int, num_microbatches: int,
+                          layer_flops: Sequence[float],
+                          boundary_act_bytes: float = 0.0,
+                          layer_param_bytes: Optional[Sequence[float]] = None,
+                          db=None, cluster_key: str = "mi355x",
+                          max_stages: Optional[int] = None
+                          ) -> Tuple[int, List[Tuple[int, int]],
+                                     List[Tuple[int, int]], float]:
+    """Profile-guided inter-op search (reference training_dp,
+    stage_construction.py:235 fed by HloCostModelProfileWorker:414):
+    enumerate (stage count P, uniform submesh (dp, tp)) x layer
+    clusterings; per-stage latency comes from the MEASURED cost curves
+    (hipBLASLt matmul flops->s; RCCL collective bytes->s when profiled,
+    alpha-beta xGMI model otherwise).
+
+    Inputs are per-MICROBATCH: ``layer_flops`` fwd+bwd flops per layer,
+    ``boundary_act_bytes`` the stage-boundary activation size.  Returns
+    (P, [(dp, tp)] * P, layer ranges, estimated step seconds).
+
+    Why the profiled curve matters: splitting a layer tp ways moves its
+    GEMMs down the measured efficiency curve (small GEMMs run far below
+    peak), which an analytic flops/peak model cannot see — this is the
+    closed profiling loop of SURVEY.md §5.1.
+    """
+    if db is None:
+        db = _default_db()
+    curve = _matmul_curve(db, cluster_key)
+    L = len(layer_flops)
+    alpha = global_config.mesh_alpha
+    beta = global_config.mesh_beta
+    M = num_microbatches
+
+    def coll_time(kind, mesh_shape, axis, nbytes):
+        if nbytes <= 0:
+            return 0.0
+        try:
+            r = db.query(cluster_key, mesh_shape)
+            return r.estimate_collective(kind, axis, nbytes)
+        except KeyError:
+            n = mesh_shape[axis]
+            factor = 2 * (n - 1) / n if kind == "all_reduce" \
+                else (n - 1) / n
+            return alpha + factor * nbytes * beta
+
+    def matmul_time(flops):
+        if curve is not None:
+            return curve.estimate(flops)
+        return flops / 1.2e15 + 5e-6
+
+    best = None
+    for P in divisors(num_devices):
+        if P > L or (max_stages and P > max_stages):
+            continue
+        per = num_devices // P
+        ranges = cluster_layers(layer_flops, P)
+        for (dp, tp) in _factorizations(per):
+            stage_costs = []
+            grad_ar = 0.0
+            for (a, b) in ranges:
+                f = sum(layer_flops[a:b])
+                # compute: per-device GEMM work down the measured curve
+                t = matmul_time(f / (dp * tp))
+                if tp > 1:
+                    # Megatron TP: 4 activation all-reduces per layer
+                    # (2 fwd + 2 bwd) over the tp axis, on this dp
+                    # shard's activation
+                    nb = 4 * (b - a) * boundary_act_bytes / max(dp, 1)
+                    t += coll_time("all_reduce", (dp, tp), 1, nb)
+                if dp > 1 and layer_param_bytes is not None:
+                    # gradient all-reduce once per STEP, mostly hidden
+                    # behind backward (grad_sync overlap discount)
+                    pb = sum(layer_param_bytes[a:b]) / tp
+                    grad_ar = max(grad_ar,
+                                  0.25 * coll_time("all_reduce", (dp, tp),
+                                                   0, pb))
+                stage_costs.append(t)
+            # cross-stage p2p: each dp replica sends its shard on its own
+            # xGMI link concurrently
+            comm = alpha + boundary_act_bytes / max(dp, 1) * beta \
+                if P > 1 else 0.0
+            cost = pipeline_makespan(stage_costs, M, comm) + grad_ar
+            if best is None or cost < best[3]:
+                best = (P, [(dp, tp)] * P, ranges, cost)
+    return best

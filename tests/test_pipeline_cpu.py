@@ -229,3 +229,94 @@ def test_heterogeneous_stage_meshes_match_serial():
     for r in results:
         for a, b in zip(r, serial):
             assert abs(a - b) < 3e-4, (r, serial)
+
+
+# ---------------------------------------------------------------------------
+# Profile-guided auto stage search (reference training_dp fed by the
+# profiled cost DB, stage_construction.py:235 + HloCostModelProfileWorker)
+# ---------------------------------------------------------------------------
+
+
+def _dummy_db(shapes=((1, 1), (2, 1), (1, 2), (4, 1), (2, 2), (1, 4),
+                      (8, 1), (4, 2), (2, 4), (1, 8))):
+    from alpa_amd.mesh_profiling import ProfilingResultDatabase
+    db = ProfilingResultDatabase()
+    for sh in shapes:
+        db.insert_dummy_mesh_result("mi355x", sh)
+    return db
+
+
+def test_profiled_stage_search_valid_layout():
+    from alpa_amd.pipeline_parallel.stage_construction import \
+        profiled_stage_search
+    P, shapes, ranges, cost = profiled_stage_search(
+        8, 16, [1e12] * 8, boundary_act_bytes=1e6,
+        layer_param_bytes=[1e8] * 8, db=_dummy_db())
+    assert len(shapes) == P and len(ranges) == P
+    assert sum(dp * tp for dp, tp in shapes) == 8
+    assert ranges[0][0] == 0 and ranges[-1][1] == 8
+    for (a0, b0), (a1, b1) in zip(ranges, ranges[1:]):
+        assert b0 == a1
+    assert cost > 0
+
+
+def test_profiled_stage_search_comm_sensitivity():
+    """A huge boundary activation makes deep pipelines pay (P-1) transfers
+    per microbatch: the search must pick fewer stages than with a free
+    boundary."""
+    from alpa_amd.pipeline_parallel.stage_construction import \
+        profiled_stage_search
+    db = _dummy_db()
+    P_free, _, _, _ = profiled_stage_search(
+        8, 64, [1e12] * 8, boundary_act_bytes=0.0, db=db)
+    P_heavy, _, _, _ = profiled_stage_search(
+        8, 64, [1e12] * 8, boundary_act_bytes=1e11, db=db)
+    assert P_heavy <= P_free
+
+
+def test_profiled_stage_search_uses_measured_curve():
+    """When the measured matmul curve has a steep latency floor (small
+    GEMMs run far below peak), splitting layers over many devices stops
+    paying — the search must keep per-device work large (few stages x
+    small tp), unlike a pure flops/peak model."""
+    from alpa_amd.mesh_profiling import (CostCurve, MeshProfilingResult,
+                                         ProfilingResultDatabase)
+    from alpa_amd.pipeline_parallel.stage_construction import \
+        profiled_stage_search
+    db = ProfilingResultDatabase()
+    r = MeshProfilingResult((1, 1))
+    c = CostCurve()
+    # latency-floor curve: below 1e12 flops everything costs ~1 ms
+    c.add(1e10, 1e-3)
+    c.add(1e12, 1.1e-3)
+    c.add(1e14, 50e-3)
+    r.op_curves["matmul_bf16"] = c
+    db.update_one_mesh("mi355x", (1, 1), r)
+    P, shapes, _, _ = profiled_stage_search(
+        8, 2, [2e12] * 4, boundary_act_bytes=1e6, db=db)
+    # 4 layers / 8 devices: fine splits (P=4 or tp=8) hit the latency
+    # floor; the winner keeps devices per (stage x shard) coarse
+    assert P * shapes[0][1] <= 4
+
+
+def _auto_pp_worker(rank, world_size, nmb):
+    method = aa.PipeshardParallel(num_micro_batches=nmb,
+                                  stage_option="auto")
+    spec = gpt_pipeline_spec(
+        CFG, microbatch_tokens=BATCH // nmb * CFG.seq_len)
+    spec.build_stage = _stage_builder
+    state = aa.TrainState.create(spec, method, lr=1e-3)
+    step = aa.parallelize(lambda m, b: None, method=method)
+    return [float(step(state, make_batch(i))) for i in range(STEPS)]
+
+
+def test_auto_stage_search_end_to_end():
+    """stage_option="auto" with spec-provided flops resolves a layout via
+    the profiled search and the resulting pipeline still matches the
+    serial oracle."""
+    serial = run_serial(2)
+    results = run_distributed(_auto_pp_worker, world_size=2, args=(2,),
+                              timeout=300)
+    for r in results:
+        for a, b in zip(r, serial):
+            assert abs(a - b) < 2e-4, (r, serial)
