@@ -38,10 +38,14 @@ class Controller:
         return sorted(self.models)
 
     def completions(self, model: str, prompt_ids: List[List[int]],
-                    max_tokens: int = 16) -> List[List[int]]:
+                    max_tokens: int = 16, num_beams: int = 1
+                    ) -> List[List[int]]:
         info = self.models[model]
         ids = torch.tensor(prompt_ids, dtype=torch.long)
-        out = info.generate_fn(ids, max_tokens)
+        if num_beams > 1:
+            out = info.generate_fn(ids, max_tokens, num_beams=num_beams)
+        else:
+            out = info.generate_fn(ids, max_tokens)
         return out.tolist()
 
     # ------------------------- HTTP frontend -------------------------
@@ -58,7 +62,8 @@ class Controller:
             try:
                 out = self.completions(
                     body["model"], body["prompt_ids"],
-                    int(body.get("max_tokens", 16)))
+                    int(body.get("max_tokens", 16)),
+                    int(body.get("num_beams", 1)))
             except KeyError as e:
                 return JSONResponse({"error": f"missing field {e}"},
                                     status_code=400)
