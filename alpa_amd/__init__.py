@@ -21,7 +21,7 @@ from .parallel_method import (AutoShardingOption, DataParallel,
                               Zero2Parallel, Zero3Parallel,
                               get_3d_parallel_method)
 
-__version__ = "0.1.0"
+from .version import __version__, check_hip_ops_version  # noqa: F401
 
 __all__ = [
     "init", "shutdown", "parallelize", "TrainState", "AdamW",

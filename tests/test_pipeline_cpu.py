@@ -320,3 +320,27 @@ def test_auto_stage_search_end_to_end():
     for r in results:
         for a, b in zip(r, serial):
             assert abs(a - b) < 2e-4, (r, serial)
+
+
+def test_local_pipeline_runner_matches_serial():
+    """The single-process sequential stage runner (debug path, reference
+    local_pipeline.py:16) reproduces the serial model's loss and grads."""
+    from alpa_amd.pipeline_parallel.local_pipeline import LocalPipelineRunner
+    spec = gpt_pipeline_spec(CFG)
+    spec.build_stage = _stage_builder
+    runner = LocalPipelineRunner(spec, num_stages=2)
+    serial = GPTModel(CFG, None, 1, torch.float32, None, init_seed=11)
+    batch = make_batch(0)
+    mbs = [{k: v[i * 2:(i + 1) * 2] for k, v in batch.items()}
+           for i in range(2)]
+    loss = runner.train_step(mbs)
+    ref = sum(serial.loss(mb["ids"], mb["labels"]) for mb in mbs) / 2
+    ref.backward()
+    assert abs(float(loss) - float(ref)) < 1e-5
+    ref_params = dict(serial.named_parameters())
+    for s, stage in enumerate(runner.stages):
+        for n, p in stage.named_parameters():
+            assert n in ref_params, n
+            if p.grad is not None and ref_params[n].grad is not None:
+                torch.testing.assert_close(p.grad, ref_params[n].grad,
+                                           rtol=1e-5, atol=1e-5)

@@ -82,3 +82,13 @@ def test_method_to_plan_shard():
     assert plan.stage_plans[0].logical_mesh_shape == (4, 2)
     m2 = plan_to_method(plan)
     assert m2.logical_mesh_shape == (4, 2)
+
+
+def test_version_guard():
+    """ABI guard (reference check_alpa_jaxlib_version): absent extension
+    -> -1 on CPU; a present extension must carry ABI_VERSION >= minimum."""
+    import alpa_amd
+    from alpa_amd.version import MIN_HIP_OPS_ABI, check_hip_ops_version
+    abi = check_hip_ops_version()
+    assert abi == -1 or abi >= MIN_HIP_OPS_ABI
+    assert alpa_amd.__version__
