@@ -296,6 +296,10 @@ def test_opt_generation_on_gpu():
     logits_full = m.forward_step(torch.cat([ids, nxt], 1), cache2)
     torch.testing.assert_close(logits_inc.float(), logits_full.float(),
                                rtol=5e-2, atol=5e-2)
+    # beam search: KV-cache reorder between steps on device
+    beam = m.beam_search(ids, max_new_tokens=6, num_beams=3, eos_token=2)
+    assert beam.shape[0] == 2 and beam.shape[1] <= 22
+    assert beam.is_cuda
 
 
 def test_bert_and_unet_on_gpu():
