@@ -327,3 +327,26 @@ def full_mesh(shape: Optional[Tuple[int, int]] = None) -> DeviceMesh:
     if shape is None:
         shape = (1, n)
     return get_device_mesh(tuple(range(n)), shape)
+
+
+def memory_stats() -> Dict[str, float]:
+    """This rank's device memory stats in GB (reference MeshHostWorker
+    get_memory_stats / get_max_memory_allocated, device_mesh.py:255-270).
+    CPU-only environments report zeros."""
+    if not torch.cuda.is_available():
+        return {"allocated_gb": 0.0, "max_allocated_gb": 0.0,
+                "reserved_gb": 0.0, "total_gb": 0.0}
+    free, total = torch.cuda.mem_get_info()
+    return {
+        "allocated_gb": torch.cuda.memory_allocated() / 1e9,
+        "max_allocated_gb": torch.cuda.max_memory_allocated() / 1e9,
+        "reserved_gb": torch.cuda.memory_reserved() / 1e9,
+        "total_gb": total / 1e9,
+    }
+
+
+def reset_memory_stats() -> None:
+    """Reset the peak-memory counter (reference reset_memory_stats,
+    device_mesh.py:268)."""
+    if torch.cuda.is_available():
+        torch.cuda.reset_peak_memory_stats()

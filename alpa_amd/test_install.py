@@ -17,10 +17,11 @@ def make_batch(hidden=128, batch=4, seed=0):
 
 
 def test_shard_parallel():
-    """1-device ShardParallel runs and the loss decreases."""
+    """ShardParallel runs and the loss decreases (any world size:
+    serial `python -m alpa_amd.test_install` or under torchrun)."""
     aa.init()
     method = aa.ShardParallel(num_micro_batches=2,
-                              logical_mesh_shape=(1, 1))
+                              logical_mesh_shape=(aa.world_size(), 1))
     state, step = get_mlp_train_state_and_step(method, hidden=128)
     batch = make_batch()
     first = float(step(state, batch))
@@ -35,7 +36,7 @@ def test_grad_accumulation_consistency():
     outs = []
     for nmb in (1, 4):
         method = aa.ShardParallel(num_micro_batches=nmb,
-                                  logical_mesh_shape=(1, 1))
+                                  logical_mesh_shape=(aa.world_size(), 1))
         state, step = get_mlp_train_state_and_step(method, hidden=128)
         step(state, make_batch())
         outs.append([p.detach().clone()

@@ -113,3 +113,11 @@ def test_flop_counter_matches_analytic_gpt():
     # by the embedding-input dX and rounding of the attention term
     assert abs(measured - analytic) / analytic < 0.15, \
         (measured, analytic)
+
+
+def test_memory_stats_shape():
+    from alpa_amd.mesh import memory_stats, reset_memory_stats
+    s = memory_stats()
+    assert set(s) == {"allocated_gb", "max_allocated_gb", "reserved_gb",
+                      "total_gb"}
+    reset_memory_stats()
