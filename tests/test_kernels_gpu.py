@@ -351,3 +351,46 @@ def test_mfma32_layout_probe(ext):
     d = ext.mfma_probe32(a.contiguous(), b.contiguous())
     d_ref = a.float() @ b.float()
     torch.testing.assert_close(d, d_ref, rtol=2e-2, atol=2e-2)
+
+
+def test_ring_attention_chunk_decomposition(ext):
+    """Ring attention on GPU decomposes full attention into per-KV-chunk
+    kernel calls merged via the kernel-emitted lse (parallel/
+    ring_attention.py); single-device check: 2 chunk calls + lse merge ==
+    one full-sequence call, fwd AND bwd."""
+    from alpa_amd.parallel.ring_attention import _merge
+    torch.manual_seed(12)
+    B, Hh, S, D = 2, 4, 512, 64
+    half = S // 2
+    scale = 1.0 / math.sqrt(D)
+    q = torch.randn(B, Hh, S, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(B, Hh, S, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(B, Hh, S, D, device="cuda", dtype=torch.bfloat16)
+    o_full, lse_full = ext.attn_fwd(q, k, v, False, scale)
+    # fwd: q vs kv-chunk0 and q vs kv-chunk1, merged
+    o0, lse0 = ext.attn_fwd(q, k[:, :, :half].contiguous(),
+                            v[:, :, :half].contiguous(), False, scale)
+    o1, lse1 = ext.attn_fwd(q, k[:, :, half:].contiguous(),
+                            v[:, :, half:].contiguous(), False, scale)
+    o_m, lse_m = _merge(o0.float(), lse0.view(B, Hh, S).float(),
+                        o1, lse1.view(B, Hh, S).float())
+    torch.testing.assert_close(o_m, o_full.float(), rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(lse_m, lse_full.view(B, Hh, S).float(),
+                               rtol=1e-3, atol=1e-3)
+    # bwd: per-chunk calls with the GLOBAL o/lse sum to the full grads
+    do = torch.randn_like(o_full)
+    dq_f, dk_f, dv_f = ext.attn_bwd(do, q, k, v, o_full,
+                                    lse_full.view(B, Hh, S), False, scale)
+    lse_g = lse_full.view(B, Hh, S)
+    dq0, dk0, dv0 = ext.attn_bwd(do, q, k[:, :, :half].contiguous(),
+                                 v[:, :, :half].contiguous(), o_full,
+                                 lse_g, False, scale)
+    dq1, dk1, dv1 = ext.attn_bwd(do, q, k[:, :, half:].contiguous(),
+                                 v[:, :, half:].contiguous(), o_full,
+                                 lse_g, False, scale)
+    torch.testing.assert_close((dq0.float() + dq1.float()), dq_f.float(),
+                               rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(torch.cat([dk0, dk1], 2).float(),
+                               dk_f.float(), rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(torch.cat([dv0, dv1], 2).float(),
+                               dv_f.float(), rtol=3e-2, atol=3e-2)
