@@ -92,6 +92,45 @@ class UlyssesAttention(nn.Module):
         return self.out(o)
 
 
+class RingSelfAttention(nn.Module):
+    """Multi-head attention over a sequence-sharded residual stream using
+    RING attention (parallel/ring_attention.py) instead of Ulysses'
+    all-to-alls: heads stay whole, K/V chunks stream around the ring and
+    partials merge via lse — per-rank peak memory stays O(S/sp) even for
+    the attention working set, the memory-optimal long-context mode.
+    Drop-in interface match with UlyssesAttention ([B, S/sp, hidden])."""
+
+    def __init__(self, hidden: int, num_heads: int,
+                 mesh: Optional[DeviceMesh] = None, sp_axis: int = 1,
+                 dtype=torch.float32, device=None, layer_idx: int = 0,
+                 init_seed: int = 0):
+        super().__init__()
+        self.mesh, self.sp_axis = mesh, sp_axis
+        self.sp = mesh.axis_size(sp_axis) if mesh is not None else 1
+        self.heads = num_heads
+        self.head_dim = hidden // num_heads
+        self.qkv = ColumnParallelLinear(hidden, 3 * hidden, None, 1,
+                                        dtype=dtype, device=device,
+                                        init_seed=init_seed,
+                                        init_tag=f"sp{layer_idx}.qkv")
+        self.out = RowParallelLinear(hidden, hidden, None, 1, dtype=dtype,
+                                     device=device, init_seed=init_seed,
+                                     init_tag=f"sp{layer_idx}.out")
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """x [B, S/sp, hidden] (this rank's seq chunk, rank-ordered)."""
+        from .ring_attention import ring_attention
+        B, Sl, _ = x.shape
+        h, d = self.heads, self.head_dim
+        qkv = self.qkv(x).view(B, Sl, h, 3, d)
+        q = qkv[:, :, :, 0].permute(0, 2, 1, 3).contiguous()
+        k = qkv[:, :, :, 1].permute(0, 2, 1, 3).contiguous()
+        v = qkv[:, :, :, 2].permute(0, 2, 1, 3).contiguous()
+        o = ring_attention(q, k, v, self.mesh, self.sp_axis, causal=True)
+        o = o.permute(0, 2, 1, 3).reshape(B, Sl, h * d)
+        return self.out(o)
+
+
 def shard_sequence(x: torch.Tensor, sp: int, idx: int,
                    dim: int = 1) -> torch.Tensor:
     """Slice a full-sequence tensor to this rank's seq shard."""

@@ -60,3 +60,52 @@ def test_sp2_matches_serial():
                                    atol=1e-5)
         torch.testing.assert_close(
             xg, x.grad[:, r * per:(r + 1) * per], rtol=1e-4, atol=1e-5)
+
+
+# ---------------------------------------------------------------------------
+# Ring-attention module (context parallelism — the second SP mode)
+# ---------------------------------------------------------------------------
+
+
+def build_ring(mesh=None, sp_axis=1):
+    from alpa_amd.parallel.sequence import RingSelfAttention
+    return RingSelfAttention(HID, HEADS, mesh, sp_axis, init_seed=9)
+
+
+def test_ring_module_serial_matches_ulysses_serial():
+    """Both SP modes degenerate to the same plain attention at sp=1 (same
+    tag-seeded weights)."""
+    x = make_x()
+    y_r = build_ring()(x)
+    y_u = build()(x)
+    torch.testing.assert_close(y_r, y_u, rtol=1e-5, atol=1e-5)
+
+
+def _ring_sp_worker(rank, world_size):
+    mesh = aa.full_mesh((1, world_size))
+    m = build_ring(mesh, 1)
+    x = make_x()
+    xl = shard_sequence(x, world_size, mesh.axis_index(1))
+    xl = xl.clone().requires_grad_(True)
+    y = m(xl)
+    y.square().mean().backward()
+    return y.detach(), xl.grad, m.qkv.weight.grad
+
+
+def test_ring_sp2_matches_serial():
+    """Ring-attention SP: forward, input grads AND weight grads equal the
+    serial module under the sum-of-per-shard losses."""
+    m = build_ring()
+    x = make_x().requires_grad_(True)
+    y = m(x)
+    per = S // 2
+    total = sum(y[:, r * per:(r + 1) * per].square().mean()
+                for r in range(2))
+    total.backward()
+    results = run_distributed(_ring_sp_worker, world_size=2, timeout=300)
+    for r, (y_shard, xg, wg) in enumerate(results):
+        expect = y[:, r * per:(r + 1) * per]
+        torch.testing.assert_close(y_shard, expect.detach(), rtol=1e-4,
+                                   atol=1e-5)
+        torch.testing.assert_close(
+            xg, x.grad[:, r * per:(r + 1) * per], rtol=1e-4, atol=1e-5)
