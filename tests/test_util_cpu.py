@@ -121,3 +121,16 @@ def test_memory_stats_shape():
     assert set(s) == {"allocated_gb", "max_allocated_gb", "reserved_gb",
                       "total_gb"}
     reset_memory_stats()
+
+
+def test_benchmark_suites_wellformed():
+    from alpa_amd.benchmark_suites import ALL_SUITES, case_command
+    from alpa_amd.models.gpt import GPT_SPECS
+    for suite in ALL_SUITES.values():
+        for c in suite.values():
+            assert c.model in GPT_SPECS
+            if c.parallel == "manual":
+                dp = c.dp or c.n_gpus // c.tp
+                assert dp * c.tp == c.n_gpus, c
+            cmd = case_command(c)
+            assert f"--model {c.model}" in cmd
