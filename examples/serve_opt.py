@@ -35,14 +35,21 @@ def main():
     if args.demo or aa.world_size() > 1:
         ids = torch.randint(0, cfg.vocab_size, (1, 8), device=aa.device())
         out = model.generate(ids, max_new_tokens=8)
+        beam = model.beam_search(ids, max_new_tokens=8, num_beams=4)
         if aa.rank() == 0:
             print("generated:", out.tolist())
+            print("beam4    :", beam.tolist())
         if args.demo:
             aa.shutdown()
             return
     c = Controller()
-    c.register_model(f"opt-{args.model}",
-                     lambda ids, mt: model.generate(ids.to(aa.device()), mt))
+    def _gen(ids, mt, num_beams=1):
+        ids = ids.to(aa.device())
+        if num_beams > 1:
+            return model.beam_search(ids, mt, num_beams=num_beams)
+        return model.generate(ids, mt)
+
+    c.register_model(f"opt-{args.model}", _gen)
     run_controller(c, port=args.port)
 
 
