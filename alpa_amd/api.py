@@ -50,6 +50,9 @@ class TrainState:
         self.grad_sync: Optional[GradSynchronizer] = None
         self.engine = None  # PipelineEngine for PipeshardParallel
         self.step_count = 0
+        #: optional fp16 loss scaling (dynamic_scale.DynamicScale);
+        #: unused for the default bf16 compute dtype
+        self.dynamic_scale = None
 
     @classmethod
     def create(cls, model_fn, method: ParallelMethod, lr: float = 1e-4,
@@ -200,18 +203,29 @@ class ParallelizedFunc:
         if state.engine is not None:
             return self._pipeline_call(state, micro)
         gs = state.grad_sync
+        ds = state.dynamic_scale
         gs.zero_grads()
         total_loss = None
         for i, mb in enumerate(micro):
             gs.begin_microbatch(is_last=(i == nmb - 1))
             loss = self.fn(state.model, mb)
-            loss.backward()
+            (ds.scale_loss(loss) if ds is not None else loss).backward()
             total_loss = loss.detach() if total_loss is None \
                 else total_loss + loss.detach()
         gs.finish()
         # grad scale: mean over microbatches; all-reduce over dp was a SUM
         dp = state.mesh.axis_size(m.dp_axis) if state.mesh is not None else 1
         scale = 1.0 / (nmb * (dp if dp > 1 else 1))
+        if ds is not None:
+            # fold the loss-scale unscale into the fused AdamW launch;
+            # overflow => skip the update + back off (reference
+            # DynamicScale, model_util.py)
+            if ds.update(ds.found_inf(
+                    p.grad for p in state.model.parameters())):
+                gs.zero_grads()
+                state.step_count += 1
+                return total_loss / nmb
+            scale *= ds.unscale_factor()
         state.optimizer.step(grad_scale=scale)
         state.step_count += 1
         return total_loss / nmb

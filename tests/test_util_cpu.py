@@ -134,3 +134,38 @@ def test_benchmark_suites_wellformed():
                 assert dp * c.tp == c.n_gpus, c
             cmd = case_command(c)
             assert f"--model {c.model}" in cmd
+
+
+def test_dynamic_scale_training():
+    """fp16-style dynamic loss scaling (reference DynamicScale,
+    model_util.py): scale folds out exactly in fp32; an overflow step is
+    skipped and the scale backs off."""
+    import torch
+    import alpa_amd as aa
+    from alpa_amd.dynamic_scale import DynamicScale
+    from alpa_amd.testing import get_mlp_train_state_and_step
+
+    def make(ds):
+        method = aa.ShardParallel(num_micro_batches=1,
+                                  logical_mesh_shape=(1, 1))
+        state, step = get_mlp_train_state_and_step(method, hidden=32)
+        state.dynamic_scale = ds
+        return state, step
+
+    g = torch.Generator().manual_seed(0)
+    batch = (torch.randn(4, 32, generator=g),
+             torch.randn(4, 32, generator=g))
+    s1, step1 = make(None)
+    s2, step2 = make(DynamicScale(init_scale=2.0 ** 4))
+    step1(s1, batch)
+    step2(s2, batch)
+    for p, q in zip(s1.model.parameters(), s2.model.parameters()):
+        torch.testing.assert_close(p, q, rtol=1e-6, atol=1e-6)
+    # overflow: inf inputs -> inf grads -> skipped step, scale halved
+    ref = [p.detach().clone() for p in s2.model.parameters()]
+    bad = (torch.full((4, 32), 1e30), torch.randn(4, 32, generator=g))
+    scale_before = s2.dynamic_scale.scale
+    step2(s2, bad)
+    assert s2.dynamic_scale.scale == scale_before * 0.5
+    for p, r in zip(s2.model.parameters(), ref):
+        torch.testing.assert_close(p.detach(), r)
