@@ -174,10 +174,50 @@ class PipelineEngine:
                    dist.P2POp(dist.irecv, buf, self.prev_peer)])
         return buf
 
+    # ------------------------- tracing -------------------------
+    def enable_tracing(self, tracer=None):
+        """Record per-(microbatch x fwd/bwd) spans; dump with
+        dump_stage_execution_trace (reference
+        dump_stage_execution_trace_internal, pipeshard_executable.py:592).
+        """
+        from ..timer import Tracer
+        self.tracer = tracer or Tracer()
+        return self.tracer
+
+    def dump_stage_execution_trace(self, path: str):
+        """Write the recorded schedule execution as chrome://tracing JSON
+        (one row per stage; fwd/bwd spans labeled by microbatch)."""
+        assert getattr(self, "tracer", None) is not None,             "call enable_tracing() before the traced step"
+        self.tracer.dump_chrome_trace(path, pid=self.s)
+
+    def _trace(self, kind, mb_idx):
+        import time as _t
+
+        class _Span:
+
+            def __init__(self, eng):
+                self.eng = eng
+
+            def __enter__(self):
+                self.t0 = _t.perf_counter()
+
+            def __exit__(self, *a):
+                tr = getattr(self.eng, "tracer", None)
+                if tr is not None:
+                    tr.log_span(f"stage{self.eng.s}.{kind}.mb{mb_idx}",
+                                self.t0, _t.perf_counter(), cat=kind,
+                                tid=self.eng.s)
+
+        return _Span(self)
+
     # ------------------------- compute steps -------------------------
     def _forward_step(self, mb_idx: int, x: Optional[torch.Tensor],
                       microbatches: List[Any]):
         """Returns (input, output): output is loss on the last stage."""
+        with self._trace("fwd", mb_idx):
+            return self._forward_step_inner(mb_idx, x, microbatches)
+
+    def _forward_step_inner(self, mb_idx, x, microbatches):
         if not self.is_first:
             x = x.requires_grad_(True)
         mb = microbatches[mb_idx]
@@ -190,6 +230,11 @@ class PipelineEngine:
 
     def _backward_step(self, x, out, out_grad,
                        is_last_bwd: bool) -> Optional[torch.Tensor]:
+        with self._trace("bwd", "x"):
+            return self._backward_step_inner(x, out, out_grad, is_last_bwd)
+
+    def _backward_step_inner(self, x, out, out_grad,
+                             is_last_bwd: bool) -> Optional[torch.Tensor]:
         if self.grad_sync is not None:
             self.grad_sync.begin_microbatch(is_last=is_last_bwd)
         if self.is_last:

@@ -344,3 +344,34 @@ def test_local_pipeline_runner_matches_serial():
             if p.grad is not None and ref_params[n].grad is not None:
                 torch.testing.assert_close(p.grad, ref_params[n].grad,
                                            rtol=1e-5, atol=1e-5)
+
+
+def _traced_pp_worker(rank, world_size, tmpdir):
+    import json
+    import os
+    method = aa.PipeshardParallel(num_micro_batches=2,
+                                  num_stages=world_size,
+                                  stage_mesh_shape=(1, 1))
+    spec = gpt_pipeline_spec(CFG)
+    spec.build_stage = _stage_builder
+    state = aa.TrainState.create(spec, method, lr=1e-3)
+    state.engine.enable_tracing()
+    step = aa.parallelize(lambda m, b: None, method=method)
+    step(state, make_batch(0))
+    path = os.path.join(tmpdir, f"trace_{rank}.json")
+    state.engine.dump_stage_execution_trace(path)
+    with open(path) as f:
+        ev = json.load(f)["traceEvents"]
+    return [(e["name"], e["cat"]) for e in ev]
+
+
+def test_stage_execution_trace(tmp_path):
+    """Per-stage chrome-trace dump of the 1F1B execution (reference
+    dump_stage_execution_trace, pipeshard_executable.py:592)."""
+    results = run_distributed(_traced_pp_worker, world_size=2,
+                              args=(str(tmp_path),), timeout=300)
+    for r, events in enumerate(results):
+        fwd = [n for n, c in events if c == "fwd"]
+        bwd = [n for n, c in events if c == "bwd"]
+        assert len(fwd) == 2 and len(bwd) == 2, events
+        assert all(n.startswith(f"stage{r}.") for n, _ in events)
