@@ -37,6 +37,12 @@ class GPTConfig:
     ffn_mult: int = 4
     layernorm_eps: float = 1e-5
     tie_embeddings: bool = False
+    #: rematerialize each block's activations in backward (reference
+    #: automatic_remat, layer_construction.py:571 — remat at layer
+    #: boundaries).  288 GB HBM3E rarely needs it; for very long seq or
+    #: 70B-scale single-node runs it trades ~33% recompute for O(L) less
+    #: activation memory
+    remat: bool = False
 
     @property
     def head_dim(self) -> int:
@@ -170,6 +176,18 @@ class Block(nn.Module):
         return h2, self.mlp(y2)
 
 
+def _run_block(blk, res, delta, remat: bool):
+    """One transformer block, optionally under activation rematerialization
+    (torch.utils.checkpoint, non-reentrant): the block's internals are
+    recomputed during backward — the layer-boundary remat the reference
+    inserts at pipeline-layer slices (remat_sliced_eqns)."""
+    if remat and torch.is_grad_enabled() and res.requires_grad:
+        from torch.utils.checkpoint import checkpoint
+        return checkpoint(blk.forward_fused, res, delta,
+                          use_reentrant=False)
+    return blk.forward_fused(res, delta)
+
+
 class GPTModel(nn.Module):
     """Full GPT LM. `mesh`/`axis` give the tensor-parallel axis (axis 1 of a
     (dp, tp) mesh); dp replication is handled by the trainer."""
@@ -207,7 +225,7 @@ class GPTModel(nn.Module):
         res = self.wte(ids) + self.wpe[:S]
         delta = None
         for blk in self.blocks:
-            res, delta = blk.forward_fused(res, delta)
+            res, delta = _run_block(blk, res, delta, self.cfg.remat)
         _, x = ops.add_layer_norm(res, delta, self.ln_f.weight,
                                   self.ln_f.bias, self.ln_f.eps)
         return self.lm_head(x)
@@ -283,7 +301,7 @@ class GPTStage(nn.Module):
             x = self.wte(ids) + self.wpe[:S]
         res, delta = x, None
         for blk in self.blocks.children():
-            res, delta = blk.forward_fused(res, delta)
+            res, delta = _run_block(blk, res, delta, self.cfg.remat)
         if not self.is_last:
             # materialize the stream value at the stage boundary
             return res + delta if delta is not None else res

@@ -50,6 +50,9 @@ def parse_args():
     p.add_argument("--parallel", choices=["auto", "manual"], default="auto",
                    help="auto = ILP auto-sharding picks (dp, tp)")
     p.add_argument("--seq", type=int, default=1024)
+    p.add_argument("--remat", action="store_true",
+                   help="activation remat at block boundaries (TFLOPS "
+                        "accounting then uses the reference's factor 96)")
     p.add_argument("--no-hipgraph", action="store_true",
                    help="disable hipGraph step capture (single-GPU only)")
     return p.parse_args()
@@ -82,6 +85,7 @@ def main():
         model_name = "2.6B" if on_gpu else "125M"
     if on_gpu:
         cfg = gpt_config(model_name, seq_len=args.seq)
+        cfg.remat = args.remat
         batch_per_gpu = args.batch_per_gpu
     else:
         # CPU smoke: tiny shapes so the script works without a GPU
@@ -185,7 +189,7 @@ def main():
 
     ms_per_step = elapsed / args.steps * 1000
     global_batch = batch_per_gpu * dp
-    tflops_step = model_tflops_per_step(cfg, global_batch)
+    tflops_step = model_tflops_per_step(cfg, global_batch, args.remat)
     value = tflops_step / (elapsed / args.steps)  # aggregate TFLOPS
     baseline_aggregate_tflops = 296.0  # 37.01 TF/GPU x 8 V100
 

@@ -96,3 +96,26 @@ def test_gpt_dp2_matches_serial():
     for i in range(STEPS):
         avg = sum(r[i] for r in results) / 2
         assert abs(avg - serial[i]) < 2e-4, (results, serial)
+
+
+def test_remat_matches_no_remat():
+    """Block-boundary activation remat (reference automatic_remat,
+    layer_construction.py:571) is numerically identical in loss AND
+    grads — only memory/time differ."""
+    from alpa_amd.models.gpt import GPTConfig, GPTModel
+    cfg_a = GPTConfig(hidden_size=64, num_layers=2, num_heads=4,
+                      seq_len=32, vocab_size=96)
+    cfg_b = GPTConfig(hidden_size=64, num_layers=2, num_heads=4,
+                      seq_len=32, vocab_size=96, remat=True)
+    ids = torch.randint(0, 96, (2, 32))
+    ma = GPTModel(cfg_a, None, 1, torch.float32, None, init_seed=3)
+    mb = GPTModel(cfg_b, None, 1, torch.float32, None, init_seed=3)
+    la = ma.loss(ids, ids)
+    lb = mb.loss(ids, ids)
+    torch.testing.assert_close(la, lb)
+    la.backward()
+    lb.backward()
+    for (n, pa), (_, pb) in zip(ma.named_parameters(),
+                                mb.named_parameters()):
+        torch.testing.assert_close(pa.grad, pb.grad, rtol=1e-6, atol=1e-6,
+                                   msg=lambda m: f"{n}: {m}")
