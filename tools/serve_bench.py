@@ -1,0 +1,97 @@
+"""Serving benchmark: OPT prefill latency + decode throughput.
+
+Measures the reference's llm_serving headline quantities (examples/
+llm_serving: generation throughput of OPT on a mesh) on MI355X: random
+init weights (no network), bf16, KV-cache decode through the gfx950
+attention kernel.
+
+  python tools/serve_bench.py --model 30B --batch 8 --prompt 128 --gen 64
+  torchrun --standalone --nproc-per-node N tools/serve_bench.py --model 66B ...
+
+Prints one JSON line: prefill_ms, decode_ms_per_token, decode_tokens_per_s
+(aggregate over the batch), model, batch, tp.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                ".."))
+import alpa_amd as aa
+from alpa_amd.models.opt import OPTModel, opt_config
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default="1.3B")
+    p.add_argument("--batch", type=int, default=8)
+    p.add_argument("--prompt", type=int, default=128)
+    p.add_argument("--gen", type=int, default=64)
+    p.add_argument("--beams", type=int, default=1)
+    args = p.parse_args()
+
+    aa.init()
+    mesh = aa.full_mesh((1, aa.world_size()))
+    on_gpu = torch.cuda.is_available()
+    dtype = torch.bfloat16 if on_gpu else torch.float32
+    cfg = opt_config(args.model,
+                     max_seq_len=args.prompt + args.gen + args.batch)
+    model = OPTModel(cfg, mesh, 1, dtype, aa.device(), init_seed=0)
+
+    ids = torch.randint(0, cfg.vocab_size, (args.batch, args.prompt),
+                        device=aa.device())
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    # warmup (prefill + a few decodes)
+    with torch.no_grad():
+        model.generate(ids, max_new_tokens=4)
+    sync()
+
+    # prefill timing
+    cache = model.new_cache(args.batch)
+    sync()
+    t0 = time.perf_counter()
+    with torch.no_grad():
+        logits = model.forward_step(ids, cache)
+    sync()
+    prefill_ms = (time.perf_counter() - t0) * 1e3
+
+    # decode timing
+    tok = model.greedy_token(logits).unsqueeze(1)
+    sync()
+    t0 = time.perf_counter()
+    with torch.no_grad():
+        for _ in range(args.gen):
+            logits = model.forward_step(tok, cache)
+            tok = model.greedy_token(logits).unsqueeze(1)
+    sync()
+    decode_s = time.perf_counter() - t0
+
+    if args.beams > 1:
+        with torch.no_grad():
+            model.beam_search(ids, max_new_tokens=8, num_beams=args.beams)
+        sync()
+
+    if aa.rank() == 0:
+        print(json.dumps({
+            "model": f"OPT-{args.model}", "batch": args.batch,
+            "tp": aa.world_size(), "prompt_len": args.prompt,
+            "gen_tokens": args.gen, "dtype": str(dtype).split(".")[-1],
+            "prefill_ms": round(prefill_ms, 2),
+            "decode_ms_per_token": round(decode_s / args.gen * 1e3, 3),
+            "decode_tokens_per_s": round(args.batch * args.gen / decode_s,
+                                         1),
+            "data": "synthetic/random-init",
+        }))
+    aa.shutdown()
+
+
+if __name__ == "__main__":
+    main()
