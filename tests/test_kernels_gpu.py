@@ -321,6 +321,15 @@ def test_bert_and_unet_on_gpu():
     loss.backward()
     assert float(loss) == float(loss)
 
+    from alpa_amd.models.vit import ViTConfig, ViTModel
+    vcfg = ViTConfig(image_size=64, patch_size=16, hidden_size=128,
+                     num_layers=2, num_heads=4, num_classes=10)
+    vit = ViTModel(vcfg, dtype=torch.bfloat16, device="cuda", init_seed=2)
+    img = torch.randn(2, 3, 64, 64, device="cuda", dtype=torch.bfloat16)
+    loss = vit.loss(img, torch.randint(0, 10, (2,), device="cuda"))
+    loss.backward()
+    assert float(loss) == float(loss)
+
 
 def test_zero_paths_single_gpu():
     """ZeRO-2/3 optimizer paths through the fused kernels (dp=1 degenerate:
