@@ -8,7 +8,8 @@ hand-written CDNA4 (gfx950) HIP kernels + RCCL collectives over xGMI, one
 process per GPU.  No JAX/XLA, no Ray, no CUDA shims.
 """
 
-from .api import TrainState, init, parallelize, shutdown
+from .api import (TrainState, grad, init, parallelize, shutdown,
+                  value_and_grad)
 from .serialization import (restore_checkpoint, restore_train_state,
                             save_checkpoint, save_train_state)
 from .global_env import global_config

@@ -263,3 +263,34 @@ def parallelize(fn: Optional[Callable] = None, *,
     if fn is None:
         return wrap
     return wrap(fn)
+
+
+def value_and_grad(fn: Callable):
+    """Functional helper mirroring the reference's ``alpa.value_and_grad``
+    (api.py:241): ``vg(model, batch) -> (loss, {name: grad})`` without
+    touching ``.grad`` accumulation state (grads come from autograd.grad).
+    The decorated training step (`parallelize`) does NOT need this; it
+    exists for eval/diagnostic code written in the reference's style."""
+
+    def vg(model: torch.nn.Module, batch):
+        params = [p for p in model.parameters() if p.requires_grad]
+        loss = fn(model, batch)
+        grads = torch.autograd.grad(loss, params, allow_unused=True)
+        named = {}
+        it = iter(grads)
+        for n, p in model.named_parameters():
+            if p.requires_grad:
+                named[n] = next(it)
+        return loss.detach(), named
+
+    return vg
+
+
+def grad(fn: Callable):
+    """``alpa.grad`` analog (api.py:241): returns only the grads dict."""
+    _vg = value_and_grad(fn)
+
+    def g(model, batch):
+        return _vg(model, batch)[1]
+
+    return g

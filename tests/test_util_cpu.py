@@ -169,3 +169,20 @@ def test_dynamic_scale_training():
     assert s2.dynamic_scale.scale == scale_before * 0.5
     for p, r in zip(s2.model.parameters(), ref):
         torch.testing.assert_close(p.detach(), r)
+
+
+def test_value_and_grad():
+    """aa.value_and_grad/grad mirror the reference api.py:241 surface."""
+    import torch
+    import alpa_amd as aa
+    from alpa_amd.testing import MLPModel
+    m = MLPModel(hidden=16)
+    batch = (torch.randn(2, 16), torch.randn(2, 16))
+    vg = aa.value_and_grad(lambda mod, b: mod.loss(*b))
+    loss, grads = vg(m, batch)
+    assert not loss.requires_grad
+    assert set(grads) == {n for n, p in m.named_parameters()
+                          if p.requires_grad}
+    g = aa.grad(lambda mod, b: mod.loss(*b))(m, batch)
+    for n in grads:
+        torch.testing.assert_close(g[n], grads[n])
