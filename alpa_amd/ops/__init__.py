@@ -128,16 +128,17 @@ def bias_gelu(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
 class _FlashAttention(torch.autograd.Function):
 
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale):
+    def forward(ctx, q, k, v, causal, scale, alibi=None):
         if scale is None:
             scale = 1.0 / math.sqrt(q.shape[-1])
         if use_hip(q):
-            o, lse = hip_ops().attn_fwd(q, k, v, causal, scale)
+            o, lse = hip_ops().attn_fwd(q, k, v, causal, scale, alibi)
         else:
-            o, lse = ref.attention_fwd(q, k, v, causal, scale)
+            o, lse = ref.attention_fwd(q, k, v, causal, scale, alibi)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
         ctx.scale = scale
+        ctx.alibi = alibi
         return o
 
     @staticmethod
@@ -145,23 +146,26 @@ class _FlashAttention(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         if use_hip(q):
             dq, dk, dv = hip_ops().attn_bwd(do.contiguous(), q, k, v, o, lse,
-                                            ctx.causal, ctx.scale)
+                                            ctx.causal, ctx.scale, ctx.alibi)
         else:
             dq, dk, dv = ref.attention_bwd(do, q, k, v, o, lse, ctx.causal,
-                                           ctx.scale)
-        return dq, dk, dv, None, None
+                                           ctx.scale, ctx.alibi)
+        return dq, dk, dv, None, None, None
 
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     causal: bool = True,
-                    scale: Optional[float] = None) -> torch.Tensor:
+                    scale: Optional[float] = None,
+                    alibi_slopes: Optional[torch.Tensor] = None
+                    ) -> torch.Tensor:
     """Fused attention. q,k,v: [B, heads, S, head_dim] (bf16 on GPU).
 
     Online-softmax tiling — never materializes the S x S score matrix
     (reference's compiled modules get this from XLA fusion; here it is the
-    hand-written gfx950 kernel).
+    hand-written gfx950 kernel).  alibi_slopes [heads]
+    (fp32) adds the BLOOM ALiBi bias inside the kernel.
     """
-    return _FlashAttention.apply(q, k, v, causal, scale)
+    return _FlashAttention.apply(q, k, v, causal, scale, alibi_slopes)
 
 
 class _FlashAttentionQKV(torch.autograd.Function):

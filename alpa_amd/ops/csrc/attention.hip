@@ -38,7 +38,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int H, int S, int Skv, int D, float scale,
-    int causal, AttnStrides st) {
+    int causal, const float* __restrict__ alibi, AttnStrides st) {
   // 8 waves x 16 q-rows = 128 q rows per block; K/V tiles of 64 kv are
   // double-buffered in LDS with register-prefetched staging (loads for
   // tile t+1 issue before computing tile t and land after it — the HBM
@@ -99,6 +99,10 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
       causal ? min(Skv, qb * (16 * NW) + 16 * NW) : Skv;
   const int n_tiles = (kv_limit + ATTN_BLOCK_K - 1) / ATTN_BLOCK_K;
   const int my_q = q_row0 + lo;  // this lane's q row
+  // ALiBi (BLOOM): per-head slope, bias = slope * (kv_pos - q_pos); the
+  // q GLOBAL position is my_q + (Skv - S) for cached decode
+  const float al_slope = (alibi != nullptr) ? alibi[bh % H] : 0.f;
+  const int al_qoff = Skv - S;
 
   // staging: load tile -> regs (two kv rows per thread so the V transpose
   // writes pair as b32)
@@ -210,7 +214,8 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int kv_idx = kvb + nt * 16 + hi * 4 + r;
-        float sv = s_acc[nt][r] * scale;
+        float sv = fmaf(al_slope, (float)(kv_idx - my_q - al_qoff),
+                        s_acc[nt][r] * scale);
         bool masked = (kv_idx >= Skv) || (causal && kv_idx > my_q) ||
                       (my_q >= S);
         sv = masked ? -INFINITY : sv;
@@ -314,7 +319,7 @@ __global__ __launch_bounds__(ATTN32_THREADS) void attn_fwd_kernel32(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int H, int S, int Skv, int D, float scale,
-    int causal, AttnStrides st) {
+    int causal, const float* __restrict__ alibi, AttnStrides st) {
   constexpr int KS_QK = Dp / 16;       // k-steps over head dim (K=16)
   constexpr int MT = 2;                // 2 kv m-tiles of 32 per 64-kv tile
   constexpr int DT = Dp / 32;          // d tiles of 32 (O^T rows)
@@ -337,6 +342,8 @@ __global__ __launch_bounds__(ATTN32_THREADS) void attn_fwd_kernel32(
   short* op = o + batch * st.ob + head * st.oh;
   const int q_row0 = qb * (32 * NW) + wave * 32;
   const int my_q = q_row0 + lo;
+  const float al_slope = (alibi != nullptr) ? alibi[head] : 0.f;
+  const int al_qoff = Skv - S;
 
   __shared__ short k_lds[2][64][Dp + LP];
   __shared__ short vt_lds[2][Dp][64 + LP];
@@ -438,7 +445,8 @@ __global__ __launch_bounds__(ATTN32_THREADS) void attn_fwd_kernel32(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int kv_idx = kvb + mt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float sv = s_acc[mt][r] * scale;
+        float sv = fmaf(al_slope, (float)(kv_idx - my_q - al_qoff),
+                        s_acc[mt][r] * scale);
         bool masked = (kv_idx >= Skv) || (causal && kv_idx > my_q) ||
                       (my_q >= S);
         sv = masked ? -INFINITY : sv;
@@ -531,7 +539,7 @@ hipError_t launch_attn_fwd_ablate(const void* q, const void* k,
   if (abl == A) {                                                           \
     attn_fwd_kernel<96, A><<<grid, block, 0, stream>>>(                     \
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,  \
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);               \
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr, st);      \
   }
   ABL_CASE(0) ABL_CASE(1) ABL_CASE(2) ABL_CASE(3) ABL_CASE(4)
 #undef ABL_CASE
@@ -554,15 +562,15 @@ hipError_t launch_attn_fwd_v2(const void* q, const void* k, const void* v,
   if (D <= 64) {
     attn_fwd_kernel32<64><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr, st);
   } else if (D <= 96) {
     attn_fwd_kernel32<96><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr, st);
   } else if (D <= 128) {
     attn_fwd_kernel32<128><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr, st);
   } else {
     return hipErrorInvalidValue;
   }
@@ -572,8 +580,8 @@ hipError_t launch_attn_fwd_v2(const void* q, const void* k, const void* v,
 hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
                            void* o, float* lse, int64_t B, int64_t H,
                            int64_t S, int64_t Skv, int64_t D, float scale,
-                           int causal, const int64_t* strides,
-                           hipStream_t stream) {
+                           int causal, const float* alibi,
+                           const int64_t* strides, hipStream_t stream) {
   dim3 grid((uint32_t)ceil_div(S, 16 * (ATTN_THREADS / 64)), (uint32_t)(B * H));
   dim3 block(ATTN_THREADS);
   AttnStrides st;
@@ -592,15 +600,15 @@ hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
   if (D <= 64) {
     attn_fwd_kernel<64><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
   } else if (D <= 96) {
     attn_fwd_kernel<96><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
   } else if (D <= 128) {
     attn_fwd_kernel<128><<<grid, block, 0, stream>>>(
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, st);
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
   } else {
     return hipErrorInvalidValue;
   }
@@ -724,7 +732,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dq, int H, int S, int Skv, int D, float scale,
-    int causal, AttnBwdStrides st) {
+    int causal, const float* __restrict__ alibi, AttnBwdStrides st) {
   // 8 waves x 16 q rows = 128 q rows per block; K/V/K^T staged per kv tile
   // with register-prefetched loads (issue before compute, write after the
   // barrier) and paired-row transposes.
@@ -779,6 +787,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     lse_r[r] = (qi < S) ? lse[(int64_t)bh * S + qi] : 0.f;
     delta_r[r] = (qi < S) ? delta[(int64_t)bh * S + qi] : 0.f;
   }
+  const float al_slope = (alibi != nullptr) ? alibi[bh % H] : 0.f;
 
   f32x4 dq_acc[DT];
 #pragma unroll
@@ -870,7 +879,10 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
         int q_idx = q_row0 + hi * 4 + r;
         bool masked = (kv_idx >= Skv) || (causal && kv_idx > q_idx) ||
                       (q_idx >= S);
-        float pv = masked ? 0.f : __expf(s_acc[nt][r] * scale - lse_r[r]);
+        float pv = masked ? 0.f
+                          : __expf(fmaf(al_slope, (float)(kv_idx - q_idx),
+                                        s_acc[nt][r] * scale) -
+                                   lse_r[r]);
         float ds = pv * (dp_acc[nt][r] - delta_r[r]) * scale;
         p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(ds);
       }
@@ -925,7 +937,8 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dk, short* __restrict__ dv, int H, int S, int Skv,
-    int D, float scale, int causal, AttnBwdStrides st) {
+    int D, float scale, int causal, const float* __restrict__ alibi,
+    AttnBwdStrides st) {
   // 8 waves x 16 kv rows = 128 kv rows per block; Q/dO tiles (row-major +
   // transposed) staged per q tile with register-prefetched loads and
   // paired-row transposes — the staging cost is the dominant term here and
@@ -952,6 +965,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
   short* dkp = dk + batch * st.dkb + head * st.dkh;
   short* dvp = dv + batch * st.dvb + head * st.dvh;
   const int kv_row0 = kvb_idx * (16 * NW) + wave * 16;
+  const float al_slope = (alibi != nullptr) ? alibi[bh % H] : 0.f;
 
   __shared__ short q_lds[64][Dp + LP];
   __shared__ short qt_lds[Dp][64 + LP];
@@ -1101,7 +1115,9 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
         bool masked = (q_idx >= S) || (kv_idx >= Skv) ||
                       (causal && kv_idx > q_idx);
         float pv = masked ? 0.f
-                          : __expf(st_acc[nt][r] * scale -
+                          : __expf(fmaf(al_slope,
+                                        (float)(kv_idx - q_idx),
+                                        st_acc[nt][r] * scale) -
                                    lse_lds[nt * 16 + lo]);
         p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(pv);
         dst_pk[nt][r] =
@@ -1205,7 +1221,7 @@ hipError_t launch_attn_bwd_dkv_ablate(const void* q, const void* k,
     attn_bwd_dkv_kernel<96, A><<<grid_kv, block, 0, stream>>>(             \
         (const short*)q, (const short*)k, (const short*)v,                 \
         (const short*)dout, lse, delta, (short*)dk, (short*)dv, (int)H,    \
-        (int)S, (int)Skv, (int)D, scale, causal, st);
+        (int)S, (int)Skv, (int)D, scale, causal, nullptr, st);
   DKV_CASE(0) DKV_CASE(1) DKV_CASE(2) DKV_CASE(3)
 #undef DKV_CASE
   return hipGetLastError();
@@ -1216,8 +1232,8 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
                            const float* lse, float* delta_ws, void* dq,
                            void* dk, void* dv, int64_t B, int64_t H,
                            int64_t S, int64_t Skv, int64_t D, float scale,
-                           int causal, const int64_t* strides,
-                           hipStream_t stream) {
+                           int causal, const float* alibi,
+                           const int64_t* strides, hipStream_t stream) {
   // strides layout: q(3) k(3) v(3) do(3) dq(3) dk(3) dv(3) o(3)
   AttnBwdStrides st;
   const int64_t* p = strides;
@@ -1258,11 +1274,11 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
     attn_bwd_dq_kernel<DP><<<grid_q, block, 0, side_stream>>>(               \
         (const short*)q, (const short*)k, (const short*)v,                   \
         (const short*)dout, lse, delta_ws, (short*)dq, (int)H, (int)S,       \
-        (int)Skv, (int)D, scale, causal, st);                                \
+        (int)Skv, (int)D, scale, causal, alibi, st);                         \
     attn_bwd_dkv_kernel<DP><<<grid_kv, block, 0, stream>>>(                  \
         (const short*)q, (const short*)k, (const short*)v,                   \
         (const short*)dout, lse, delta_ws, (short*)dk, (short*)dv, (int)H,   \
-        (int)S, (int)Skv, (int)D, scale, causal, st);                        \
+        (int)S, (int)Skv, (int)D, scale, causal, alibi, st);                 \
   } while (0)
   if (D <= 64) {
     LAUNCH_BWD(64);

@@ -49,7 +49,8 @@ hipError_t launch_ce_bwd(const float*, const void*, const int64_t*,
                          const float*, void*, int64_t, int64_t, hipStream_t);
 hipError_t launch_attn_fwd(const void*, const void*, const void*, void*,
                            float*, int64_t, int64_t, int64_t, int64_t,
-                           int64_t, float, int, const int64_t*, hipStream_t);
+                           int64_t, float, int, const float*,
+                           const int64_t*, hipStream_t);
 hipError_t launch_attn_fwd_v2(const void*, const void*, const void*, void*,
                               float*, int64_t, int64_t, int64_t, int64_t,
                               int64_t, float, int, const int64_t*,
@@ -70,8 +71,8 @@ hipError_t launch_attn_bwd_dkv_ablate(const void*, const void*,
 hipError_t launch_attn_bwd(const void*, const void*, const void*,
                            const void*, const void*, const float*, float*,
                            void*, void*, void*, int64_t, int64_t, int64_t,
-                           int64_t, int64_t, float, int, const int64_t*,
-                           hipStream_t);
+                           int64_t, int64_t, float, int, const float*,
+                           const int64_t*, hipStream_t);
 }
 
 namespace {
@@ -291,9 +292,22 @@ void check_bf16_strided4(const at::Tensor& t, const char* name) {
 // q/k/v/o are logical [B, H, S, D] views with arbitrary B/H/S strides and a
 // contiguous D — so the packed qkv layout [B, S, heads, 3*D] feeds the
 // kernel directly, with no permute/contiguous copies on the hot path.
+
+// ALiBi slopes: optional per-head fp32 device tensor [H]; nullptr = off
+static const float* alibi_ptr(const c10::optional<at::Tensor>& alibi,
+                              int64_t H) {
+  if (!alibi.has_value() || !alibi->defined()) return nullptr;
+  TORCH_CHECK(alibi->scalar_type() == at::kFloat &&
+                  alibi->is_contiguous() && alibi->numel() == H,
+              "alibi slopes must be contiguous fp32 [H]");
+  return (const float*)alibi->const_data_ptr();
+}
+
 at::Tensor attn_fwd_out(const at::Tensor& q, const at::Tensor& k,
                         const at::Tensor& v, at::Tensor o, at::Tensor lse,
-                        bool causal, double scale) {
+                        bool causal, double scale,
+                        const c10::optional<at::Tensor>& alibi =
+                            c10::nullopt) {
   check_bf16_strided4(q, "q");
   check_bf16_strided4(k, "k");
   check_bf16_strided4(v, "v");
@@ -309,8 +323,8 @@ at::Tensor attn_fwd_out(const at::Tensor& q, const at::Tensor& k,
   HIP_OK(launch_attn_fwd(q.const_data_ptr(), k.const_data_ptr(),
                          v.const_data_ptr(), o.mutable_data_ptr(),
                          (float*)lse.mutable_data_ptr(), B, H, S, Skv, D,
-                         (float)scale, causal ? 1 : 0, strides,
-                         cur_stream()));
+                         (float)scale, causal ? 1 : 0, alibi_ptr(alibi, H),
+                         strides, cur_stream()));
   return o;
 }
 
@@ -353,11 +367,13 @@ std::vector<at::Tensor> attn_fwd_v2(const at::Tensor& q,
 
 std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                  const at::Tensor& v, bool causal,
-                                 double scale) {
+                                 double scale,
+                                 const c10::optional<at::Tensor>& alibi =
+                                     c10::nullopt) {
   int64_t B = q.size(0), H = q.size(1), S = q.size(2);
   auto o = at::empty_like(q.contiguous());
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
-  attn_fwd_out(q, k, v, o, lse, causal, scale);
+  attn_fwd_out(q, k, v, o, lse, causal, scale, alibi);
   return {o, lse};
 }
 
@@ -411,7 +427,8 @@ std::vector<at::Tensor> attn_bwd_blocked(const at::Tensor& dout_, const at::Tens
 void attn_bwd_out(const at::Tensor& dout, const at::Tensor& q,
                   const at::Tensor& k, const at::Tensor& v,
                   const at::Tensor& o, const at::Tensor& lse, at::Tensor dq,
-                  at::Tensor dk, at::Tensor dv, bool causal, double scale) {
+                  at::Tensor dk, at::Tensor dv, bool causal, double scale,
+                  const c10::optional<at::Tensor>& alibi = c10::nullopt) {
   check_bf16_strided4(q, "q");
   check_bf16_strided4(dout, "dout");
   check_bf16_strided4(dq, "dq");
@@ -435,18 +452,20 @@ void attn_bwd_out(const at::Tensor& dout, const at::Tensor& q,
                          (float*)delta.mutable_data_ptr(),
                          dq.mutable_data_ptr(), dk.mutable_data_ptr(),
                          dv.mutable_data_ptr(), B, H, S, Skv, D,
-                         (float)scale, causal ? 1 : 0, strides,
-                         cur_stream()));
+                         (float)scale, causal ? 1 : 0, alibi_ptr(alibi, H),
+                         strides, cur_stream()));
 }
 
 std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& q,
                                  const at::Tensor& k, const at::Tensor& v,
                                  const at::Tensor& o, const at::Tensor& lse,
-                                 bool causal, double scale) {
+                                 bool causal, double scale,
+                                 const c10::optional<at::Tensor>& alibi =
+                                     c10::nullopt) {
   auto dq = at::empty_like(q.contiguous());
   auto dk = at::empty_like(k.contiguous());
   auto dv = at::empty_like(v.contiguous());
-  attn_bwd_out(dout, q, k, v, o, lse, dq, dk, dv, causal, scale);
+  attn_bwd_out(dout, q, k, v, o, lse, dq, dk, dv, causal, scale, alibi);
   return {dq, dk, dv};
 }
 
@@ -505,13 +524,25 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adamw_step_raw", &adamw_step_raw, "multi-tensor AdamW (gfx950)");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE fwd (gfx950)");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE bwd (gfx950)");
-  m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)");
-  m.def("attn_fwd_out", &attn_fwd_out, "flash attention fwd, strided out");
+  m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"),
+        py::arg("scale"), py::arg("alibi") = py::none());
+  m.def("attn_fwd_out", &attn_fwd_out, "flash attention fwd, strided out",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"),
+        py::arg("lse"), py::arg("causal"), py::arg("scale"),
+        py::arg("alibi") = py::none());
   m.def("attn_fwd_v2", &attn_fwd_v2, "32x32-MFMA fwd experiment");
   m.def("attn_fwd_ablate", &attn_fwd_ablate, "ablation variants (perf)");
   m.def("attn_bwd_dkv_ablate", &attn_bwd_dkv_ablate, "dkv ablation (perf)");
-  m.def("attn_bwd", &attn_bwd, "flash attention bwd (gfx950 MFMA)");
-  m.def("attn_bwd_out", &attn_bwd_out, "flash attention bwd, strided out");
+  m.def("attn_bwd", &attn_bwd, "flash attention bwd (gfx950 MFMA)",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("o"), py::arg("lse"), py::arg("causal"), py::arg("scale"),
+        py::arg("alibi") = py::none());
+  m.def("attn_bwd_out", &attn_bwd_out, "flash attention bwd, strided out",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("o"), py::arg("lse"), py::arg("dq"), py::arg("dk"),
+        py::arg("dv"), py::arg("causal"), py::arg("scale"),
+        py::arg("alibi") = py::none());
   m.def("attn_bwd_blocked", &attn_bwd_blocked,
         "attention bwd (blocked hipBLASLt reference)");
   m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 layout probe");

@@ -103,13 +103,26 @@ def bias_gelu_bwd(dy: torch.Tensor, x: torch.Tensor, bias: torch.Tensor
     return dx.to(x.dtype), dbias
 
 
+def _alibi_bias(alibi, Hh, S, Skv, device):
+    """[1, Hh, S, Skv] ALiBi bias: slope_h * (kv_pos - q_pos); q global
+    position is offset by Skv - S (cached decode)."""
+    qpos = torch.arange(S, device=device, dtype=torch.float32) + (Skv - S)
+    kpos = torch.arange(Skv, device=device, dtype=torch.float32)
+    rel = kpos.view(1, 1, 1, Skv) - qpos.view(1, 1, S, 1)
+    return alibi.float().view(1, Hh, 1, 1) * rel
+
+
 def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
-                  causal: bool = True, softmax_scale: float | None = None
+                  causal: bool = True, softmax_scale: float | None = None,
+                  alibi: torch.Tensor | None = None
                   ) -> Tuple[torch.Tensor, torch.Tensor]:
-    """q,k,v: [B, Hh, S, D]. Returns (o, lse[B,Hh,S]) in fp32 math."""
+    """q,k,v: [B, Hh, S, D]. Returns (o, lse[B,Hh,S]) in fp32 math.
+    alibi: optional per-head slopes [Hh] (BLOOM bias)."""
     B, Hh, S, D = q.shape
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(D)
     s = torch.matmul(_up(q), _up(k).transpose(-1, -2)) * scale
+    if alibi is not None:
+        s = s + _alibi_bias(alibi, Hh, S, k.shape[2], q.device)
     if causal:
         mask = torch.triu(torch.ones(S, k.shape[2], dtype=torch.bool,
                                      device=q.device), diagonal=1)
@@ -124,12 +137,15 @@ def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 
 def attention_bwd(do: torch.Tensor, q: torch.Tensor, k: torch.Tensor,
                   v: torch.Tensor, o: torch.Tensor, lse: torch.Tensor,
-                  causal: bool = True, softmax_scale: float | None = None
+                  causal: bool = True, softmax_scale: float | None = None,
+                  alibi: torch.Tensor | None = None
                   ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     B, Hh, S, D = q.shape
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(D)
     qf, kf, vf, dof = _up(q), _up(k), _up(v), _up(do)
     s = torch.matmul(qf, kf.transpose(-1, -2)) * scale
+    if alibi is not None:
+        s = s + _alibi_bias(alibi, Hh, S, k.shape[2], q.device)
     if causal:
         mask = torch.triu(torch.ones(S, k.shape[2], dtype=torch.bool,
                                      device=q.device), diagonal=1)
