@@ -22,11 +22,20 @@ import torch
 sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
                                 ".."))
 import alpa_amd as aa
+from alpa_amd.models.bloom import BloomModel, bloom_config
+from alpa_amd.models.codegen import CodeGenModel, codegen_config
 from alpa_amd.models.opt import OPTModel, opt_config
+
+FAMILIES = {
+    "opt": (OPTModel, opt_config),
+    "bloom": (BloomModel, bloom_config),
+    "codegen": (CodeGenModel, codegen_config),
+}
 
 
 def main():
     p = argparse.ArgumentParser()
+    p.add_argument("--family", default="opt", choices=sorted(FAMILIES))
     p.add_argument("--model", default="1.3B")
     p.add_argument("--batch", type=int, default=8)
     p.add_argument("--prompt", type=int, default=128)
@@ -38,9 +47,10 @@ def main():
     mesh = aa.full_mesh((1, aa.world_size()))
     on_gpu = torch.cuda.is_available()
     dtype = torch.bfloat16 if on_gpu else torch.float32
-    cfg = opt_config(args.model,
-                     max_seq_len=args.prompt + args.gen + args.batch)
-    model = OPTModel(cfg, mesh, 1, dtype, aa.device(), init_seed=0)
+    model_cls, config_fn = FAMILIES[args.family]
+    cfg = config_fn(args.model,
+                    max_seq_len=args.prompt + args.gen + args.batch)
+    model = model_cls(cfg, mesh, 1, dtype, aa.device(), init_seed=0)
 
     ids = torch.randint(0, cfg.vocab_size, (args.batch, args.prompt),
                         device=aa.device())
@@ -81,7 +91,7 @@ def main():
 
     if aa.rank() == 0:
         print(json.dumps({
-            "model": f"OPT-{args.model}", "batch": args.batch,
+            "model": f"{args.family}-{args.model}", "batch": args.batch,
             "tp": aa.world_size(), "prompt_len": args.prompt,
             "gen_tokens": args.gen, "dtype": str(dtype).split(".")[-1],
             "prefill_ms": round(prefill_ms, 2),
