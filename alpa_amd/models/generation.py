@@ -1,4 +1,4 @@
-"""Shared autoregressive generation: greedy + beam search.
+"""Shared autoregressive generation: greedy, sampling + beam search.
 
 Capability analog of the reference's huggingface GenerationMixin adapter
 (``examples/llm_serving/model/wrapper.py:501`` wraps alpa executables in
@@ -45,6 +45,32 @@ class GenerationMixin:
         idx = gathered[best, torch.arange(best.shape[0],
                                           device=best.device), 1]
         return idx.long()
+
+    @torch.no_grad()
+    def sample_token(self, logits: torch.Tensor, temperature: float = 1.0,
+                     top_k: int = 0, top_p: float = 1.0,
+                     generator: Optional[torch.Generator] = None
+                     ) -> torch.Tensor:
+        """Temperature / top-k / top-p sampling, vocab-parallel-safe: the
+        candidate set is the global top-max(top_k, 64) (covering all mass
+        that survives top-p in practice); every TP rank draws the SAME
+        sample (shared generator seed), so SPMD decode stays in lockstep.
+        """
+        kk = max(top_k if top_k > 0 else 0, 64)
+        lp, idx = self._log_probs_topk(logits.float() / max(temperature,
+                                                            1e-6), kk)
+        if top_k > 0:
+            lp = lp[:, :top_k]
+            idx = idx[:, :top_k]
+        probs = torch.softmax(lp, dim=-1)
+        if top_p < 1.0:
+            cum = probs.cumsum(dim=-1)
+            # keep the smallest prefix reaching top_p (always >= 1 token)
+            cut = (cum - probs) >= top_p
+            probs = probs.masked_fill(cut, 0.0)
+            probs = probs / probs.sum(dim=-1, keepdim=True)
+        choice = torch.multinomial(probs, 1, generator=generator)
+        return idx.gather(1, choice)[:, 0]
 
     @torch.no_grad()
     def _log_probs_topk(self, logits: torch.Tensor, k: int):
@@ -139,16 +165,28 @@ class GenerationMixin:
 
     @torch.no_grad()
     def generate(self, prompt_ids: torch.Tensor, max_new_tokens: int,
-                 eos_token: Optional[int] = None) -> torch.Tensor:
-        """Greedy generation: prefill + cached decode loop.  prompt_ids
-        [B, S0]; returns [B, S0 + max_new_tokens]."""
+                 eos_token: Optional[int] = None,
+                 do_sample: bool = False, temperature: float = 1.0,
+                 top_k: int = 0, top_p: float = 1.0,
+                 generator: Optional[torch.Generator] = None
+                 ) -> torch.Tensor:
+        """Greedy (default) or sampled generation: prefill + cached decode
+        loop.  prompt_ids [B, S0]; returns [B, S0 + max_new_tokens].
+        With do_sample, pass the SAME generator seed on every TP rank."""
         B, S0 = prompt_ids.shape
         cache = self.new_cache(B)
+
+        def pick(lg):
+            if do_sample:
+                return self.sample_token(lg, temperature, top_k, top_p,
+                                         generator)
+            return self.greedy_token(lg)
+
         logits = self.forward_step(prompt_ids, cache)
-        toks = [self.greedy_token(logits)]
+        toks = [pick(logits)]
         for _ in range(max_new_tokens - 1):
             logits = self.forward_step(toks[-1].unsqueeze(1), cache)
-            toks.append(self.greedy_token(logits))
+            toks.append(pick(logits))
             if eos_token is not None and bool((toks[-1] == eos_token).all()):
                 break
         return torch.cat([prompt_ids] + [t.unsqueeze(1) for t in toks],

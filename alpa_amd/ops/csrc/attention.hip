@@ -33,7 +33,7 @@ struct AttnStrides {
 // ABL: ablation level for perf diagnosis (0 = full kernel; higher skips
 // later phases; asm keep-alives prevent dead-code elimination of earlier
 // phases — guide methodology rule 17)
-template <int Dp, int ABL = 0>
+template <int Dp, int ABL = 0, bool AL = false>
 __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
@@ -101,7 +101,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   const int my_q = q_row0 + lo;  // this lane's q row
   // ALiBi (BLOOM): per-head slope, bias = slope * (kv_pos - q_pos); the
   // q GLOBAL position is my_q + (Skv - S) for cached decode
-  const float al_slope = (alibi != nullptr) ? alibi[bh % H] : 0.f;
+  const float al_slope = AL ? alibi[bh % H] : 0.f;
   const int al_qoff = Skv - S;
 
   // staging: load tile -> regs (two kv rows per thread so the V transpose
@@ -214,8 +214,9 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int kv_idx = kvb + nt * 16 + hi * 4 + r;
-        float sv = fmaf(al_slope, (float)(kv_idx - my_q - al_qoff),
-                        s_acc[nt][r] * scale);
+        float sv = s_acc[nt][r] * scale;
+        if (AL)  // compile-time: the non-ALiBi instantiation is untouched
+          sv = fmaf(al_slope, (float)(kv_idx - my_q - al_qoff), sv);
         bool masked = (kv_idx >= Skv) || (causal && kv_idx > my_q) ||
                       (my_q >= S);
         sv = masked ? -INFINITY : sv;
@@ -314,7 +315,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
 // ===========================================================================
 #define ATTN32_THREADS 1024
 
-template <int Dp>
+template <int Dp, bool AL = false>
 __global__ __launch_bounds__(ATTN32_THREADS) void attn_fwd_kernel32(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
@@ -342,7 +343,7 @@ __global__ __launch_bounds__(ATTN32_THREADS) void attn_fwd_kernel32(
   short* op = o + batch * st.ob + head * st.oh;
   const int q_row0 = qb * (32 * NW) + wave * 32;
   const int my_q = q_row0 + lo;
-  const float al_slope = (alibi != nullptr) ? alibi[head] : 0.f;
+  const float al_slope = AL ? alibi[head] : 0.f;
   const int al_qoff = Skv - S;
 
   __shared__ short k_lds[2][64][Dp + LP];
@@ -445,8 +446,8 @@ __global__ __launch_bounds__(ATTN32_THREADS) void attn_fwd_kernel32(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         int kv_idx = kvb + mt * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
-        float sv = fmaf(al_slope, (float)(kv_idx - my_q - al_qoff),
-                        s_acc[mt][r] * scale);
+        float sv = s_acc[mt][r] * scale;
+        if (AL) sv = fmaf(al_slope, (float)(kv_idx - my_q - al_qoff), sv);
         bool masked = (kv_idx >= Skv) || (causal && kv_idx > my_q) ||
                       (my_q >= S);
         sv = masked ? -INFINITY : sv;
@@ -598,17 +599,35 @@ hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
   st.oh = strides[10];
   st.os = strides[11];
   if (D <= 64) {
-    attn_fwd_kernel<64><<<grid, block, 0, stream>>>(
-        (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
+    if (alibi)
+      attn_fwd_kernel<64, 0, true><<<grid, block, 0, stream>>>(
+          (const short*)q, (const short*)k, (const short*)v, (short*)o,
+          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
+    else
+      attn_fwd_kernel<64><<<grid, block, 0, stream>>>(
+          (const short*)q, (const short*)k, (const short*)v, (short*)o,
+          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr,
+          st);
   } else if (D <= 96) {
-    attn_fwd_kernel<96><<<grid, block, 0, stream>>>(
-        (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
+    if (alibi)
+      attn_fwd_kernel<96, 0, true><<<grid, block, 0, stream>>>(
+          (const short*)q, (const short*)k, (const short*)v, (short*)o,
+          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
+    else
+      attn_fwd_kernel<96><<<grid, block, 0, stream>>>(
+          (const short*)q, (const short*)k, (const short*)v, (short*)o,
+          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr,
+          st);
   } else if (D <= 128) {
-    attn_fwd_kernel<128><<<grid, block, 0, stream>>>(
-        (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
+    if (alibi)
+      attn_fwd_kernel<128, 0, true><<<grid, block, 0, stream>>>(
+          (const short*)q, (const short*)k, (const short*)v, (short*)o,
+          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
+    else
+      attn_fwd_kernel<128><<<grid, block, 0, stream>>>(
+          (const short*)q, (const short*)k, (const short*)v, (short*)o,
+          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr,
+          st);
   } else {
     return hipErrorInvalidValue;
   }
@@ -726,7 +745,7 @@ __global__ void attn_bwd_preprocess_kernel(const short* __restrict__ dout,
 //   dP = dO V^T        (A=dO regs, B=V_lds row-major)
 //   dS = P*(dP-delta)*scale
 //   dQ += dS K         (A=dS via p_lds, B=Kt_lds transposed)
-template <int Dp>
+template <int Dp, bool AL = false>
 __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -787,7 +806,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     lse_r[r] = (qi < S) ? lse[(int64_t)bh * S + qi] : 0.f;
     delta_r[r] = (qi < S) ? delta[(int64_t)bh * S + qi] : 0.f;
   }
-  const float al_slope = (alibi != nullptr) ? alibi[bh % H] : 0.f;
+  const float al_slope = AL ? alibi[bh % H] : 0.f;
 
   f32x4 dq_acc[DT];
 #pragma unroll
@@ -879,10 +898,9 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
         int q_idx = q_row0 + hi * 4 + r;
         bool masked = (kv_idx >= Skv) || (causal && kv_idx > q_idx) ||
                       (q_idx >= S);
-        float pv = masked ? 0.f
-                          : __expf(fmaf(al_slope, (float)(kv_idx - q_idx),
-                                        s_acc[nt][r] * scale) -
-                                   lse_r[r]);
+        float sraw = s_acc[nt][r] * scale;
+        if (AL) sraw = fmaf(al_slope, (float)(kv_idx - q_idx), sraw);
+        float pv = masked ? 0.f : __expf(sraw - lse_r[r]);
         float ds = pv * (dp_acc[nt][r] - delta_r[r]) * scale;
         p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(ds);
       }
@@ -931,7 +949,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
 //   dP^T = V dO^T      (A=V regs, B=dO_lds row-major)
 //   dS^T = P^T*(dP^T - delta[q])*scale
 //   dK += dS^T Q       (A=dS^T via p_lds, B=Qt_lds)
-template <int Dp, int ABL = 0>
+template <int Dp, int ABL = 0, bool AL = false>
 __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -965,7 +983,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
   short* dkp = dk + batch * st.dkb + head * st.dkh;
   short* dvp = dv + batch * st.dvb + head * st.dvh;
   const int kv_row0 = kvb_idx * (16 * NW) + wave * 16;
-  const float al_slope = (alibi != nullptr) ? alibi[bh % H] : 0.f;
+  const float al_slope = AL ? alibi[bh % H] : 0.f;
 
   __shared__ short q_lds[64][Dp + LP];
   __shared__ short qt_lds[Dp][64 + LP];
@@ -1114,11 +1132,9 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
         int kv_idx = kv_row0 + hi * 4 + r;
         bool masked = (q_idx >= S) || (kv_idx >= Skv) ||
                       (causal && kv_idx > q_idx);
-        float pv = masked ? 0.f
-                          : __expf(fmaf(al_slope,
-                                        (float)(kv_idx - q_idx),
-                                        st_acc[nt][r] * scale) -
-                                   lse_lds[nt * 16 + lo]);
+        float sraw = st_acc[nt][r] * scale;
+        if (AL) sraw = fmaf(al_slope, (float)(kv_idx - q_idx), sraw);
+        float pv = masked ? 0.f : __expf(sraw - lse_lds[nt * 16 + lo]);
         p_lds[wave][hi * 4 + r][nt * 16 + lo] = f2bf(pv);
         dst_pk[nt][r] =
             f2bf(pv * (dpt_acc[nt][r] - delta_lds[nt * 16 + lo]) * scale);
@@ -1269,16 +1285,21 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
               (uint32_t)(B * H));
   dim3 grid_kv((uint32_t)ceil_div(Skv, 16 * (ATTN_BWD_THREADS / 64)),
                (uint32_t)(B * H));
-#define LAUNCH_BWD(DP)                                                       \
+#define LAUNCH_BWD_T(DP, ALB, ALP)                                         \
   do {                                                                       \
-    attn_bwd_dq_kernel<DP><<<grid_q, block, 0, side_stream>>>(               \
+    attn_bwd_dq_kernel<DP, ALB><<<grid_q, block, 0, side_stream>>>(          \
         (const short*)q, (const short*)k, (const short*)v,                   \
         (const short*)dout, lse, delta_ws, (short*)dq, (int)H, (int)S,       \
-        (int)Skv, (int)D, scale, causal, alibi, st);                         \
-    attn_bwd_dkv_kernel<DP><<<grid_kv, block, 0, stream>>>(                  \
+        (int)Skv, (int)D, scale, causal, ALP, st);                           \
+    attn_bwd_dkv_kernel<DP, 0, ALB><<<grid_kv, block, 0, stream>>>(          \
         (const short*)q, (const short*)k, (const short*)v,                   \
         (const short*)dout, lse, delta_ws, (short*)dk, (short*)dv, (int)H,   \
-        (int)S, (int)Skv, (int)D, scale, causal, alibi, st);                 \
+        (int)S, (int)Skv, (int)D, scale, causal, ALP, st);                   \
+  } while (0)
+#define LAUNCH_BWD(DP)                                                       \
+  do {                                                                       \
+    if (alibi) LAUNCH_BWD_T(DP, true, alibi);                                \
+    else LAUNCH_BWD_T(DP, false, nullptr);                                   \
   } while (0)
   if (D <= 64) {
     LAUNCH_BWD(64);

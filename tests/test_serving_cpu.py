@@ -308,3 +308,45 @@ def test_tp2_bloom_matches_serial():
     results = run_distributed(_tp_bloom_worker, world_size=2, timeout=300)
     for r in results:
         assert torch.equal(torch.as_tensor(r), serial)
+
+
+def test_sampling_top_k_top_p():
+    """sample_token respects top-k/top-p supports and temperature; greedy
+    is the temperature->0 limit."""
+    torch.manual_seed(15)
+    m = build_opt()
+    logits = torch.randn(4, CFG.vocab_size) * 3
+    g = torch.Generator().manual_seed(0)
+    for _ in range(5):
+        t = m.sample_token(logits, top_k=5, generator=g)
+        ref_top5 = logits.topk(5, -1).indices
+        for b in range(4):
+            assert t[b] in ref_top5[b]
+    # top_p=tiny keeps only the argmax
+    t = m.sample_token(logits, top_p=1e-9, generator=g)
+    assert torch.equal(t, logits.argmax(-1))
+    # near-zero temperature concentrates on the argmax
+    t = m.sample_token(logits, temperature=1e-5, generator=g)
+    assert torch.equal(t, logits.argmax(-1))
+
+
+def _tp_sample_worker(rank, world_size):
+    mesh = aa.mesh.full_mesh((1, world_size))
+    m = build_opt(mesh, axis=1)
+    torch.manual_seed(16)
+    ids = torch.randint(0, CFG.vocab_size, (2, 5))
+    g = torch.Generator().manual_seed(99)
+    return m.generate(ids, 5, do_sample=True, top_k=10, generator=g)
+
+
+def test_tp2_sampling_matches_serial():
+    """Sampled decode under TP: identical candidate sets + shared RNG keep
+    every rank (and the serial oracle) in lockstep."""
+    m = build_opt()
+    torch.manual_seed(16)
+    ids = torch.randint(0, CFG.vocab_size, (2, 5))
+    g = torch.Generator().manual_seed(99)
+    serial = m.generate(ids, 5, do_sample=True, top_k=10, generator=g)
+    results = run_distributed(_tp_sample_worker, world_size=2, timeout=300)
+    for r in results:
+        assert torch.equal(torch.as_tensor(r), serial)
