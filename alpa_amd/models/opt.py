@@ -147,6 +147,35 @@ class OPTBlock(nn.Module):
         x = x + self.fc2(h)
         return x
 
+    @torch.no_grad()
+    def _attn_decode_varlen(self, x, cache_k, cache_v, lens):
+        """One token per slot at PER-SLOT positions (continuous
+        batching): scatter k/v at lens[b], attend with kv_lens=lens+1."""
+        B, S, _ = x.shape
+        assert S == 1
+        h, d = self.heads_per_rank, self.head_dim
+        qkv = self.qkv(x).view(B, 1, h, 3, d)
+        q = qkv[:, :, :, 0].permute(0, 2, 1, 3)
+        k = qkv[:, :, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, :, 2].permute(0, 2, 1, 3)
+        bidx = torch.arange(B, device=x.device)
+        cache_k[bidx, :, lens] = k[:, :, 0]
+        cache_v[bidx, :, lens] = v[:, :, 0]
+        total = int(lens.max()) + 1
+        o = ops.flash_attention_varlen(q.contiguous(),
+                                       cache_k[:, :, :total],
+                                       cache_v[:, :, :total], lens + 1)
+        o = o.permute(0, 2, 1, 3).reshape(B, 1, h * d)
+        return self.out(o)
+
+    @torch.no_grad()
+    def forward_decode_varlen(self, x, cache_k, cache_v, lens):
+        x = x + self._attn_decode_varlen(self.ln1(x), cache_k, cache_v,
+                                         lens)
+        h = self.fc1(self.ln2(x))
+        h = torch.nn.functional.relu(h)
+        return x + self.fc2(h)
+
 
 class OPTModel(nn.Module, GenerationMixin):
     """TP-sharded OPT decoder with generation support (greedy/beam from
@@ -204,4 +233,16 @@ class OPTModel(nn.Module, GenerationMixin):
         cache.length += S
         x = self.ln_f(x[:, -1:])
         return self.lm_head(x)[:, 0]
+
+    @torch.no_grad()
+    def forward_decode(self, tok: torch.Tensor, cache: KVCache,
+                       lens: torch.Tensor) -> torch.Tensor:
+        """Varlen decode (continuous batching): tok [B, 1] next token per
+        slot, lens [B] tokens already cached per slot.  Returns logits
+        [B, vocab/tp].  The caller owns per-slot length bookkeeping
+        (cache.length is unused on this path)."""
+        x = self.wte(tok) + self.wpe[2 + lens].unsqueeze(1)
+        for i, blk in enumerate(self.blocks):
+            x = blk.forward_decode_varlen(x, cache.k[i], cache.v[i], lens)
+        return self.lm_head(self.ln_f(x))[:, 0]
 

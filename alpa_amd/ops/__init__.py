@@ -168,6 +168,28 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return _FlashAttention.apply(q, k, v, causal, scale, alibi_slopes)
 
 
+@torch.no_grad()
+def flash_attention_varlen(q: torch.Tensor, k: torch.Tensor,
+                           v: torch.Tensor, kv_lens: torch.Tensor,
+                           scale: Optional[float] = None,
+                           alibi_slopes: Optional[torch.Tensor] = None
+                           ) -> torch.Tensor:
+    """Decode attention with PER-BATCH KV lengths (continuous batching):
+    slot b's queries attend to k[b, :, :kv_lens[b]].  Inference-only
+    (no backward); the gfx950 kernel masks per element past each slot's
+    length."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    lens = kv_lens.to(dtype=torch.int32)
+    if use_hip(q):
+        o, _ = hip_ops().attn_fwd(q.contiguous(), k.contiguous(),
+                                  v.contiguous(), False, scale,
+                                  alibi_slopes, lens.contiguous())
+        return o
+    o, _ = ref.attention_fwd(q, k, v, False, scale, alibi_slopes, lens)
+    return o
+
+
 class _FlashAttentionQKV(torch.autograd.Function):
     """Packed-qkv attention: input [B, S, h*3d] (per-head [q|k|v] layout),
     output [B, S, h*d].  The strided gfx950 kernel reads/writes these

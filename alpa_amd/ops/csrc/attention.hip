@@ -33,12 +33,13 @@ struct AttnStrides {
 // ABL: ablation level for perf diagnosis (0 = full kernel; higher skips
 // later phases; asm keep-alives prevent dead-code elimination of earlier
 // phases — guide methodology rule 17)
-template <int Dp, int ABL = 0, bool AL = false>
+template <int Dp, int ABL = 0, bool AL = false, bool VL = false>
 __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int H, int S, int Skv, int D, float scale,
-    int causal, const float* __restrict__ alibi, AttnStrides st) {
+    int causal, const float* __restrict__ alibi,
+    const int* __restrict__ kv_lens, AttnStrides st) {
   // 8 waves x 16 q-rows = 128 q rows per block; K/V tiles of 64 kv are
   // double-buffered in LDS with register-prefetched staging (loads for
   // tile t+1 issue before computing tile t and land after it — the HBM
@@ -101,8 +102,11 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   const int my_q = q_row0 + lo;  // this lane's q row
   // ALiBi (BLOOM): per-head slope, bias = slope * (kv_pos - q_pos); the
   // q GLOBAL position is my_q + (Skv - S) for cached decode
+  // varlen (continuous batching): this batch slot's real kv length;
+  // tiles past it are masked per element (Skv stays the tile loop bound)
+  const int my_skv = VL ? kv_lens[batch] : Skv;
   const float al_slope = AL ? alibi[bh % H] : 0.f;
-  const int al_qoff = Skv - S;
+  const int al_qoff = my_skv - S;
 
   // staging: load tile -> regs (two kv rows per thread so the V transpose
   // writes pair as b32)
@@ -217,7 +221,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
         float sv = s_acc[nt][r] * scale;
         if (AL)  // compile-time: the non-ALiBi instantiation is untouched
           sv = fmaf(al_slope, (float)(kv_idx - my_q - al_qoff), sv);
-        bool masked = (kv_idx >= Skv) || (causal && kv_idx > my_q) ||
+        bool masked = (kv_idx >= my_skv) || (causal && kv_idx > my_q) ||
                       (my_q >= S);
         sv = masked ? -INFINITY : sv;
         s_acc[nt][r] = sv;
@@ -540,7 +544,8 @@ hipError_t launch_attn_fwd_ablate(const void* q, const void* k,
   if (abl == A) {                                                           \
     attn_fwd_kernel<96, A><<<grid, block, 0, stream>>>(                     \
         (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,  \
-        (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr, st);      \
+        (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr, nullptr, \
+        st);                                                                \
   }
   ABL_CASE(0) ABL_CASE(1) ABL_CASE(2) ABL_CASE(3) ABL_CASE(4)
 #undef ABL_CASE
@@ -582,9 +587,21 @@ hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
                            void* o, float* lse, int64_t B, int64_t H,
                            int64_t S, int64_t Skv, int64_t D, float scale,
                            int causal, const float* alibi,
-                           const int64_t* strides, hipStream_t stream) {
+                           const int* kv_lens, const int64_t* strides,
+                           hipStream_t stream) {
   dim3 grid((uint32_t)ceil_div(S, 16 * (ATTN_THREADS / 64)), (uint32_t)(B * H));
   dim3 block(ATTN_THREADS);
+#define FWD_VARIANT(DP, ALB, VLB, ALP, VLP)                                \
+  attn_fwd_kernel<DP, 0, ALB, VLB><<<grid, block, 0, stream>>>(            \
+      (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,   \
+      (int)H, (int)S, (int)Skv, (int)D, scale, causal, ALP, VLP, st)
+#define FWD_DISPATCH(DP)                                                   \
+  do {                                                                     \
+    if (alibi && kv_lens) FWD_VARIANT(DP, true, true, alibi, kv_lens);     \
+    else if (alibi) FWD_VARIANT(DP, true, false, alibi, nullptr);          \
+    else if (kv_lens) FWD_VARIANT(DP, false, true, nullptr, kv_lens);      \
+    else FWD_VARIANT(DP, false, false, nullptr, nullptr);                  \
+  } while (0)
   AttnStrides st;
   st.qb = strides[0];
   st.qh = strides[1];
@@ -599,38 +616,16 @@ hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
   st.oh = strides[10];
   st.os = strides[11];
   if (D <= 64) {
-    if (alibi)
-      attn_fwd_kernel<64, 0, true><<<grid, block, 0, stream>>>(
-          (const short*)q, (const short*)k, (const short*)v, (short*)o,
-          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
-    else
-      attn_fwd_kernel<64><<<grid, block, 0, stream>>>(
-          (const short*)q, (const short*)k, (const short*)v, (short*)o,
-          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr,
-          st);
+    FWD_DISPATCH(64);
   } else if (D <= 96) {
-    if (alibi)
-      attn_fwd_kernel<96, 0, true><<<grid, block, 0, stream>>>(
-          (const short*)q, (const short*)k, (const short*)v, (short*)o,
-          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
-    else
-      attn_fwd_kernel<96><<<grid, block, 0, stream>>>(
-          (const short*)q, (const short*)k, (const short*)v, (short*)o,
-          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr,
-          st);
+    FWD_DISPATCH(96);
   } else if (D <= 128) {
-    if (alibi)
-      attn_fwd_kernel<128, 0, true><<<grid, block, 0, stream>>>(
-          (const short*)q, (const short*)k, (const short*)v, (short*)o,
-          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, alibi, st);
-    else
-      attn_fwd_kernel<128><<<grid, block, 0, stream>>>(
-          (const short*)q, (const short*)k, (const short*)v, (short*)o,
-          lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr,
-          st);
+    FWD_DISPATCH(128);
   } else {
     return hipErrorInvalidValue;
   }
+#undef FWD_DISPATCH
+#undef FWD_VARIANT
   return hipGetLastError();
 }
 

@@ -49,7 +49,7 @@ hipError_t launch_ce_bwd(const float*, const void*, const int64_t*,
                          const float*, void*, int64_t, int64_t, hipStream_t);
 hipError_t launch_attn_fwd(const void*, const void*, const void*, void*,
                            float*, int64_t, int64_t, int64_t, int64_t,
-                           int64_t, float, int, const float*,
+                           int64_t, float, int, const float*, const int*,
                            const int64_t*, hipStream_t);
 hipError_t launch_attn_fwd_v2(const void*, const void*, const void*, void*,
                               float*, int64_t, int64_t, int64_t, int64_t,
@@ -303,10 +303,23 @@ static const float* alibi_ptr(const c10::optional<at::Tensor>& alibi,
   return (const float*)alibi->const_data_ptr();
 }
 
+// kv_lens: optional per-batch int32 [B] real KV lengths (continuous
+// batching decode); elements past a slot's length are masked
+static const int* kv_lens_ptr(const c10::optional<at::Tensor>& kv_lens,
+                              int64_t B) {
+  if (!kv_lens.has_value() || !kv_lens->defined()) return nullptr;
+  TORCH_CHECK(kv_lens->scalar_type() == at::kInt &&
+                  kv_lens->is_contiguous() && kv_lens->numel() == B,
+              "kv_lens must be contiguous int32 [B]");
+  return (const int*)kv_lens->const_data_ptr();
+}
+
 at::Tensor attn_fwd_out(const at::Tensor& q, const at::Tensor& k,
                         const at::Tensor& v, at::Tensor o, at::Tensor lse,
                         bool causal, double scale,
                         const c10::optional<at::Tensor>& alibi =
+                            c10::nullopt,
+                        const c10::optional<at::Tensor>& kv_lens =
                             c10::nullopt) {
   check_bf16_strided4(q, "q");
   check_bf16_strided4(k, "k");
@@ -324,7 +337,7 @@ at::Tensor attn_fwd_out(const at::Tensor& q, const at::Tensor& k,
                          v.const_data_ptr(), o.mutable_data_ptr(),
                          (float*)lse.mutable_data_ptr(), B, H, S, Skv, D,
                          (float)scale, causal ? 1 : 0, alibi_ptr(alibi, H),
-                         strides, cur_stream()));
+                         kv_lens_ptr(kv_lens, B), strides, cur_stream()));
   return o;
 }
 
@@ -369,11 +382,13 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                  const at::Tensor& v, bool causal,
                                  double scale,
                                  const c10::optional<at::Tensor>& alibi =
+                                     c10::nullopt,
+                                 const c10::optional<at::Tensor>& kv_lens =
                                      c10::nullopt) {
   int64_t B = q.size(0), H = q.size(1), S = q.size(2);
   auto o = at::empty_like(q.contiguous());
   auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
-  attn_fwd_out(q, k, v, o, lse, causal, scale, alibi);
+  attn_fwd_out(q, k, v, o, lse, causal, scale, alibi, kv_lens);
   return {o, lse};
 }
 
@@ -526,11 +541,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE bwd (gfx950)");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("causal"),
-        py::arg("scale"), py::arg("alibi") = py::none());
+        py::arg("scale"), py::arg("alibi") = py::none(),
+        py::arg("kv_lens") = py::none());
   m.def("attn_fwd_out", &attn_fwd_out, "flash attention fwd, strided out",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("o"),
         py::arg("lse"), py::arg("causal"), py::arg("scale"),
-        py::arg("alibi") = py::none());
+        py::arg("alibi") = py::none(), py::arg("kv_lens") = py::none());
   m.def("attn_fwd_v2", &attn_fwd_v2, "32x32-MFMA fwd experiment");
   m.def("attn_fwd_ablate", &attn_fwd_ablate, "ablation variants (perf)");
   m.def("attn_bwd_dkv_ablate", &attn_bwd_dkv_ablate, "dkv ablation (perf)");

@@ -114,15 +114,31 @@ def _alibi_bias(alibi, Hh, S, Skv, device):
 
 def attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                   causal: bool = True, softmax_scale: float | None = None,
-                  alibi: torch.Tensor | None = None
+                  alibi: torch.Tensor | None = None,
+                  kv_lens: torch.Tensor | None = None
                   ) -> Tuple[torch.Tensor, torch.Tensor]:
     """q,k,v: [B, Hh, S, D]. Returns (o, lse[B,Hh,S]) in fp32 math.
-    alibi: optional per-head slopes [Hh] (BLOOM bias)."""
+    alibi: optional per-head slopes [Hh] (BLOOM bias).
+    kv_lens: optional int [B] per-batch real KV lengths (varlen decode:
+    positions >= kv_lens[b] are masked; alibi q position offsets per
+    batch)."""
     B, Hh, S, D = q.shape
     scale = softmax_scale if softmax_scale is not None else 1.0 / math.sqrt(D)
     s = torch.matmul(_up(q), _up(k).transpose(-1, -2)) * scale
+    Skv = k.shape[2]
     if alibi is not None:
-        s = s + _alibi_bias(alibi, Hh, S, k.shape[2], q.device)
+        if kv_lens is not None:
+            kpos = torch.arange(Skv, device=q.device, dtype=torch.float32)
+            qpos = (kv_lens.to(q.device).float().view(B, 1, 1, 1) - S +
+                    torch.arange(S, device=q.device,
+                                 dtype=torch.float32).view(1, 1, S, 1))
+            s = s + alibi.float().view(1, Hh, 1, 1) * (
+                kpos.view(1, 1, 1, Skv) - qpos)
+        else:
+            s = s + _alibi_bias(alibi, Hh, S, Skv, q.device)
+    if kv_lens is not None:
+        mask = torch.arange(Skv, device=q.device).view(1, 1, 1, Skv) >=             kv_lens.to(q.device).view(B, 1, 1, 1)
+        s = s.masked_fill(mask, float("-inf"))
     if causal:
         mask = torch.triu(torch.ones(S, k.shape[2], dtype=torch.bool,
                                      device=q.device), diagonal=1)
