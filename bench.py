@@ -43,7 +43,11 @@ def parse_args():
     p.add_argument("--model", type=str, default="auto",
                    help="GPT spec name (125M..76B) or 'auto'")
     p.add_argument("--batch-per-gpu", type=int, default=32)
-    p.add_argument("--nmb", type=int, default=4, help="num micro batches")
+    p.add_argument("--nmb", type=int, default=0,
+                   help="num micro batches; 0 = auto (1: without a "
+                        "pipeline, microbatching only shrinks GEMMs — "
+                        "measured 760 vs 738 TF — and grad-sync overlap "
+                        "spans the whole backward)")
     p.add_argument("--dp", type=int, default=0, help="data-parallel degree "
                    "(0 = all GPUs)")
     p.add_argument("--tp", type=int, default=1, help="tensor-parallel degree")
