@@ -59,7 +59,10 @@ def parse_args():
                         "accounting then uses the reference's factor 96)")
     p.add_argument("--no-hipgraph", action="store_true",
                    help="disable hipGraph step capture (single-GPU only)")
-    return p.parse_args()
+    args = p.parse_args()
+    if args.nmb == 0:
+        args.nmb = 1
+    return args
 
 
 def _enable_tunableop():
