@@ -143,10 +143,9 @@ class CodeGenModel(nn.Module, GenerationMixin):
                  init_seed: int = 0):
         super().__init__()
         self.cfg = cfg
-        assert cfg.head_dim <= 128, (
-            f"head_dim {cfg.head_dim} > 128: the gfx950 attention kernel "
-            "tiles head_dim in registers up to 128 (split-D is a planned "
-            "extension); pick a config with head_dim <= 128")
+        assert cfg.head_dim <= 256, (
+            f"head_dim {cfg.head_dim} > 256 unsupported; dims in "
+            "(128, 256] use the blocked hipBLASLt attention path")
         self.mesh, self.axis = mesh, axis
         tp = mesh.axis_size(axis) if mesh is not None else 1
         self.heads_per_rank = cfg.num_heads // tp

@@ -132,7 +132,14 @@ class _FlashAttention(torch.autograd.Function):
         if scale is None:
             scale = 1.0 / math.sqrt(q.shape[-1])
         if use_hip(q):
-            o, lse = hip_ops().attn_fwd(q, k, v, causal, scale, alibi)
+            if q.shape[-1] > 128:
+                # fused kernel tiles head_dim in registers up to 128;
+                # bigger heads (CodeGen-6B/16B: 256) take the blocked
+                # hipBLASLt path (no alibi there — BLOOM heads are <=128)
+                assert alibi is None, "alibi requires head_dim <= 128"
+                o, lse = hip_ops().attn_fwd_blocked(q, k, v, causal, scale)
+            else:
+                o, lse = hip_ops().attn_fwd(q, k, v, causal, scale, alibi)
         else:
             o, lse = ref.attention_fwd(q, k, v, causal, scale, alibi)
         ctx.save_for_backward(q, k, v, o, lse)
@@ -145,8 +152,13 @@ class _FlashAttention(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         if use_hip(q):
-            dq, dk, dv = hip_ops().attn_bwd(do.contiguous(), q, k, v, o, lse,
-                                            ctx.causal, ctx.scale, ctx.alibi)
+            if q.shape[-1] > 128:
+                dq, dk, dv = hip_ops().attn_bwd_blocked(
+                    do.contiguous(), q, k, v, o, lse, ctx.causal, ctx.scale)
+            else:
+                dq, dk, dv = hip_ops().attn_bwd(
+                    do.contiguous(), q, k, v, o, lse, ctx.causal, ctx.scale,
+                    ctx.alibi)
         else:
             dq, dk, dv = ref.attention_bwd(do, q, k, v, o, lse, ctx.causal,
                                            ctx.scale, ctx.alibi)

@@ -11,7 +11,9 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+import os
+sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                ".."))
 from alpa_amd.ops._backend import hip_ops
 
 
