@@ -107,7 +107,8 @@ def profiled_stage_search(num_devices: int, num_microbatches: int,
                           boundary_act_bytes: float = 0.0,
                           layer_param_bytes: Optional[Sequence[float]] = None,
                           db=None, cluster_key: str = "mi355x",
-                          max_stages: Optional[int] = None
+                          max_stages: Optional[int] = None,
+                          memory_budget: Optional[float] = None
                           ) -> Tuple[int, List[Tuple[int, int]],
                                      List[Tuple[int, int]], float]:
     """Profile-guided inter-op search (reference training_dp,
@@ -125,6 +126,13 @@ def profiled_stage_search(num_devices: int, num_microbatches: int,
     GEMMs down the measured efficiency curve (small GEMMs run far below
     peak), which an analytic flops/peak model cannot see — this is the
     closed profiling loop of SURVEY.md §5.1.
+
+    ``memory_budget`` (bytes/device) adds the reference's memory
+    feasibility (max_n_succ_stages, stage_profiling.py): per-device
+    stage memory = weight state (bf16 w+g + fp32 m,v = 6x the bf16
+    param bytes, sharded over tp) + 1F1B in-flight activations (stage s
+    holds up to min(P - s, M) live microbatches); infeasible combos are
+    skipped.
     """
     if db is None:
         db = _default_db()
@@ -178,6 +186,18 @@ def profiled_stage_search(num_devices: int, num_microbatches: int,
                                   0.25 * coll_time("all_reduce", (dp, tp),
                                                    0, pb))
                 stage_costs.append(t)
+            if memory_budget is not None and layer_param_bytes is not None:
+                feasible = True
+                for si, (a, b) in enumerate(ranges):
+                    state = 6.0 * sum(layer_param_bytes[a:b]) / tp
+                    # ~3 boundary-sized live tensors per layer per
+                    # in-flight microbatch (residual + attn + mlp acts)
+                    act = min(P - si, M) * (b - a) * 3.0 *                         boundary_act_bytes / max(dp, 1)
+                    if state + act > memory_budget:
+                        feasible = False
+                        break
+                if not feasible:
+                    continue
             # cross-stage p2p: each dp replica sends its shard on its own
             # xGMI link concurrently
             comm = alpha + boundary_act_bytes / max(dp, 1) * beta \

@@ -375,3 +375,27 @@ def test_stage_execution_trace(tmp_path):
         bwd = [n for n, c in events if c == "bwd"]
         assert len(fwd) == 2 and len(bwd) == 2, events
         assert all(n.startswith(f"stage{r}.") for n, _ in events)
+
+
+def test_profiled_stage_search_memory_feasibility():
+    """The memory budget (reference max_n_succ_stages) rules out
+    layouts whose per-device weight state does not fit: a model too big
+    for one device forces P > 1 (or tp > 1)."""
+    from alpa_amd.pipeline_parallel.stage_construction import \
+        profiled_stage_search
+    db = _dummy_db()
+    pb = [10e9] * 8  # 80 GB of bf16 params -> 480 GB optimizer state
+    got = profiled_stage_search(
+        8, 8, [1e12] * 8, boundary_act_bytes=1e6, layer_param_bytes=pb,
+        db=db, memory_budget=100e9)
+    assert got is not None
+    P, shapes, ranges, cost = got
+    per_dev_state = 6.0 * sum(pb) / (P * shapes[0][1])
+    assert per_dev_state <= 100e9, (P, shapes)
+    assert P * shapes[0][0] * shapes[0][1] == 8
+    # a tighter budget forces an even finer split: 480 GB state needs
+    # P x tp >= 8 to fit under 70 GB/device
+    P1, sh1, _, _ = profiled_stage_search(
+        8, 8, [1e12] * 8, boundary_act_bytes=1e6, layer_param_bytes=pb,
+        db=db, memory_budget=70e9)
+    assert P1 * sh1[0][1] >= 8, (P1, sh1)
