@@ -25,10 +25,15 @@ def main():
 
     aa.init()
     cfg = gpt_config(args.model, seq_len=512)
+    # --stages N pins the layout; without it the profile-guided search
+    # (measured cost DB) picks (stages x submesh) itself
     method = aa.PipeshardParallel(num_micro_batches=args.nmb,
-                                  num_stages=args.stages or aa.world_size(),
+                                  num_stages=args.stages,
+                                  stage_option="auto",
                                   schedule="1f1b")
-    state = aa.TrainState.create(gpt_pipeline_spec(cfg), method, lr=3e-4)
+    spec = gpt_pipeline_spec(
+        cfg, microbatch_tokens=args.batch * cfg.seq_len // args.nmb)
+    state = aa.TrainState.create(spec, method, lr=3e-4)
     step = aa.parallelize(lambda m, b: None, method=method)
 
     for i in range(args.steps):
