@@ -36,6 +36,9 @@ class PipelinePlan:
     schedule: str = "1f1b"
     layer_ranges: List[Tuple[int, int]] = field(default_factory=list)
     stage_mesh_shape: Tuple[int, int] = (1, 1)
+    #: heterogeneous per-stage shapes (reference auto-search submeshes);
+    #: overrides stage_mesh_shape when set
+    stage_mesh_shapes: Optional[List[Tuple[int, int]]] = None
 
 
 @dataclass
@@ -60,7 +63,10 @@ class ParallelPlan:
             pipeline_plan=PipelinePlan(
                 num_stages=pp["num_stages"], schedule=pp["schedule"],
                 layer_ranges=[tuple(r) for r in pp["layer_ranges"]],
-                stage_mesh_shape=tuple(pp["stage_mesh_shape"]))
+                stage_mesh_shape=tuple(pp["stage_mesh_shape"]),
+                stage_mesh_shapes=[tuple(x) for x in
+                                   pp["stage_mesh_shapes"]]
+                if pp.get("stage_mesh_shapes") else None)
             if pp else None,
             stage_plans=[
                 StagePlan(logical_mesh_shape=tuple(s["logical_mesh_shape"]),
@@ -75,11 +81,13 @@ def plan_to_method(plan: ParallelPlan):
     plan_to_method, parallel_plan.py:57)."""
     from .parallel_method import PipeshardParallel, ShardParallel
     if plan.pipeline_plan is not None and plan.pipeline_plan.num_stages > 1:
+        pp = plan.pipeline_plan
         return PipeshardParallel(
             num_micro_batches=plan.num_micro_batches,
-            num_stages=plan.pipeline_plan.num_stages,
-            stage_mesh_shape=plan.pipeline_plan.stage_mesh_shape,
-            schedule=plan.pipeline_plan.schedule,
+            num_stages=pp.num_stages,
+            stage_mesh_shape=tuple(pp.stage_mesh_shape),
+            stage_mesh_shapes=pp.stage_mesh_shapes,
+            schedule=pp.schedule,
             stage_option="manual")
     shape = plan.stage_plans[0].logical_mesh_shape if plan.stage_plans \
         else (plan.world_size, 1)
@@ -93,13 +101,16 @@ def method_to_plan(method, world_size: int,
     from .parallel_method import PipeshardParallel
     if isinstance(method, PipeshardParallel):
         from .pipeline_parallel.compile import resolve_stage_layout
-        P, shape = resolve_stage_layout(method, world_size)
+        P, shapes = resolve_stage_layout(method, world_size)
+        hetero = len(set(shapes)) > 1
         return ParallelPlan(
             world_size=world_size,
             num_micro_batches=method.num_micro_batches,
-            pipeline_plan=PipelinePlan(num_stages=P,
-                                       schedule=method.schedule,
-                                       stage_mesh_shape=shape))
+            pipeline_plan=PipelinePlan(
+                num_stages=P, schedule=method.schedule,
+                stage_mesh_shape=tuple(shapes[0]),
+                stage_mesh_shapes=[tuple(sh) for sh in shapes]
+                if hetero else None))
     sp = StagePlan(
         logical_mesh_shape=method.logical_mesh_shape or (world_size, 1))
     if sharding_plan is not None:

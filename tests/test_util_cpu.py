@@ -186,3 +186,26 @@ def test_value_and_grad():
     g = aa.grad(lambda mod, b: mod.loss(*b))(m, batch)
     for n in grads:
         torch.testing.assert_close(g[n], grads[n])
+
+
+def test_pipeline_plan_roundtrip(tmp_path):
+    """Pipeline plan save/replay incl. heterogeneous per-stage shapes
+    (reference LoadSolutionParallelArgs workflow)."""
+    import alpa_amd as aa
+    from alpa_amd.parallel_plan import (ParallelPlan, method_to_plan,
+                                        plan_to_method)
+    m = aa.PipeshardParallel(num_micro_batches=4, num_stages=2,
+                             stage_mesh_shape=(2, 2))
+    plan = method_to_plan(m, 8)
+    plan.save(str(tmp_path / "p.json"))
+    loaded = ParallelPlan.load(str(tmp_path / "p.json"))
+    m2 = plan_to_method(loaded)
+    assert m2.num_stages == 2 and tuple(m2.stage_mesh_shape) == (2, 2)
+    # heterogeneous
+    mh = aa.PipeshardParallel(num_micro_batches=2,
+                              stage_mesh_shapes=[(1, 2), (1, 2), (1, 4)])
+    plan = method_to_plan(mh, 8)
+    plan.save(str(tmp_path / "h.json"))
+    m3 = plan_to_method(ParallelPlan.load(str(tmp_path / "h.json")))
+    assert [tuple(sh) for sh in m3.stage_mesh_shapes] == \
+        [(1, 2), (1, 2), (1, 4)]
