@@ -33,7 +33,8 @@ struct AttnStrides {
 // ABL: ablation level for perf diagnosis (0 = full kernel; higher skips
 // later phases; asm keep-alives prevent dead-code elimination of earlier
 // phases — guide methodology rule 17)
-template <int Dp, int ABL = 0, bool AL = false, bool VL = false>
+template <int Dp, int ABL = 0, bool AL = false, bool VL = false,
+          bool DB = true>
 __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
@@ -67,8 +68,11 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   short* op = o + batch * st.ob + head * st.oh;
   const int q_row0 = qb * (16 * NW) + wave * 16;
 
-  __shared__ short k_lds[2][ATTN_BLOCK_K][Dp + LP];
-  __shared__ short vt_lds[2][Dp][ATTN_BLOCK_K + LP];
+  // DB=false: single-buffered K/V (64 KB total for Dp=96 instead of
+  // 91 KB) fits TWO blocks per CU — inter-block latency hiding replaces
+  // the intra-block prefetch (experiment; see profiles/ PMC notes)
+  __shared__ short k_lds[DB ? 2 : 1][ATTN_BLOCK_K][Dp + LP];
+  __shared__ short vt_lds[DB ? 2 : 1][Dp][ATTN_BLOCK_K + LP];
   __shared__ short p_lds[NW][16][ATTN_BLOCK_K + LP];
 
   // ---- Q fragments in registers ----
@@ -153,6 +157,19 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     }
   };
 
+  auto advance = [&](int ti, int cur) {
+    if (ti + 1 >= n_tiles) return;
+    if (DB) {
+      write_tile(1 - cur);  // waits on the prefetched loads here
+      __syncthreads();
+    } else {
+      __syncthreads();      // everyone done reading buf 0
+      issue_loads(ti + 1);
+      write_tile(0);
+      __syncthreads();
+    }
+  };
+
   if (n_tiles > 0) {
     issue_loads(0);
     write_tile(0);
@@ -161,19 +178,19 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
 
   for (int ti = 0; ti < n_tiles; ++ti) {
     const int kvb = ti * ATTN_BLOCK_K;
-    const int cur = ti & 1;
-    if (ti + 1 < n_tiles) issue_loads(ti + 1);  // lands under the compute
+    const int cur = DB ? (ti & 1) : 0;
+    if (DB && ti + 1 < n_tiles) issue_loads(ti + 1);  // lands under compute
 
     // causal: tiles entirely above this wave's q rows contribute nothing
     if (causal && kvb > q_row0 + 15) {
-      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      advance(ti, cur);
       continue;
     }
 
     if (ABL >= 4) {  // staging-only
       float keep = bf2f(k_lds[cur][lane][0]) + bf2f(vt_lds[cur][lane][0]);
       asm volatile("" ::"v"(keep));
-      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      advance(ti, cur);
       continue;
     }
 
@@ -196,7 +213,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
 #pragma unroll
       for (int nt = 0; nt < NTILES; ++nt)
         asm volatile("" ::"v"(s_acc[nt][0]), "v"(s_acc[nt][3]));
-      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      advance(ti, cur);
       continue;
     }
     if (ABL == 2) {  // no softmax math: raw S write + one read
@@ -207,7 +224,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
           p_lds[wave][lo][nt * 16 + hi * 4 + r] = f2bf(s_acc[nt][r]);
       bf16x8 a0 = *reinterpret_cast<const bf16x8*>(&p_lds[wave][lo][hi * 8]);
       asm volatile("" ::"v"(a0));
-      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      advance(ti, cur);
       continue;
     }
 
@@ -262,7 +279,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
 
     if (ABL >= 1) {
       asm volatile("" ::"v"(l_state), "v"(m_state));
-      if (ti + 1 < n_tiles) { write_tile(1 - cur); __syncthreads(); }
+      advance(ti, cur);
       continue;
     }
 
@@ -280,10 +297,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
       }
     }
 
-    if (ti + 1 < n_tiles) {
-      write_tile(1 - cur);  // waits on the prefetched loads here
-      __syncthreads();
-    }
+    advance(ti, cur);
   }
 
   // ---- epilogue: lane holds O^T[d = 16*dt + hi*4 + r][q = lo] ----
@@ -626,6 +640,34 @@ hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
   }
 #undef FWD_DISPATCH
 #undef FWD_VARIANT
+  return hipGetLastError();
+}
+
+// Single-buffer experiment entry (plain path only: no alibi/varlen)
+hipError_t launch_attn_fwd_sbuf(const void* q, const void* k, const void* v,
+                                void* o, float* lse, int64_t B, int64_t H,
+                                int64_t S, int64_t Skv, int64_t D,
+                                float scale, int causal,
+                                const int64_t* strides,
+                                hipStream_t stream) {
+  dim3 grid((uint32_t)ceil_div(S, 16 * (ATTN_THREADS / 64)),
+            (uint32_t)(B * H));
+  dim3 block(ATTN_THREADS);
+  AttnStrides st;
+  st.qb = strides[0]; st.qh = strides[1]; st.qs = strides[2];
+  st.kb = strides[3]; st.kh = strides[4]; st.ks = strides[5];
+  st.vb = strides[6]; st.vh = strides[7]; st.vs = strides[8];
+  st.ob = strides[9]; st.oh = strides[10]; st.os = strides[11];
+#define SBUF_CASE(DP)                                                      \
+  attn_fwd_kernel<DP, 0, false, false, false><<<grid, block, 0, stream>>>( \
+      (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,   \
+      (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr, nullptr,   \
+      st)
+  if (D <= 64) SBUF_CASE(64);
+  else if (D <= 96) SBUF_CASE(96);
+  else if (D <= 128) SBUF_CASE(128);
+  else return hipErrorInvalidValue;
+#undef SBUF_CASE
   return hipGetLastError();
 }
 

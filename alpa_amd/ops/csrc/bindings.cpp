@@ -51,6 +51,10 @@ hipError_t launch_attn_fwd(const void*, const void*, const void*, void*,
                            float*, int64_t, int64_t, int64_t, int64_t,
                            int64_t, float, int, const float*, const int*,
                            const int64_t*, hipStream_t);
+hipError_t launch_attn_fwd_sbuf(const void*, const void*, const void*,
+                                void*, float*, int64_t, int64_t, int64_t,
+                                int64_t, int64_t, float, int,
+                                const int64_t*, hipStream_t);
 hipError_t launch_attn_fwd_v2(const void*, const void*, const void*, void*,
                               float*, int64_t, int64_t, int64_t, int64_t,
                               int64_t, float, int, const int64_t*,
@@ -359,6 +363,25 @@ at::Tensor attn_fwd_ablate(const at::Tensor& q, const at::Tensor& k,
   return o;
 }
 
+std::vector<at::Tensor> attn_fwd_sbuf(const at::Tensor& q,
+                                      const at::Tensor& k,
+                                      const at::Tensor& v, bool causal,
+                                      double scale) {
+  int64_t B = q.size(0), H = q.size(1), S = q.size(2);
+  auto o = at::empty_like(q.contiguous());
+  auto lse = at::empty({B, H, S}, q.options().dtype(at::kFloat));
+  int64_t strides[12] = {q.stride(0), q.stride(1), q.stride(2),
+                         k.stride(0), k.stride(1), k.stride(2),
+                         v.stride(0), v.stride(1), v.stride(2),
+                         o.stride(0), o.stride(1), o.stride(2)};
+  HIP_OK(launch_attn_fwd_sbuf(q.const_data_ptr(), k.const_data_ptr(),
+                              v.const_data_ptr(), o.mutable_data_ptr(),
+                              (float*)lse.mutable_data_ptr(), B, H, S,
+                              k.size(2), q.size(3), (float)scale,
+                              causal ? 1 : 0, strides, cur_stream()));
+  return {o, lse};
+}
+
 std::vector<at::Tensor> attn_fwd_v2(const at::Tensor& q,
                                     const at::Tensor& k,
                                     const at::Tensor& v, bool causal,
@@ -589,6 +612,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd_v2", &attn_fwd_v2, "32x32-MFMA fwd experiment");
   m.def("attn_fwd_ablate", &attn_fwd_ablate, "ablation variants (perf)");
   m.def("attn_bwd_dkv_ablate", &attn_bwd_dkv_ablate, "dkv ablation (perf)");
+  m.def("attn_fwd_sbuf", &attn_fwd_sbuf,
+        "single-buffer LDS fwd variant (2 blocks/CU experiment)");
   m.def("attn_fwd_blocked", &attn_fwd_blocked,
         "blocked fwd for head_dim > 128 (hipBLASLt scores)");
   m.def("attn_bwd", &attn_bwd, "flash attention bwd (gfx950 MFMA)",

@@ -49,6 +49,15 @@ def main():
     t = timeit(lambda: ext.attn_fwd(q, k, v, True, scale), args.iters)
     print(f"fwd:  {t*1e6:8.1f} us  {fwd_flops/t/1e12:7.1f} TF")
 
+    if hasattr(ext, "attn_fwd_sbuf"):
+        o1, l1 = ext.attn_fwd(q, k, v, True, scale)
+        o3, l3 = ext.attn_fwd_sbuf(q, k, v, True, scale)
+        err = (o1.float() - o3.float()).abs().max().item()
+        t = timeit(lambda: ext.attn_fwd_sbuf(q, k, v, True, scale),
+                   args.iters)
+        print(f"sbuf: {t*1e6:8.1f} us  {fwd_flops/t/1e12:7.1f} TF  "
+              f"(maxdiff vs db: {err:.4f})")
+
     if hasattr(ext, "attn_fwd_v2"):
         o1, l1 = ext.attn_fwd(q, k, v, True, scale)
         o2, l2 = ext.attn_fwd_v2(q, k, v, True, scale)
