@@ -54,7 +54,10 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   constexpr int TOTAL_G = ATTN_BLOCK_K * GPR;
   constexpr int G_PER_T = (TOTAL_G + ATTN_THREADS - 1) / ATTN_THREADS;
 
-  const int qb = blockIdx.x;
+  // causal: later q blocks touch more kv tiles — launch them FIRST so
+  // the heavy blocks don't land in the tail wave of the dispatch
+  const int qb = causal ? ((int)gridDim.x - 1 - (int)blockIdx.x)
+                        : (int)blockIdx.x;
   const int bh = blockIdx.y;
   const int batch = bh / H, head = bh % H;
   const int wave = threadIdx.x >> 6;
@@ -800,7 +803,8 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
   constexpr int GPR = Dp / 8;
   constexpr int PAIRS = 32 * GPR;       // (64 rows / 2) * groups
 
-  const int qb = blockIdx.x;
+  const int qb = causal ? ((int)gridDim.x - 1 - (int)blockIdx.x)
+                        : (int)blockIdx.x;
   const int bh = blockIdx.y;
   const int batch = bh / H, head = bh % H;
   const int wave = threadIdx.x >> 6;
