@@ -30,6 +30,17 @@ struct AttnStrides {
   int64_t ob, oh, os;
 };
 
+// XCD-aware workgroup swizzle (gfx950: 8 XCDs, hardware assigns
+// workgroup i to XCD i%8, each XCD has its own L2): remap so each XCD
+// gets a CONTIGUOUS run of logical blocks — all blocks of one
+// (batch,head) then share one L2 and its K/V (fwd) / Q,dO (bwd) tiles
+// are loaded from HBM once per XCD instead of once per block.
+__device__ inline int xcd_swizzle(int flat, int total) {
+  constexpr int NXCD = 8;
+  if (total % NXCD != 0) return flat;
+  return (flat % NXCD) * (total / NXCD) + flat / NXCD;
+}
+
 // ABL: ablation level for perf diagnosis (0 = full kernel; higher skips
 // later phases; asm keep-alives prevent dead-code elimination of earlier
 // phases — guide methodology rule 17)
@@ -54,11 +65,15 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   constexpr int TOTAL_G = ATTN_BLOCK_K * GPR;
   constexpr int G_PER_T = (TOTAL_G + ATTN_THREADS - 1) / ATTN_THREADS;
 
-  // causal: later q blocks touch more kv tiles — launch them FIRST so
-  // the heavy blocks don't land in the tail wave of the dispatch
-  const int qb = causal ? ((int)gridDim.x - 1 - (int)blockIdx.x)
-                        : (int)blockIdx.x;
-  const int bh = blockIdx.y;
+  // XCD co-location + causal ordering: swizzle the flattened id so one
+  // XCD owns contiguous (bh, qb) work; within a bh, later q blocks
+  // (more kv tiles under causal) go first to avoid a heavy tail
+  const int flat_ = xcd_swizzle(
+      (int)(blockIdx.y * gridDim.x + blockIdx.x),
+      (int)(gridDim.x * gridDim.y));
+  const int qb_raw = flat_ % (int)gridDim.x;
+  const int qb = causal ? ((int)gridDim.x - 1 - qb_raw) : qb_raw;
+  const int bh = flat_ / (int)gridDim.x;
   const int batch = bh / H, head = bh % H;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -803,9 +818,12 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
   constexpr int GPR = Dp / 8;
   constexpr int PAIRS = 32 * GPR;       // (64 rows / 2) * groups
 
-  const int qb = causal ? ((int)gridDim.x - 1 - (int)blockIdx.x)
-                        : (int)blockIdx.x;
-  const int bh = blockIdx.y;
+  const int flat_ = xcd_swizzle(
+      (int)(blockIdx.y * gridDim.x + blockIdx.x),
+      (int)(gridDim.x * gridDim.y));
+  const int qb_raw = flat_ % (int)gridDim.x;
+  const int qb = causal ? ((int)gridDim.x - 1 - qb_raw) : qb_raw;
+  const int bh = flat_ / (int)gridDim.x;
   const int batch = bh / H, head = bh % H;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
@@ -1010,8 +1028,11 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
   constexpr int GPR = Dp / 8;
   constexpr int PAIRS = 32 * GPR;
 
-  const int kvb_idx = blockIdx.x;
-  const int bh = blockIdx.y;
+  const int flat_ = xcd_swizzle(
+      (int)(blockIdx.y * gridDim.x + blockIdx.x),
+      (int)(gridDim.x * gridDim.y));
+  const int kvb_idx = flat_ % (int)gridDim.x;
+  const int bh = flat_ / (int)gridDim.x;
   const int batch = bh / H, head = bh % H;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
