@@ -171,9 +171,19 @@ class ExpertParallelMLP(nn.Module):
         routed = routed.view(self.ep, self.e_local, C, H) \
             .transpose(0, 1).reshape(self.e_local, self.ep * C, H)
 
-        h = torch.bmm(routed, self.w1.to(routed.dtype))
-        h = torch.nn.functional.gelu(h, approximate="tanh")
-        h = torch.bmm(h, self.w2.to(h.dtype))
+        # Per-expert 2-D GEMMs instead of one batched bmm: torch.bmm's
+        # BACKWARD hits hipErrorIllegalAddress in hipBLASLt at MoE-scale
+        # bf16 shapes on this stack (repro: bmm([8,8192,1024],
+        # [8,1024,4096]).backward() on random tensors, ROCm 7.2 /
+        # torch 2.10) — the plain mm path is unaffected and equally
+        # fast at these sizes
+        w1 = self.w1.to(routed.dtype)
+        w2 = self.w2.to(routed.dtype)
+        h = torch.stack([
+            torch.nn.functional.gelu(routed[j] @ w1[j],
+                                     approximate="tanh") @ w2[j]
+            for j in range(self.e_local)
+        ])
 
         # return all-to-all: back to the source ranks' [E, C, H] layout
         h = h.view(self.e_local, self.ep, C, H).transpose(0, 1) \
