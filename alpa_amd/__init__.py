@@ -17,10 +17,11 @@ from .mesh import (DeviceMesh, VirtualMesh, device, full_mesh, full_virtual_mesh
                    get_device_mesh, init_distributed, local_rank, rank,
                    world_size)
 from .optim import AdamW
-from .parallel_method import (AutoShardingOption, DataParallel,
-                              ParallelMethod, PipeshardParallel, ShardParallel,
+from .parallel_method import (AutoShardingOption, CreateStateParallel,
+                              DataParallel, FollowParallel, ParallelMethod,
+                              PipeshardParallel, ShardParallel,
                               Zero2Parallel, Zero3Parallel,
-                              get_3d_parallel_method)
+                              get_3d_parallel_method, parallelize_inference)
 
 from .version import __version__, check_hip_ops_version  # noqa: F401
 
@@ -30,5 +31,8 @@ __all__ = [
     "get_device_mesh", "init_distributed", "local_rank", "rank", "world_size",
     "ParallelMethod", "ShardParallel", "DataParallel", "Zero2Parallel",
     "Zero3Parallel", "PipeshardParallel", "AutoShardingOption",
-    "get_3d_parallel_method", "global_config",
+    "CreateStateParallel", "FollowParallel", "parallelize_inference",
+    "get_3d_parallel_method", "global_config", "grad", "value_and_grad",
+    "save_checkpoint", "restore_checkpoint", "save_train_state",
+    "restore_train_state",
 ]
