@@ -163,3 +163,131 @@ def shard_ring(x: torch.Tensor, mesh: Optional[DeviceMesh],
     i = mesh.axis_index(axis)
     per = x.shape[dim] // n
     return x.narrow(dim, i * per, per).contiguous()
+
+
+# ---------------------------------------------------------------------------
+# Zigzag ring attention: causal load balancing
+# ---------------------------------------------------------------------------
+
+
+def shard_zigzag(x: torch.Tensor, mesh: Optional[DeviceMesh],
+                 axis: int = 1, dim: int = 2) -> torch.Tensor:
+    """Zigzag sharding: the sequence splits into 2n chunks; rank i holds
+    the CONCAT of chunks i and 2n-1-i.  Under causal attention the two
+    chunks' workloads sum to the same total on every rank — the plain
+    ring's idle-rank-0 imbalance disappears."""
+    if mesh is None or mesh.axis_size(axis) == 1:
+        return x
+    n = mesh.axis_size(axis)
+    i = mesh.axis_index(axis)
+    per = x.shape[dim] // (2 * n)
+    a = x.narrow(dim, i * per, per)
+    b = x.narrow(dim, (2 * n - 1 - i) * per, per)
+    return torch.cat([a, b], dim=dim).contiguous()
+
+
+def unshard_zigzag_grad_positions(n: int, i: int):
+    """Chunk indices held by rank i under zigzag."""
+    return (i, 2 * n - 1 - i)
+
+
+def _zz_pair_fwd(q, kv_k, kv_v, my_c, in_c, scale):
+    """Partial attention of one local q chunk against one arriving kv
+    chunk with chunk-level causality (in_c < my_c: full; ==: causal;
+    >: None)."""
+    if in_c > my_c:
+        return None
+    return _chunk_fwd(q, kv_k, kv_v, in_c == my_c, scale)
+
+
+class _ZigzagRingAttention(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, q, k, v, mesh: DeviceMesh, axis: int, scale):
+        if scale is None:
+            scale = 1.0 / math.sqrt(q.shape[-1])
+        n = mesh.axis_size(axis) if mesh is not None else 1
+        if n == 1 or not is_distributed():
+            o, lse = _chunk_fwd(q, k, v, True, scale)
+            ctx.save_for_backward(q, k, v, o, lse)
+            ctx.meta = (mesh, axis, scale, 1, 0, ())
+            return o
+        peers = mesh.axis_ranks(axis)
+        r = mesh.axis_index(axis)
+        B, H, S2, D = q.shape
+        Sc = S2 // 2
+        my_cs = unshard_zigzag_grad_positions(n, r)
+        qs = [q[:, :, :Sc], q[:, :, Sc:]]
+        k_cur, v_cur = k.contiguous(), v.contiguous()
+        o_acc = [torch.zeros(B, H, Sc, D, dtype=torch.float32,
+                             device=q.device) for _ in range(2)]
+        lse_acc = [torch.full((B, H, Sc), float("-inf"),
+                              dtype=torch.float32, device=q.device)
+                   for _ in range(2)]
+        for t in range(n):
+            src = (r - t) % n
+            in_cs = unshard_zigzag_grad_positions(n, src)
+            for qi in range(2):
+                for ki in range(2):
+                    res = _zz_pair_fwd(
+                        qs[qi], k_cur[:, :, ki * Sc:(ki + 1) * Sc],
+                        v_cur[:, :, ki * Sc:(ki + 1) * Sc],
+                        my_cs[qi], in_cs[ki], scale)
+                    if res is not None:
+                        o_acc[qi], lse_acc[qi] = _merge(
+                            o_acc[qi], lse_acc[qi], res[0],
+                            res[1].float())
+            if t < n - 1:
+                k_cur, v_cur = _ring_pass(peers, r, [k_cur, v_cur])
+        o = torch.cat(o_acc, dim=2).to(q.dtype)
+        lse = torch.cat(lse_acc, dim=2)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.meta = (mesh, axis, scale, n, r, peers)
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        mesh, axis, scale, n, r, peers = ctx.meta
+        if n == 1:
+            dq, dk, dv = _chunk_bwd(do, q, k, v, o, lse, True, scale)
+            return dq, dk, dv, None, None, None
+        B, H, S2, D = q.shape
+        Sc = S2 // 2
+        my_cs = unshard_zigzag_grad_positions(n, r)
+        k_cur, v_cur = k.contiguous(), v.contiguous()
+        dq_acc = torch.zeros(q.shape, dtype=torch.float32, device=q.device)
+        dk_acc = torch.zeros(k.shape, dtype=torch.float32, device=k.device)
+        dv_acc = torch.zeros(v.shape, dtype=torch.float32, device=v.device)
+        for t in range(n):
+            src = (r - t) % n
+            in_cs = unshard_zigzag_grad_positions(n, src)
+            for qi in range(2):
+                qsl = slice(qi * Sc, (qi + 1) * Sc)
+                for ki in range(2):
+                    if in_cs[ki] > my_cs[qi]:
+                        continue
+                    ksl = slice(ki * Sc, (ki + 1) * Sc)
+                    dq_t, dk_t, dv_t = _chunk_bwd(
+                        do[:, :, qsl], q[:, :, qsl], k_cur[:, :, ksl],
+                        v_cur[:, :, ksl], o[:, :, qsl], lse[:, :, qsl],
+                        in_cs[ki] == my_cs[qi], scale)
+                    dq_acc[:, :, qsl] += dq_t.float()
+                    dk_acc[:, :, ksl] += dk_t.float()
+                    dv_acc[:, :, ksl] += dv_t.float()
+            if t < n - 1:
+                k_cur, v_cur, dk_acc, dv_acc = _ring_pass(
+                    peers, r, [k_cur, v_cur, dk_acc, dv_acc])
+        dk_acc, dv_acc = _ring_pass(peers, r, [dk_acc, dv_acc])
+        return (dq_acc.to(q.dtype), dk_acc.to(k.dtype),
+                dv_acc.to(v.dtype), None, None, None)
+
+
+def zigzag_ring_attention(q: torch.Tensor, k: torch.Tensor,
+                          v: torch.Tensor, mesh: Optional[DeviceMesh],
+                          axis: int = 1,
+                          scale: Optional[float] = None) -> torch.Tensor:
+    """Causal ring attention with zigzag load balancing: q/k/v are this
+    rank's shard_zigzag output ([B, h, 2*(S/2n), d]).  Always causal —
+    that is what the balancing is for."""
+    return _ZigzagRingAttention.apply(q, k, v, mesh, axis, scale)
