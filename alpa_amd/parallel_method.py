@@ -133,6 +133,9 @@ class PipeshardParallel(ParallelMethod):
     stage_mesh_shapes: Optional[list] = None
     layer_option: str = "auto"  # "auto" | "manual"
     stage_option: str = "uniform"  # "uniform" | "auto" | "manual"
+    #: per-device memory budget (bytes) for the auto stage search's
+    #: feasibility check (reference max_n_succ_stages)
+    memory_budget_per_device: Optional[float] = None
     schedule: str = "1f1b"  # "1f1b" | "gpipe" | "inference"
 
     def resolve_mesh(self) -> DeviceMesh:

@@ -41,10 +41,21 @@ def resolve_stage_layout(method, n: Optional[int] = None,
         if (method.stage_option == "auto" and spec is not None
                 and spec.layer_flops is not None
                 and method.stage_mesh_shape is None):
-            from .stage_construction import profiled_stage_search
-            P, shapes, _ranges, _cost = profiled_stage_search(
+            from .stage_construction import (profiled_stage_search,
+                                             training_dp_search)
+            budget = getattr(method, "memory_budget_per_device", None)
+            got = training_dp_search(
                 n, method.num_micro_batches, spec.layer_flops,
-                spec.boundary_act_bytes, spec.layer_param_bytes)
+                spec.boundary_act_bytes, spec.layer_param_bytes,
+                memory_budget=budget)
+            if got is None:  # infeasible under the budget: uniform search
+                got = profiled_stage_search(
+                    n, method.num_micro_batches, spec.layer_flops,
+                    spec.boundary_act_bytes, spec.layer_param_bytes)
+            P, shapes, ranges, _cost = got
+            # the DP pairs its layer ranges WITH the submeshes — stash
+            # them so the compiler does not re-cluster uniformly
+            method._auto_layer_ranges = list(ranges)
             return P, shapes
         from .stage_construction import auto_num_stages
         P = auto_num_stages(n, method.num_micro_batches)
@@ -88,8 +99,12 @@ def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
             next_peer = int(stage_meshes[my_stage + 1].grid[coord])
 
     # layer clustering (auto DP over costs; reference layer_construction.py:342)
-    costs = spec.layer_costs or uniform_layer_costs(spec.num_layers)
-    ranges = cluster_layers(costs, P)
+    # — unless the auto stage search already chose ranges paired with
+    # its (possibly heterogeneous) submeshes
+    ranges = getattr(method, "_auto_layer_ranges", None)
+    if ranges is None or len(ranges) != P:
+        costs = spec.layer_costs or uniform_layer_costs(spec.num_layers)
+        ranges = cluster_layers(costs, P)
     layer_range = ranges[my_stage]
 
     dtype = getattr(torch, global_config.compute_dtype) \

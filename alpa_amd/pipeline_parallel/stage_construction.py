@@ -206,3 +206,128 @@ def profiled_stage_search(num_devices: int, num_microbatches: int,
             if best is None or cost < best[3]:
                 best = (P, [(dp, tp)] * P, ranges, cost)
     return best
+
+
+def training_dp_search(num_devices: int, num_microbatches: int,
+                       layer_flops: Sequence[float],
+                       boundary_act_bytes: float = 0.0,
+                       layer_param_bytes: Optional[Sequence[float]] = None,
+                       db=None, cluster_key: str = "mi355x",
+                       memory_budget: Optional[float] = None
+                       ) -> Optional[Tuple[int, List[Tuple[int, int]],
+                                           List[Tuple[int, int]], float]]:
+    """The reference's inter-op training DP (training_dp_impl,
+    stage_construction.py:235): minimize ``total + (M-1)·max_stage`` over
+    HETEROGENEOUS stage layouts — F[s][i][j] = min total latency to place
+    layers[i:] on j devices in s stages, enumerating (submesh size d x
+    logical shape (dp, tp)) per stage, outer loop over max-stage-cost
+    thresholds (Alpa paper eqn 3).  Per-stage latency/memory come from
+    the same profiled cost model as profiled_stage_search; memory
+    feasibility counts 1F1B in-flight microbatches per SUCCEEDING stage
+    count (max_n_succ_stages).
+
+    Returns (P, [(dp, tp)] per stage, layer ranges, est. step seconds) —
+    shapes may DIFFER per stage (the reference's auto-search submeshes,
+    e.g. (1,2),(1,2),(1,4)); the hetero pipeline runtime executes them
+    via tile resharding at the boundaries.
+    """
+    if db is None:
+        db = _default_db()
+    curve = _matmul_curve(db, cluster_key)
+    L = len(layer_flops)
+    alpha = global_config.mesh_alpha
+    beta = global_config.mesh_beta
+    M = num_microbatches
+
+    def coll_time(kind, mesh_shape, axis, nbytes):
+        if nbytes <= 0:
+            return 0.0
+        try:
+            r = db.query(cluster_key, mesh_shape)
+            return r.estimate_collective(kind, axis, nbytes)
+        except KeyError:
+            nn = mesh_shape[axis]
+            factor = 2 * (nn - 1) / nn if kind == "all_reduce" \
+                else (nn - 1) / nn
+            return alpha + factor * nbytes * beta
+
+    def matmul_time(flops):
+        return curve.estimate(flops) if curve is not None \
+            else flops / 1.2e15 + 5e-6
+
+    # pre-compute per (i, k, shape) stage cost; feasibility depends on
+    # the succeeding-stage count and is checked inside the DP
+    shapes_by_d = {d: _factorizations(d) for d in divisors(num_devices)}
+    pre = [[0.0] * (L + 1) for _ in range(L)]
+    for i in range(L):
+        acc = 0.0
+        for k in range(i, L):
+            acc += layer_flops[k]
+            pre[i][k + 1] = acc
+
+    def stage_cost(i, k, dp, tp):
+        f = pre[i][k]
+        t = matmul_time(f / (dp * tp))
+        if tp > 1:
+            nb = 4 * (k - i) * boundary_act_bytes / max(dp, 1)
+            t += coll_time("all_reduce", (dp, tp), 1, nb)
+        return t
+
+    def feasible(i, k, dp, tp, succ):
+        if memory_budget is None or layer_param_bytes is None:
+            return True
+        state = 6.0 * sum(layer_param_bytes[i:k]) / tp
+        act = min(succ + 1, M) * (k - i) * 3.0 * \
+            boundary_act_bytes / max(dp, 1)
+        return state + act <= memory_budget
+
+    INF = float("inf")
+    # candidate thresholds: all single-stage costs
+    cand = sorted({round(stage_cost(i, k, dp, tp), 12)
+                   for i in range(L) for k in range(i + 1, L + 1)
+                   for d in shapes_by_d for dp, tp in shapes_by_d[d]})
+    best = None
+    max_s = min(num_devices, L)
+    for thr in cand:
+        if best is not None and (M - 1) * thr >= best[3]:
+            break  # larger thresholds can only be worse
+        # F[s][i][j]: layers[i:] on j devices in exactly s stages
+        F = [[[INF] * (num_devices + 1) for _ in range(L + 1)]
+             for _ in range(max_s + 1)]
+        choice = {}
+        F[0][L][0] = 0.0
+        for s in range(1, max_s + 1):
+            for i in range(L - 1, -1, -1):
+                for j in range(1, num_devices + 1):
+                    for d in shapes_by_d:
+                        if d > j:
+                            continue
+                        for (dp, tp) in shapes_by_d[d]:
+                            for k in range(i + 1, L + 1):
+                                nxt = F[s - 1][k][j - d]
+                                if nxt == INF:
+                                    continue
+                                c = stage_cost(i, k, dp, tp)
+                                if c > thr or not feasible(
+                                        i, k, dp, tp, s - 1):
+                                    continue
+                                tot = c + nxt
+                                if tot < F[s][i][j]:
+                                    F[s][i][j] = tot
+                                    choice[(s, i, j)] = (k, d, dp, tp)
+        for s in range(1, max_s + 1):
+            tot = F[s][0][num_devices]
+            if tot == INF:
+                continue
+            comm = (alpha + boundary_act_bytes * beta) * (s - 1) / max(M, 1)
+            cost = tot + (M - 1) * thr + comm
+            if best is None or cost < best[3]:
+                ranges, shapes = [], []
+                i, j = 0, num_devices
+                for ss in range(s, 0, -1):
+                    k, d, dp, tp = choice[(ss, i, j)]
+                    ranges.append((i, k))
+                    shapes.append((dp, tp))
+                    i, j = k, j - d
+                best = (s, shapes, ranges, cost)
+    return best

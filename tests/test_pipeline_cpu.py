@@ -432,3 +432,39 @@ def test_inference_pipeline_matches_serial():
     assert len(got) == 1
     for a, b in zip(got[0], want):
         assert abs(a - b) < 2e-4, (got, want)
+
+
+def test_training_dp_heterogeneous_under_memory_pressure():
+    """The reference-style training DP (F[s][i][j]) emits HETEROGENEOUS
+    submeshes when memory demands it: param-heavy tail layers need a
+    wider tp shard than the light head layers."""
+    from alpa_amd.pipeline_parallel.stage_construction import \
+        training_dp_search
+    flops = [1e12] * 4 + [1e12] * 4
+    pb = [1e8] * 4 + [40e9] * 4   # tail: 160 GB bf16 params -> 960 GB state
+    got = training_dp_search(8, 8, flops, boundary_act_bytes=1e6,
+                             layer_param_bytes=pb, db=_dummy_db(),
+                             memory_budget=200e9)
+    assert got is not None
+    P, shapes, ranges, cost = got
+    assert sum(dp * tp for dp, tp in shapes) == 8
+    assert ranges[0][0] == 0 and ranges[-1][1] == 8
+    # every stage must satisfy the budget
+    for (a, b), (dp, tp) in zip(ranges, shapes):
+        assert 6.0 * sum(pb[a:b]) / tp <= 200e9 + 1e9, (ranges, shapes)
+    # the tail stages hold the heavy layers with bigger tp than a
+    # uniform all-dp layout could afford
+    heavy = [tp for (a, b), (dp, tp) in zip(ranges, shapes) if b > 4]
+    assert max(heavy) >= 4, (ranges, shapes)
+
+
+def test_training_dp_no_worse_than_uniform():
+    from alpa_amd.pipeline_parallel.stage_construction import (
+        profiled_stage_search, training_dp_search)
+    db = _dummy_db()
+    flops = [2e12, 1e12, 1e12, 3e12, 1e12, 1e12, 2e12, 1e12]
+    uni = profiled_stage_search(8, 16, flops, boundary_act_bytes=1e7,
+                                layer_param_bytes=[1e9] * 8, db=db)
+    dp = training_dp_search(8, 16, flops, boundary_act_bytes=1e7,
+                            layer_param_bytes=[1e9] * 8, db=db)
+    assert dp[3] <= uni[3] * 1.05, (dp[3], uni[3])
