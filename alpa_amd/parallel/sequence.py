@@ -103,12 +103,15 @@ class RingSelfAttention(nn.Module):
     def __init__(self, hidden: int, num_heads: int,
                  mesh: Optional[DeviceMesh] = None, sp_axis: int = 1,
                  dtype=torch.float32, device=None, layer_idx: int = 0,
-                 init_seed: int = 0):
+                 init_seed: int = 0, zigzag: bool = False):
         super().__init__()
         self.mesh, self.sp_axis = mesh, sp_axis
         self.sp = mesh.axis_size(sp_axis) if mesh is not None else 1
         self.heads = num_heads
         self.head_dim = hidden // num_heads
+        #: zigzag mode: the residual stream is shard_zigzag-sharded
+        #: (rank i holds chunks i and 2n-1-i) — equal causal work per rank
+        self.zigzag = zigzag
         self.qkv = ColumnParallelLinear(hidden, 3 * hidden, None, 1,
                                         dtype=dtype, device=device,
                                         init_seed=init_seed,
@@ -118,15 +121,20 @@ class RingSelfAttention(nn.Module):
                                      init_tag=f"sp{layer_idx}.out")
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        """x [B, S/sp, hidden] (this rank's seq chunk, rank-ordered)."""
-        from .ring_attention import ring_attention
+        """x [B, S/sp, hidden] (this rank's seq chunk; rank-ordered, or
+        shard_zigzag layout in zigzag mode)."""
+        from .ring_attention import ring_attention, zigzag_ring_attention
         B, Sl, _ = x.shape
         h, d = self.heads, self.head_dim
         qkv = self.qkv(x).view(B, Sl, h, 3, d)
         q = qkv[:, :, :, 0].permute(0, 2, 1, 3).contiguous()
         k = qkv[:, :, :, 1].permute(0, 2, 1, 3).contiguous()
         v = qkv[:, :, :, 2].permute(0, 2, 1, 3).contiguous()
-        o = ring_attention(q, k, v, self.mesh, self.sp_axis, causal=True)
+        if self.zigzag and self.sp > 1:
+            o = zigzag_ring_attention(q, k, v, self.mesh, self.sp_axis)
+        else:
+            o = ring_attention(q, k, v, self.mesh, self.sp_axis,
+                               causal=True)
         o = o.permute(0, 2, 1, 3).reshape(B, Sl, h * d)
         return self.out(o)
 

@@ -109,3 +109,30 @@ def test_ring_sp2_matches_serial():
                                    atol=1e-5)
         torch.testing.assert_close(
             xg, x.grad[:, r * per:(r + 1) * per], rtol=1e-4, atol=1e-5)
+
+
+def _zz_module_worker(rank, world_size):
+    from alpa_amd.parallel.ring_attention import shard_zigzag
+    from alpa_amd.parallel.sequence import RingSelfAttention
+    mesh = aa.full_mesh((1, world_size))
+    m = RingSelfAttention(HID, HEADS, mesh, 1, init_seed=9, zigzag=True)
+    x = make_x()
+    xl = shard_zigzag(x, mesh, axis=1, dim=1).requires_grad_(True)
+    y = m(xl)
+    y.square().mean().backward()
+    return y.detach()
+
+
+def test_zigzag_module_matches_serial():
+    from alpa_amd.parallel.sequence import RingSelfAttention
+    m = RingSelfAttention(HID, HEADS, None, 1, init_seed=9)
+    x = make_x()
+    y = m(x)
+    results = run_distributed(_zz_module_worker, world_size=2, timeout=300)
+    per = S // 4
+    for r, ys in enumerate(results):
+        lo = y[:, r * per:(r + 1) * per]
+        hi = y[:, (3 - r) * per:(4 - r) * per]
+        want = torch.cat([lo, hi], dim=1)
+        torch.testing.assert_close(torch.as_tensor(ys), want.detach(),
+                                   rtol=1e-4, atol=1e-5)
