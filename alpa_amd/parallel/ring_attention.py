@@ -72,14 +72,23 @@ def _ring_pass(peers: Tuple[int, ...], idx: int,
 
 
 def _merge(o_acc, lse_acc, o_t, lse_t):
-    """Combine two normalized partials: o = Σ exp(lse_i - lse)·o_i."""
+    """Combine two normalized partials: o = Σ exp(lse_i - lse)·o_i.
+    Rows with no attendable keys anywhere (both lse = -inf — cannot
+    happen under causal self-attention, where every row sees at least
+    itself) resolve to o = 0 / lse = -inf like the kernel's empty-row
+    epilogue, instead of 0/0 NaNs."""
     m = torch.maximum(lse_acc, lse_t)
-    w1 = torch.exp(lse_acc - m)
-    w2 = torch.exp(lse_t - m)
-    l = w1 + w2
+    dead = torch.isinf(m) & (m < 0)
+    m_safe = torch.where(dead, torch.zeros_like(m), m)
+    w1 = torch.exp(lse_acc - m_safe)
+    w2 = torch.exp(lse_t - m_safe)
+    l = (w1 + w2).clamp_min(1e-38)
     o = (w1.unsqueeze(-1) * o_acc + w2.unsqueeze(-1) * o_t.float()) / \
         l.unsqueeze(-1)
-    return o, m + torch.log(l)
+    o = torch.where(dead.unsqueeze(-1), torch.zeros_like(o), o)
+    lse = torch.where(dead, torch.full_like(m, float("-inf")),
+                      m_safe + torch.log(l))
+    return o, lse
 
 
 class _RingAttention(torch.autograd.Function):
