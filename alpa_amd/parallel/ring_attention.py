@@ -31,7 +31,7 @@ from typing import List, Optional, Tuple
 import torch
 import torch.distributed as dist
 
-from ..mesh import DeviceMesh, is_distributed, rank
+from ..mesh import DeviceMesh, is_distributed
 from ..ops import reference as ref
 from ..ops._backend import hip_ops, use_hip
 
