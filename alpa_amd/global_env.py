@@ -29,6 +29,11 @@ class GlobalConfig:
     # ---------- compute ----------
     #: default compute dtype for models/benchmarks
     compute_dtype: str = "bfloat16"
+    #: EXPERIMENTAL: fp8(e4m3) forward GEMMs in the parallel linear
+    #: layers (bf16 backward; ops/fp8.py).  NOT valid for BASELINE
+    #: comparisons — env ALPA_AMD_FP8=1
+    fp8_gemm: bool = field(
+        default_factory=lambda: os.environ.get("ALPA_AMD_FP8", "0") == "1")
     #: use hand-written HIP kernels when their extension is available
     use_hip_kernels: bool = True
     #: fail loudly if running on GPU without the HIP extension (anti-silent-fallback)

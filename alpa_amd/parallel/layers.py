@@ -102,6 +102,17 @@ def _sharded_normal_(w: torch.Tensor, full_shape, shard_dim: int,
     del full
 
 
+def _fp8_ok(x, weight) -> bool:
+    """Gate for the EXPERIMENTAL fp8-forward GEMM path (ops/fp8.py):
+    opt-in via global_config.fp8_gemm, CUDA only, dims 16-aligned."""
+    from ..global_env import global_config
+    if not global_config.fp8_gemm or not x.is_cuda:
+        return False
+    from ..ops.fp8 import fp8_available
+    return (fp8_available(x) and x.shape[-1] % 16 == 0 and
+            weight.shape[0] % 16 == 0)
+
+
 class ColumnParallelLinear(nn.Module):
     """Y = X @ W^T + b with W row-sharded over out_features (each rank holds
     out_features/tp rows).  Output stays sharded along the feature dim.
@@ -139,6 +150,11 @@ class ColumnParallelLinear(nn.Module):
 
     def forward(self, x):
         x = copy_to_tp(x, self.mesh, self.axis)
+        if self._use_fp8(x):
+            from ..ops.fp8 import fp8_linear
+            if self.gelu:
+                return ops.bias_gelu(fp8_linear(x, self.weight), self.bias)
+            return fp8_linear(x, self.weight, self.bias)
         y = torch.matmul(x, self.weight.t())
         if self.gelu:
             assert self.bias is not None
@@ -146,6 +162,9 @@ class ColumnParallelLinear(nn.Module):
         if self.bias is not None:
             y = y + self.bias
         return y
+
+    def _use_fp8(self, x):
+        return _fp8_ok(x, self.weight)
 
 
 class RowParallelLinear(nn.Module):
@@ -177,7 +196,11 @@ class RowParallelLinear(nn.Module):
             self.register_parameter("bias", None)
 
     def forward(self, x):
-        y = torch.matmul(x, self.weight.t())
+        if _fp8_ok(x, self.weight):
+            from ..ops.fp8 import fp8_linear
+            y = fp8_linear(x, self.weight)
+        else:
+            y = torch.matmul(x, self.weight.t())
         y = reduce_from_tp(y, self.mesh, self.axis)
         if self.bias is not None:
             y = y + self.bias
