@@ -21,6 +21,29 @@ import torch
 
 E4M3_MAX = 448.0
 
+# epoch counter: optimizers bump it after each parameter update so the
+# per-module quantized-weight caches invalidate (the fused AdamW kernel
+# writes through raw pointers and does not touch torch's version
+# counters)
+_EPOCH = 0
+
+
+def bump_epoch() -> None:
+    global _EPOCH
+    _EPOCH += 1
+
+
+def quantize_weight_cached(module, weight: torch.Tensor):
+    """(wq, ws) for `module.weight`, re-quantized once per optimizer
+    epoch instead of per forward call — the weight-side half of the
+    naive recipe's per-GEMM amax+cast overhead disappears."""
+    cached = getattr(module, "_fp8_cache", None)
+    if cached is not None and cached[0] == _EPOCH:
+        return cached[1], cached[2]
+    wq, ws = _quantize(weight)
+    module._fp8_cache = (_EPOCH, wq, ws)
+    return wq, ws
+
 
 def _quantize(t: torch.Tensor):
     """Per-tensor symmetric scaling into float8_e4m3fn; returns
@@ -41,11 +64,12 @@ class _Fp8Linear(torch.autograd.Function):
     """y = x @ w.T (+ bias): fp8 forward GEMM, exact bf16 backward."""
 
     @staticmethod
-    def forward(ctx, x, w, bias):
+    def forward(ctx, x, w, bias, wq=None, ws=None):
         shape = x.shape
         x2 = x.reshape(-1, shape[-1])
         xq, xs = _quantize(x2)
-        wq, ws = _quantize(w)
+        if wq is None:
+            wq, ws = _quantize(w)
         # _scaled_mm wants B column-major: w.T with w row-major qualifies
         y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
                              bias=bias, out_dtype=x.dtype)
@@ -60,12 +84,17 @@ class _Fp8Linear(torch.autograd.Function):
         dx = (dy2 @ w).reshape(*dy.shape[:-1], w.shape[1])
         dw = dy2.t() @ x2
         db = dy2.sum(0) if ctx.has_bias else None
-        return dx, dw, db
+        return dx, dw, db, None, None
 
 
 def fp8_linear(x: torch.Tensor, weight: torch.Tensor,
-               bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+               bias: Optional[torch.Tensor] = None,
+               module=None) -> torch.Tensor:
     """Drop-in for F.linear with an fp8 forward GEMM.  Requires CUDA and
     dims divisible by 16 (hipBLASLt fp8 tile constraint); callers gate on
-    `fp8_available` and fall back to bf16 matmul otherwise."""
+    `fp8_available` and fall back to bf16 matmul otherwise.  Passing the
+    owning `module` enables the per-epoch quantized-weight cache."""
+    if module is not None:
+        wq, ws = quantize_weight_cached(module, weight)
+        return _Fp8Linear.apply(x, weight, bias, wq, ws)
     return _Fp8Linear.apply(x, weight, bias)

@@ -49,6 +49,8 @@ class AdamW:
         if grads is None:
             grads = [p.grad for p in self.params]
         self.step_count += 1
+        from .ops import fp8 as _fp8
+        _fp8.bump_epoch()  # invalidate fp8 quantized-weight caches
         ops.fused_adamw(self.params, grads, self.exp_avgs, self.exp_avg_sqs,
                         self.step_count, self.lr, self.beta1, self.beta2,
                         self.eps, self.weight_decay, grad_scale)

@@ -153,8 +153,9 @@ class ColumnParallelLinear(nn.Module):
         if self._use_fp8(x):
             from ..ops.fp8 import fp8_linear
             if self.gelu:
-                return ops.bias_gelu(fp8_linear(x, self.weight), self.bias)
-            return fp8_linear(x, self.weight, self.bias)
+                return ops.bias_gelu(fp8_linear(x, self.weight,
+                                                module=self), self.bias)
+            return fp8_linear(x, self.weight, self.bias, module=self)
         y = torch.matmul(x, self.weight.t())
         if self.gelu:
             assert self.bias is not None
@@ -198,7 +199,7 @@ class RowParallelLinear(nn.Module):
     def forward(self, x):
         if _fp8_ok(x, self.weight):
             from ..ops.fp8 import fp8_linear
-            y = fp8_linear(x, self.weight)
+            y = fp8_linear(x, self.weight, module=self)
         else:
             y = torch.matmul(x, self.weight.t())
         y = reduce_from_tp(y, self.mesh, self.axis)
