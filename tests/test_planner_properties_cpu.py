@@ -1,0 +1,71 @@
+"""Property-based tests for the planning DPs and the MoE dispatcher.
+
+- cluster_layers is verified OPTIMAL against brute force over all
+  contiguous partitions (the reference's layer DP claims min-max
+  optimality, layer_construction.py:342).
+- The GShard slot assignment never collides, never exceeds capacity,
+  and accounts for every kept token exactly once.
+"""
+import itertools
+
+import torch
+from hypothesis import given, settings, strategies as st
+
+from alpa_amd.pipeline_parallel.layer_clustering import cluster_layers
+
+
+def _brute_min_max(costs, k):
+    L = len(costs)
+    best = float("inf")
+    for cuts in itertools.combinations(range(1, L), k - 1):
+        bounds = (0,) + cuts + (L,)
+        m = max(sum(costs[a:b]) for a, b in zip(bounds, bounds[1:]))
+        best = min(best, m)
+    return best
+
+
+@settings(max_examples=80, deadline=None)
+@given(st.lists(st.floats(min_value=0.1, max_value=10.0), min_size=2,
+                max_size=9),
+       st.integers(min_value=1, max_value=4))
+def test_cluster_layers_min_max_optimal(costs, k):
+    k = min(k, len(costs))
+    ranges = cluster_layers(costs, k)
+    assert ranges[0][0] == 0 and ranges[-1][1] == len(costs)
+    for (a0, b0), (a1, b1) in zip(ranges, ranges[1:]):
+        assert b0 == a1
+    got = max(sum(costs[a:b]) for a, b in ranges)
+    assert got <= _brute_min_max(costs, k) + 1e-9
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.integers(min_value=1, max_value=64),
+       st.sampled_from([2, 4, 8]),
+       st.floats(min_value=0.25, max_value=4.0),
+       st.integers(min_value=0, max_value=2 ** 31 - 1))
+def test_moe_slot_assignment_invariants(n_tokens, n_experts, cap_factor,
+                                        seed):
+    from alpa_amd.parallel.expert import ExpertParallelMLP, top2_gating
+    g = torch.Generator().manual_seed(seed)
+    logits = torch.randn(n_tokens, n_experts, generator=g)
+    C = max(1, int(cap_factor * n_tokens * 2 / n_experts))
+    w1g, i1, w2g, i2 = top2_gating(logits, C)
+    m = ExpertParallelMLP.__new__(ExpertParallelMLP)
+    m.E = n_experts
+    zeros = torch.zeros(1, n_experts, dtype=torch.long)
+    pos1, keep1, counts = m._assign_slots(i1, C, zeros)
+    pos2, keep2, _ = m._assign_slots(i2, C, counts)
+    slots = torch.cat([(i1 * C + pos1)[keep1], (i2 * C + pos2)[keep2]])
+    # no collisions, all within [0, E*C)
+    assert slots.numel() == slots.unique().numel()
+    if slots.numel():
+        assert int(slots.min()) >= 0
+        assert int(slots.max()) < n_experts * C
+    # per-expert occupancy never exceeds capacity
+    for e in range(n_experts):
+        used = int((i1[keep1] == e).sum() + (i2[keep2] == e).sum())
+        assert used <= C
+    # kept top-1 tokens: exactly the first C per expert in token order
+    for e in range(n_experts):
+        hits = (i1 == e).nonzero().flatten()
+        assert int(keep1[hits].sum()) == min(len(hits), C)
