@@ -408,3 +408,31 @@ def test_continuous_batching_eos_frees_slot():
     expect = solo[:solo.index(eos) + 1]
     assert r.done and r.output == expect
     assert cb.num_active == 0
+
+
+def _tp_batcher_worker(rank, world_size):
+    from alpa_amd.serve.batching import ContinuousBatcher, GenRequest
+    mesh = aa.mesh.full_mesh((1, world_size))
+    m = build_opt(mesh, axis=1)
+    torch.manual_seed(23)
+    prompts = [torch.randint(0, CFG.vocab_size, (n,)) for n in (4, 7)]
+    cb = ContinuousBatcher(m, max_batch=2)
+    reqs = [GenRequest(p, max_new_tokens=4) for p in prompts]
+    for r in reqs:
+        cb.submit(r)
+    cb.run_all()
+    return [r.output for r in reqs]
+
+
+def test_tp2_continuous_batching_matches_serial():
+    """The slot batcher under TP: vocab-parallel greedy + varlen decode
+    stay in SPMD lockstep and match solo serial decodes."""
+    m = build_opt()
+    torch.manual_seed(23)
+    prompts = [torch.randint(0, CFG.vocab_size, (n,)) for n in (4, 7)]
+    want = [m.generate(p.view(1, -1), max_new_tokens=4)[0, len(p):].tolist()
+            for p in prompts]
+    results = run_distributed(_tp_batcher_worker, world_size=2,
+                              timeout=300)
+    for r in results:
+        assert r == want, (r, want)
