@@ -69,3 +69,29 @@ def test_moe_slot_assignment_invariants(n_tokens, n_experts, cap_factor,
     for e in range(n_experts):
         hits = (i1 == e).nonzero().flatten()
         assert int(keep1[hits].sum()) == min(len(hits), C)
+
+
+@settings(max_examples=12, deadline=None)
+@given(st.sampled_from([2, 4, 8]),
+       st.sampled_from([1024, 2560, 4096]),
+       st.sampled_from([4096, 16384, 65536]),
+       st.booleans())
+def test_ilp_plan_wellformed(n, hidden, tokens, force_dp):
+    """Any (devices, model size, batch) input yields a well-formed ILP
+    plan: a valid mesh factorization, a strategy per node, finite
+    objective; force_data_parallel restricts to batch-dim sharding."""
+    from alpa_amd.shard_parallel.auto_sharding import solve_gpt_sharding
+    plan = solve_gpt_sharding(n, hidden=hidden, layers=2, vocab=2048,
+                              tokens=tokens,
+                              force_data_parallel=force_dp)
+    d0, d1 = plan.mesh_shape
+    assert d0 * d1 == n
+    assert plan.objective == plan.objective and plan.objective >= 0
+    assert plan.choices
+    if force_dp:
+        assert d1 == 1
+        # batch-dim-only sharding: the weight axis of every chosen
+        # strategy is None ("colNone"/"rowNone" = replicated weight)
+        for name, strat in plan.choices.items():
+            assert "col0" not in strat and "col1" not in strat and \
+                "row0" not in strat and "row1" not in strat, (name, strat)
