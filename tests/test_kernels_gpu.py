@@ -616,3 +616,19 @@ def test_fp8_gpt_step():
         assert all(g == g for g in loss_bf16_grad)
     finally:
         global_config.fp8_gemm = False
+
+
+def test_moe_scale_regression():
+    """Pin the hipBLASLt bmm-backward fault workaround: the expert FFN
+    at realistic MoE scale (this exact config reproduced
+    hipErrorIllegalAddress through torch.bmm backward) must train."""
+    from alpa_amd.parallel.expert import ExpertParallelMLP
+    torch.manual_seed(24)
+    m = ExpertParallelMLP(1024, 4096, 8, None, 1, capacity_factor=2.0,
+                          dtype=torch.bfloat16, device="cuda")
+    x = torch.randn(16, 1024, 1024, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = m(x)
+    y.float().square().mean().backward()
+    torch.cuda.synchronize()
+    assert x.grad is not None and float(x.grad.abs().sum()) > 0
