@@ -134,6 +134,9 @@ class MLP(nn.Module):
         return self.fc2(self.fc1(x))
 
 
+_PLAIN = object()  # forward() sentinel: single-arg plain-path call
+
+
 class LayerNorm(nn.Module):
 
     def __init__(self, hidden: int, eps: float, dtype, device):
@@ -144,8 +147,16 @@ class LayerNorm(nn.Module):
         self.bias = nn.Parameter(torch.zeros(hidden, dtype=dtype,
                                              device=device))
 
-    def forward(self, x):
-        return ops.layer_norm(x, self.weight, self.bias, self.eps)
+    def forward(self, x, delta=_PLAIN):
+        """Single-arg: plain LayerNorm.  Two-arg (res, delta): fused
+        residual-add + LN, returning only the normalized stream (the
+        model epilogue).  Dispatching through __call__ keeps nn.Module
+        hooks (ZeRO-3 gather/release) working."""
+        if delta is _PLAIN:
+            return ops.layer_norm(x, self.weight, self.bias, self.eps)
+        _, y = ops.add_layer_norm(x, delta, self.weight, self.bias,
+                                  self.eps)
+        return y
 
 
 class Block(nn.Module):
@@ -159,10 +170,17 @@ class Block(nn.Module):
         self.ln2 = LayerNorm(cfg.hidden_size, cfg.layernorm_eps, dtype, device)
         self.mlp = MLP(cfg, mesh, axis, dtype, device, layer_idx, init_seed)
 
-    def forward(self, x):
-        x = x + self.attn(self.ln1(x))
-        x = x + self.mlp(self.ln2(x))
-        return x
+    def forward(self, x, delta=_PLAIN):
+        """Single-arg: plain block on a materialized stream.  Two-arg
+        (res, delta): the fused residual-threaded path.  Dispatching
+        through forward/__call__ (instead of calling forward_fused
+        directly) keeps nn.Module hooks working — ZeRO-3's gather/release
+        hooks depend on it."""
+        if delta is _PLAIN:
+            x = x + self.attn(self.ln1(x))
+            x = x + self.mlp(self.ln2(x))
+            return x
+        return self.forward_fused(x, delta)
 
     def forward_fused(self, res, delta):
         """Residual-threaded form: both residual adds fuse into the
@@ -183,9 +201,8 @@ def _run_block(blk, res, delta, remat: bool):
     inserts at pipeline-layer slices (remat_sliced_eqns)."""
     if remat and torch.is_grad_enabled() and res.requires_grad:
         from torch.utils.checkpoint import checkpoint
-        return checkpoint(blk.forward_fused, res, delta,
-                          use_reentrant=False)
-    return blk.forward_fused(res, delta)
+        return checkpoint(blk, res, delta, use_reentrant=False)
+    return blk(res, delta)
 
 
 class GPTModel(nn.Module):
@@ -226,8 +243,7 @@ class GPTModel(nn.Module):
         delta = None
         for blk in self.blocks:
             res, delta = _run_block(blk, res, delta, self.cfg.remat)
-        _, x = ops.add_layer_norm(res, delta, self.ln_f.weight,
-                                  self.ln_f.bias, self.ln_f.eps)
+        x = self.ln_f(res, delta)
         return self.lm_head(x)
 
     def loss(self, ids: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
@@ -305,8 +321,7 @@ class GPTStage(nn.Module):
         if not self.is_last:
             # materialize the stream value at the stage boundary
             return res + delta if delta is not None else res
-        _, x = ops.add_layer_norm(res, delta, self.ln_f.weight,
-                                  self.ln_f.bias, self.ln_f.eps)
+        x = self.ln_f(res, delta)
         logits = self.lm_head(x)
         N = logits.shape[0] * logits.shape[1]
         logits = logits.reshape(N, -1)

@@ -39,6 +39,7 @@ class _BlockShard:
 
     def __init__(self, module: nn.Module, dp: int, rank_idx: int):
         self.module = module
+        self.rank_idx = rank_idx
         self.params = [p for p in module.parameters() if p.requires_grad]
         self.shapes = [p.shape for p in self.params]
         self.numels = [p.numel() for p in self.params]
@@ -88,6 +89,14 @@ class _BlockShard:
 
     def release(self):
         self.full.untyped_storage().resize_(0)
+
+    def writeback_shard(self):
+        """Copy this rank's slice of the (gathered) full buffer back into
+        the authoritative shard — needed after external writes into the
+        param views (checkpoint restore)."""
+        with torch.no_grad():
+            self.shard.copy_(self.full[self.rank_idx * self.shard_n:
+                                       (self.rank_idx + 1) * self.shard_n])
 
     def prepare_grads(self):
         if self.grad_flat.untyped_storage().nbytes() == 0:
@@ -161,6 +170,25 @@ class Zero3Manager:
     def zero_grads(self):
         for b in self.blocks:
             b.grad_shard.zero_()
+
+    def materialized(self):
+        """Context manager: gather every block's params (so external code
+        like checkpoint save/restore can read/write them), write the
+        shards back and release on exit."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def cm():
+            for b in self.blocks:
+                b.gather(self.group)
+            try:
+                yield
+            finally:
+                for b in self.blocks:
+                    b.writeback_shard()
+                    b.release()
+
+        return cm()
 
     def gathered_bytes(self) -> int:
         """Currently-resident gathered bytes (memory assertion hook)."""

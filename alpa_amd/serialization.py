@@ -89,9 +89,11 @@ def save_checkpoint(ckpt_dir: str, state_tree: Any, step: int,
         tdir = os.path.join(ckpt_dir, dirname)
         os.makedirs(tdir, exist_ok=True)
         if spec.is_writer:
+            # clone before .numpy(): a shared export would pin the
+            # source storage (ZeRO-3 resizes param storages to 0)
             arr = leaf.detach().to(torch.float32).cpu().numpy() \
                 if leaf.dtype == torch.bfloat16 \
-                else leaf.detach().cpu().numpy()
+                else leaf.detach().cpu().clone().numpy()
             shard_name = f"shard_{host}.0"
             np.save(os.path.join(tdir, shard_name + ".npy"), arr)
             meta = {
@@ -288,20 +290,32 @@ def _train_state_tree_and_specs(state):
     return tree, full_specs
 
 
+def _maybe_materialized(state):
+    """ZeRO-3 states keep param storages released between uses; gather
+    them around checkpoint IO (and write shards back on exit)."""
+    mgr = getattr(state, "zero3_manager", None)
+    if mgr is not None:
+        return mgr.materialized()
+    import contextlib
+    return contextlib.nullcontext()
+
+
 def save_train_state(ckpt_dir: str, state, step: Optional[int] = None
                      ) -> None:
     """Save model + optimizer of a TrainState (reference
     save_checkpoint, serialization.py:75)."""
     step = step if step is not None else state.step_count
-    tree, full_specs = _train_state_tree_and_specs(state)
-    save_checkpoint(ckpt_dir, tree, step, full_specs)
+    with _maybe_materialized(state):
+        tree, full_specs = _train_state_tree_and_specs(state)
+        save_checkpoint(ckpt_dir, tree, step, full_specs)
 
 
 def restore_train_state(ckpt_dir: str, state, step: int) -> None:
     """Restore in place, resharding to this rank's placement (reference
     restore_checkpoint, serialization.py:137)."""
-    tree, full_specs = _train_state_tree_and_specs(state)
-    loaded = restore_checkpoint(ckpt_dir, step, tree, full_specs)
+    with _maybe_materialized(state):
+        tree, full_specs = _train_state_tree_and_specs(state)
+        loaded = restore_checkpoint(ckpt_dir, step, tree, full_specs)
     if isinstance(loaded.get("step"), int):
         state.optimizer.step_count = loaded["step"]
         state.step_count = loaded["step"]
