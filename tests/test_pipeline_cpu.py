@@ -468,3 +468,62 @@ def test_training_dp_no_worse_than_uniform():
     dp = training_dp_search(8, 16, flops, boundary_act_bytes=1e7,
                             layer_param_bytes=[1e9] * 8, db=db)
     assert dp[3] <= uni[3] * 1.05, (dp[3], uni[3])
+
+
+CFG4 = GPTConfig(hidden_size=64, num_layers=4, num_heads=4, seq_len=32,
+                 vocab_size=96)
+
+
+def _stage_builder4(layer_range, is_first, is_last, mesh, axis, dtype,
+                    device):
+    from alpa_amd.models.gpt import GPTStage
+    return GPTStage(CFG4, layer_range, is_first, is_last, mesh, axis,
+                    dtype, device, init_seed=11)
+
+
+def _auto_uneven_worker(rank, world_size):
+    # planner inputs that make the training DP choose 2 stages with
+    # UNEVEN layer ranges (param-heavy tail forces tp on fewer layers)
+    method = aa.PipeshardParallel(num_micro_batches=2,
+                                  stage_option="auto",
+                                  memory_budget_per_device=100e9)
+    spec = gpt_pipeline_spec(CFG4)
+    spec.build_stage = _stage_builder4
+    spec.layer_flops = [1e12, 1e12, 2e12, 2e12]
+    spec.layer_param_bytes = [1e8, 1e8, 20e9, 20e9]
+    spec.boundary_act_bytes = 1e8
+    state = aa.TrainState.create(spec, method, lr=1e-3)
+    # the DP's ranges must have reached the engine (not re-clustered)
+    ranges = method._auto_layer_ranges
+    assert ranges[0] != ranges[-1] and ranges[-1][1] == 4, ranges
+    step = aa.parallelize(lambda m, b: None, method=method)
+    g = torch.Generator().manual_seed(0)
+    ids = torch.randint(0, 96, (4, 32), generator=g)
+    return [float(step(state, {"ids": ids, "labels": ids}))
+            for _ in range(2)], ranges
+
+
+def test_auto_search_uneven_ranges_end_to_end():
+    """The training DP's (uneven) layer ranges travel with its submeshes
+    into the compiled pipeline; training still matches the serial
+    oracle."""
+    serial = GPTModel(CFG4, None, 1, torch.float32, None, init_seed=11)
+    method = aa.ShardParallel(num_micro_batches=2,
+                              logical_mesh_shape=(1, 1))
+    state = aa.TrainState.create(
+        lambda mesh=None, axis=1, dtype=torch.float32, device=None:
+        GPTModel(CFG4, mesh, axis, dtype, device, init_seed=11),
+        method, lr=1e-3)
+    sstep = aa.parallelize(lambda m, b: m.loss(b["ids"], b["labels"]),
+                           method=method)
+    g = torch.Generator().manual_seed(0)
+    ids = torch.randint(0, 96, (4, 32), generator=g)
+    want = [float(sstep(state, {"ids": ids, "labels": ids}))
+            for _ in range(2)]
+    results = run_distributed(_auto_uneven_worker, world_size=4,
+                              timeout=300)
+    for losses, ranges in results:
+        assert ranges[0][1] - ranges[0][0] != ranges[-1][1] - \
+            ranges[-1][0], ranges  # genuinely uneven
+        for a, b in zip(losses, want):
+            assert abs(a - b) < 2e-4, (losses, want)
