@@ -48,3 +48,17 @@ def _bcast_worker(rank, world_size):
 def test_mesh_broadcast():
     res = run_distributed(_bcast_worker, world_size=2)
     assert all(float(r) == 2.0 for r in res)
+
+
+def test_virtual_mesh_slicing():
+    """VirtualMesh (reference VirtualPhysicalMesh slice_2d:1888): slice
+    an unallocated mesh into submeshes and reshape without creating
+    process groups."""
+    from alpa_amd.mesh import VirtualMesh, full_virtual_mesh
+    vm = VirtualMesh(tuple(range(8)), (2, 4))
+    sub = vm.slice_ranks(0, 4)
+    assert sub.ranks == (0, 1, 2, 3) and sub.num_devices == 4
+    r = sub.reshape((2, 2))
+    assert r.shape == (2, 2) and r.ranks == (0, 1, 2, 3)
+    assert r.rank_grid().tolist() == [[0, 1], [2, 3]]
+    assert full_virtual_mesh(8).num_devices == 8
