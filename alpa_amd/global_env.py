@@ -38,8 +38,10 @@ class GlobalConfig:
     use_hip_kernels: bool = True
     #: fail loudly if running on GPU without the HIP extension (anti-silent-fallback)
     require_hip_kernels_on_gpu: bool = True
-    #: capture steady-state step in a hipGraph when possible
-    use_hip_graphs: bool = False
+    #: capture the steady-state step in a hipGraph when possible
+    #: (single mesh, no collectives inside the graph; bench.py consults
+    #: this in addition to its --no-hipgraph flag)
+    use_hip_graphs: bool = True
 
     # ---------- collectives ----------
     #: bucket size for gradient all-reduce / reduce-scatter (bytes).

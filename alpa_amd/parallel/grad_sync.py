@@ -131,7 +131,9 @@ class GradSynchronizer:
         if b.no_sync or self.dp == 1 or not is_distributed():
             return
         group = self.mesh.axis_group(self.axis)
-        stream = self.mesh.comm_stream
+        from ..global_env import global_config
+        stream = self.mesh.comm_stream \
+            if global_config.overlap_grad_sync else None
         if stream is not None:
             # comm stream waits for the producing compute work
             stream.wait_stream(torch.cuda.current_stream())

@@ -165,7 +165,9 @@ def main():
     # execution mode: after capture, one replay = fwd+bwd x nmb + fused
     # AdamW with zero host-side dispatch.
     graph = None
-    if on_gpu and aa.world_size() == 1 and not args.no_hipgraph:
+    from alpa_amd.global_env import global_config as _gc
+    if (on_gpu and aa.world_size() == 1 and not args.no_hipgraph
+            and _gc.use_hip_graphs):
         try:
             torch.cuda.synchronize()
             graph = torch.cuda.CUDAGraph()

@@ -210,3 +210,26 @@ def test_zero3_with_remat_matches_serial():
     ref = run(cfg_p, aa.ShardParallel(logical_mesh_shape=(1, 1)))
     for a, b in zip(got, ref):
         assert abs(a - b) < 1e-5, (got, ref)
+
+
+def _dp_nooverlap_worker(rank, world_size):
+    from alpa_amd.global_env import global_config
+    global_config.overlap_grad_sync = False
+    try:
+        return _dp_worker(rank, world_size, 1, (world_size, 1))
+    finally:
+        global_config.overlap_grad_sync = True
+
+
+def test_dp2_without_overlap_matches_serial():
+    """global_config.overlap_grad_sync=False takes the synchronous
+    (async_op) collective path; results identical."""
+    serial_losses, serial_params = run_serial(num_micro_batches=1)
+    results = run_distributed(_dp_nooverlap_worker, world_size=2)
+    for i in range(len(serial_losses)):
+        avg = sum(r[0][i] for r in results) / len(results)
+        assert abs(avg - serial_losses[i]) < 1e-5
+    for _, params in results:
+        for p, sp in zip(params, serial_params):
+            torch.testing.assert_close(torch.as_tensor(p), sp, rtol=1e-5,
+                                       atol=1e-6)
