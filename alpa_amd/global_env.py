@@ -68,7 +68,6 @@ class GlobalConfig:
     collect_trace: bool = False
 
     # ---------- benchmark ----------
-    use_dummy_value_for_benchmarking: bool = False
 
     # ---------- paths ----------
     prof_database_path: str = field(

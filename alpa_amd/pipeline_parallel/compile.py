@@ -14,7 +14,8 @@ import numpy as np
 import torch
 
 from ..global_env import global_config
-from ..mesh import DeviceMesh, device, get_device_mesh, rank, world_size
+from ..mesh import (DeviceMesh, device, get_device_mesh,
+                    is_distributed, rank, world_size)
 from ..parallel.grad_sync import GradSynchronizer
 from .layer_clustering import cluster_layers, uniform_layer_costs
 from .runtime import PipelineEngine
@@ -44,6 +45,11 @@ def resolve_stage_layout(method, n: Optional[int] = None,
             from .stage_construction import (profiled_stage_search,
                                              training_dp_search)
             budget = getattr(method, "memory_budget_per_device", None)
+            if budget is None and torch.cuda.is_available():
+                # usable fraction of this device's HBM (the reference's
+                # XLA memory-fraction knob)
+                free, total = torch.cuda.mem_get_info()
+                budget = global_config.memory_fraction * total
             got = training_dp_search(
                 n, method.num_micro_batches, spec.layer_flops,
                 spec.boundary_act_bytes, spec.layer_param_bytes,
@@ -172,6 +178,14 @@ def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
     # grad scale divides by the LAST stage's dp (the loss-definition dp);
     # equals every stage's dp in the uniform case
     engine.loss_dp = stage_shapes[-1][0]
+
+    if global_config.collect_trace:
+        engine.enable_tracing()
+    if global_config.pipeline_check_alive and is_distributed():
+        # fail fast (within the dist timeout) if any rank died during
+        # stage construction — detection-only, like the reference
+        import torch.distributed as dist
+        dist.barrier()
 
     # tied-weight cross-stage allreduce groups (reference N15): one group
     # per (tied set x mesh coordinate); every rank creates every group

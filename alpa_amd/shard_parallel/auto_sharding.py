@@ -147,10 +147,14 @@ def solve_gpt_sharding(num_devices: int, hidden: int, layers: int,
                        vocab: int, tokens: int,
                        memory_budget: Optional[float] = None,
                        force_data_parallel: bool = False,
-                       time_limit: float = 30.0) -> ShardingPlan:
+                       time_limit: Optional[float] = None) -> ShardingPlan:
     """Outer loop over logical mesh shapes + inner ILP (reference
     compile_shard_executable tries logical shapes; ILP scores each)."""
     from .mesh_search import factorizations
+    if time_limit is None:
+        # per-mesh-shape slice of the configured total (reference uses a
+        # 600 s PuLP limit, auto_sharding.py:829-836)
+        time_limit = min(30.0, global_config.solver_timeout / 8)
     best: Optional[ShardingPlan] = None
     for (d0, d1) in factorizations(num_devices):
         if force_data_parallel and d1 != 1:
