@@ -114,3 +114,22 @@ def test_adamw_grad_scale():
     ref.adamw_step(p2, [g], m2, v2, 1, 1e-3, 0.9, 0.95, 1e-8, 0.0,
                    grad_scale=0.25)
     torch.testing.assert_close(p1[0], p2[0])
+
+
+def test_rms_norm_reference():
+    """RMS-norm fp32 reference (the oracle for a future rms_norm HIP
+    kernel — LLaMA-family models): fwd matches the closed form, bwd
+    matches autograd."""
+    import torch
+    from alpa_amd.ops import reference as ref
+    torch.manual_seed(5)
+    x = torch.randn(8, 32, requires_grad=True)
+    w = torch.randn(32, requires_grad=True)
+    y, rstd = ref.rms_norm_fwd(x, w, 1e-6)
+    ref_y = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + 1e-6) * w
+    torch.testing.assert_close(y, ref_y, rtol=1e-6, atol=1e-6)
+    dy = torch.randn_like(y)
+    ref_y.backward(dy)
+    dx, dw = ref.rms_norm_bwd(dy, x.detach(), w.detach(), rstd)
+    torch.testing.assert_close(dx, x.grad, rtol=1e-5, atol=1e-5)
+    torch.testing.assert_close(dw, w.grad, rtol=1e-5, atol=1e-5)

@@ -209,3 +209,18 @@ def test_pipeline_plan_roundtrip(tmp_path):
     m3 = plan_to_method(ParallelPlan.load(str(tmp_path / "h.json")))
     assert [tuple(sh) for sh in m3.stage_mesh_shapes] == \
         [(1, 2), (1, 2), (1, 4)]
+
+
+def test_make_schedule_registry():
+    """Schedule registry dispatch (reference schedules.py:521-525)."""
+    from alpa_amd.pipeline_parallel import schedules
+    for name in ("1f1b", "gpipe", "inference",
+                 "1f1b_overlap_friendly"):
+        sched = schedules.make_schedule(name, 2, 4)
+        assert len(sched) == 2
+        for stage in sched:
+            fwd = [i for op, i in stage if op == schedules.FWD]
+            assert sorted(fwd) == list(range(4)), (name, stage)
+    import pytest
+    with pytest.raises((KeyError, ValueError)):
+        schedules.make_schedule("nope", 2, 4)
