@@ -6,7 +6,9 @@ Driver contract:
 
 Measures Alpa's headline metric on Alpa's headline config: model TFLOPS for
 GPT-2.6B (H=2560, L=32, heads=32, vocab 51200, seq 1024), batch 32 per GPU,
-num_micro_batches=4, bf16, synthetic data, random-init weights.  TFLOPS
+bf16, synthetic data, random-init weights.  Microbatching defaults to 1
+(no pipeline here; the reference's nmb=4 belongs to its dp2xop2xpp2
+plan — see --nmb).  TFLOPS
 accounting follows the reference formula exactly
 (``benchmark/alpa/util.py:65-89``):
   factor*B*S*H^2*L*(1+S/(6H)) + 6*B*S*H*V, factor=72 fwd+bwd (96 w/ remat).
