@@ -85,6 +85,7 @@ class ExpertParallelMLP(nn.Module):
         super().__init__()
         self.mesh, self.axis = mesh, axis
         self.ep = mesh.axis_size(axis) if mesh is not None else 1
+        assert num_experts >= 2, "top-2 gating needs >= 2 experts"
         assert num_experts % self.ep == 0
         self.E = num_experts
         self.e_local = num_experts // self.ep
