@@ -32,6 +32,19 @@ def shard_batch(batch: Any, dp: int, dp_idx: int) -> Any:
     return batch
 
 
+def _pin(batch: Any) -> Any:
+    """Pin host tensors so the H2D copy can actually run async —
+    non_blocking from PAGEABLE memory silently degrades to a
+    synchronous copy."""
+    if torch.is_tensor(batch):
+        return batch.pin_memory() if not batch.is_pinned() else batch
+    if isinstance(batch, dict):
+        return {k: _pin(v) for k, v in batch.items()}
+    if isinstance(batch, (list, tuple)):
+        return type(batch)(_pin(v) for v in batch)
+    return batch
+
+
 def _to_device(batch: Any, device, non_blocking=True) -> Any:
     if torch.is_tensor(batch):
         return batch.to(device, non_blocking=non_blocking)
@@ -80,13 +93,16 @@ class DataLoader:
             b = shard_batch(b, self.dp, self.dp_idx)
             if self.device is not None:
                 if copy_stream is not None:
+                    host = _pin(b)
                     with torch.cuda.stream(copy_stream):
-                        b = _to_device(b, self.device)
+                        b = _to_device(host, self.device)
                     ev = torch.cuda.Event()
                     ev.record(copy_stream)
-                    return (b, ev)
+                    # keep the pinned host tensors alive until the copy
+                    # is consumed (async H2D reads them after return)
+                    return (b, ev, host)
                 b = _to_device(b, self.device)
-            return (b, None)
+            return (b, None, None)
 
         for _ in range(self.prefetch_size):
             item = load_next()
@@ -94,7 +110,7 @@ class DataLoader:
                 break
             queue.append(item)
         while queue:
-            b, ev = queue.popleft()
+            b, ev, _host = queue.popleft()
             nxt = load_next()
             if nxt is not None:
                 queue.append(nxt)
