@@ -1,3 +1,4 @@
-from .controller import Controller, run_controller
+from .controller import (Controller, SpmdGenerateService,
+                         run_controller)
 
-__all__ = ["Controller", "run_controller"]
+__all__ = ["Controller", "SpmdGenerateService", "run_controller"]

@@ -82,3 +82,55 @@ def run_controller(controller: Controller, host: str = "127.0.0.1",
     import uvicorn
     uvicorn.run(controller.asgi_app(), host=host, port=port,
                 log_level="warning")
+
+
+# ---------------------------------------------------------------------------
+# SPMD serving: rank 0 owns HTTP, other ranks execute the same generate
+# calls (reference: DeviceMeshGroupManager replica actors execute driver
+# RPCs, controller.py:59)
+# ---------------------------------------------------------------------------
+
+
+def _bcast(obj):
+    import torch.distributed as dist
+    lst = [obj]
+    dist.broadcast_object_list(lst, src=0)
+    return lst[0]
+
+
+class SpmdGenerateService:
+    """Wraps a generate fn so every rank of a TP group runs it in
+    lockstep: rank 0 (driving HTTP) broadcasts each request's inputs;
+    other ranks sit in `serve_worker_loop` executing them.  A `None`
+    broadcast shuts the workers down."""
+
+    def __init__(self, generate_fn):
+        from ..mesh import is_distributed, rank
+        self.generate_fn = generate_fn
+        self._dist = is_distributed()
+        self._is_driver = rank() == 0
+
+    def __call__(self, ids, max_tokens, num_beams: int = 1):
+        assert self._is_driver
+        if self._dist:
+            _bcast(("gen", ids.cpu(), int(max_tokens), int(num_beams)))
+        return self._run(ids, max_tokens, num_beams)
+
+    def _run(self, ids, max_tokens, num_beams):
+        if num_beams > 1:
+            return self.generate_fn(ids, max_tokens, num_beams=num_beams)
+        return self.generate_fn(ids, max_tokens)
+
+    def shutdown_workers(self):
+        if self._dist and self._is_driver:
+            _bcast(None)
+
+    def serve_worker_loop(self):
+        """Non-zero ranks: execute broadcast requests until shutdown."""
+        assert not self._is_driver
+        while True:
+            msg = _bcast(None)
+            if msg is None:
+                return
+            _, ids, max_tokens, num_beams = msg
+            self._run(ids, max_tokens, num_beams)

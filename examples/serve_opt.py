@@ -15,7 +15,8 @@ sys.path.insert(0, os.path.join(os.path.dirname(os.path.abspath(__file__)),
                                 ".."))
 import alpa_amd as aa
 from alpa_amd.models.opt import OPTModel, opt_config
-from alpa_amd.serve import Controller, run_controller
+from alpa_amd.serve import (Controller, SpmdGenerateService,
+                            run_controller)
 
 
 def main():
@@ -43,13 +44,21 @@ def main():
             aa.shutdown()
             return
     c = Controller()
+
     def _gen(ids, mt, num_beams=1):
         ids = ids.to(aa.device())
         if num_beams > 1:
             return model.beam_search(ids, mt, num_beams=num_beams)
         return model.generate(ids, mt)
 
-    c.register_model(f"opt-{args.model}", _gen)
+    svc = SpmdGenerateService(_gen)
+    if aa.rank() != 0:
+        # non-zero TP ranks execute the driver's broadcast requests
+        svc.serve_worker_loop()
+        aa.shutdown()
+        return
+
+    c.register_model(f"opt-{args.model}", svc)
     run_controller(c, port=args.port)
 
 

@@ -475,3 +475,33 @@ def test_bloom_tp2_checkpoint_serial_restore(tmp_path):
     for n, p in m.named_parameters():
         torch.testing.assert_close(p.detach(), ref[n], rtol=1e-6,
                                    atol=1e-6, msg=lambda msg: f"{n}: {msg}")
+
+
+def _spmd_serve_worker(rank, world_size):
+    from alpa_amd.serve import Controller, SpmdGenerateService
+    mesh = aa.mesh.full_mesh((1, world_size))
+    m = build_opt(mesh, axis=1)
+    svc = SpmdGenerateService(lambda ids, mt: m.generate(ids, mt))
+    torch.manual_seed(31)
+    ids = torch.randint(0, CFG.vocab_size, (1, 5))
+    if rank == 0:
+        c = Controller()
+        c.register_model("opt", svc)
+        out = c.completions("opt", ids.tolist(), max_tokens=4)
+        svc.shutdown_workers()
+        return out
+    svc.serve_worker_loop()
+    return None
+
+
+def test_spmd_serving_tp2():
+    """Rank 0 drives the controller; worker ranks execute the same
+    generate via broadcast (reference DeviceMeshGroupManager replicas) —
+    output equals the serial model."""
+    torch.manual_seed(31)
+    ids = torch.randint(0, CFG.vocab_size, (1, 5))
+    want = build_opt().generate(ids, 4).tolist()
+    results = run_distributed(_spmd_serve_worker, world_size=2,
+                              timeout=300)
+    outs = [r for r in results if r is not None]
+    assert len(outs) == 1 and outs[0] == want
