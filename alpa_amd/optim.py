@@ -51,9 +51,17 @@ class AdamW:
         self.step_count += 1
         from .ops import fp8 as _fp8
         _fp8.bump_epoch()  # invalidate fp8 quantized-weight caches
-        ops.fused_adamw(self.params, grads, self.exp_avgs, self.exp_avg_sqs,
-                        self.step_count, self.lr, self.beta1, self.beta2,
-                        self.eps, self.weight_decay, grad_scale)
+        # params without a grad this step (e.g. an unused head when
+        # training without the grad-sync bucket views) are skipped
+        quads = [(p, g, m, v) for p, g, m, v in
+                 zip(self.params, grads, self.exp_avgs, self.exp_avg_sqs)
+                 if g is not None]
+        if not quads:
+            return
+        ps, gs, ms, vs = (list(t) for t in zip(*quads))
+        ops.fused_adamw(ps, gs, ms, vs, self.step_count, self.lr,
+                        self.beta1, self.beta2, self.eps,
+                        self.weight_decay, grad_scale)
 
     def step_sharded(self, grad_sync, grad_scale: float = 1.0):
         """ZeRO-2 path: installed by parallel/zero.py (sharded state over
