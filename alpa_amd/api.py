@@ -129,6 +129,10 @@ class TrainState:
 
             def __init__(self):
                 self.step_count = 0
+                # exposed for checkpointing (serialization.py saves the
+                # sharded moments of both halves)
+                self.z3 = z3
+                self.rest_opt = rest_opt
 
             def step(self, grads=None, grad_scale: float = 1.0):
                 z3.step(grad_scale=grad_scale)
@@ -219,9 +223,24 @@ class ParallelizedFunc:
         if ds is not None:
             # fold the loss-scale unscale into the fused AdamW launch;
             # overflow => skip the update + back off (reference
-            # DynamicScale, model_util.py)
-            if ds.update(ds.found_inf(
-                    p.grad for p in state.model.parameters())):
+            # DynamicScale, model_util.py).  The flag is made globally
+            # consistent with an all-reduce MAX over the FULL mesh group:
+            # under TP the per-rank weight-shard grads differ, and the
+            # reference computes the is-finite reduction inside the SPMD
+            # program for the same reason.  A local non-finite grad
+            # implies a non-finite reduced grad (inf/nan propagate
+            # through the sum), so scanning local views stays sound
+            # under ZeRO-2 bucket sharding too.
+            local_inf = ds.found_inf(
+                p.grad for p in state.model.parameters())
+            if is_distributed() and state.mesh is not None:
+                import torch.distributed as dist
+                flag = torch.tensor(
+                    [1.0 if local_inf else 0.0],
+                    device=device() if torch.cuda.is_available() else "cpu")
+                state.mesh.all_reduce(flag, axis=None, op=dist.ReduceOp.MAX)
+                local_inf = bool(flag.item() > 0)
+            if ds.update(local_inf):
                 gs.zero_grads()
                 state.step_count += 1
                 return total_loss / nmb

@@ -21,6 +21,13 @@ class AdamW:
 
     `grads` may be supplied (e.g. views into flat grad-sync buckets) or
     defaults to `p.grad`.
+
+    Known deviation from torch.optim.AdamW: bias correction uses ONE
+    global step counter.  A param that skips steps (no grad) and later
+    rejoins is corrected with the global t — slightly weaker correction
+    for its first real updates.  Training steps through `parallelize`
+    always populate every grad via the bucket views, so the deviation
+    only affects the raw-optimizer path with genuinely unused params.
     """
 
     def __init__(self, params: Iterable[torch.nn.Parameter], lr: float = 1e-4,

@@ -76,9 +76,37 @@ class ZeroOptimizer:
     def params(self):
         return self.gs.params
 
+    def moment_slices(self):
+        """Per-parameter views of the sharded Adam moments, for
+        checkpointing in the reference's reshard-on-load format
+        (serialization.py): yields (param, lo, hi, m_view, v_view) where
+        [lo, hi) is this rank's owned range within the param's OWN flat
+        index space and the views alias exp_avgs/exp_avg_sqs."""
+        out = []
+        for bi, b in enumerate(self.gs.buckets):
+            total = b.flat.numel()
+            if self.gs.reduce_scatter:
+                shard_n = total // self.dp
+                s_lo = self.rank_idx * shard_n
+                s_hi = s_lo + shard_n
+            else:
+                s_lo, s_hi = 0, total
+            off = 0
+            for p in b.params:
+                n = p.numel()
+                a, c = max(off, s_lo), min(off + n, s_hi)
+                if a < c:
+                    out.append((p, a - off, c - off,
+                                self.exp_avgs[bi][a - s_lo:c - s_lo],
+                                self.exp_avg_sqs[bi][a - s_lo:c - s_lo]))
+                off += n
+        return out
+
     @torch.no_grad()
     def step(self, grads=None, grad_scale: float = 1.0):
         self.step_count += 1
+        from ..ops import fp8 as _fp8
+        _fp8.bump_epoch()  # invalidate fp8 quantized-weight caches
         ops.fused_adamw(self.param_shards, self.grad_shards, self.exp_avgs,
                         self.exp_avg_sqs, self.step_count, self.lr,
                         self.beta1, self.beta2, self.eps, self.weight_decay,

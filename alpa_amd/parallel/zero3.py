@@ -213,9 +213,26 @@ class Zero3Optimizer:
         self.exp_avg_sqs = [torch.zeros_like(p, dtype=torch.float32)
                             for p in pshards]
 
+    def moment_slices(self):
+        """Per-parameter views of the sharded moments (see
+        ZeroOptimizer.moment_slices): (param, lo, hi, m_view, v_view)."""
+        out = []
+        for bi, blk in enumerate(self.m.blocks):
+            s_lo = blk.rank_idx * blk.shard_n
+            s_hi = s_lo + blk.shard_n
+            for p, off, n in zip(blk.params, blk.offsets, blk.numels):
+                a, c = max(off, s_lo), min(off + n, s_hi)
+                if a < c:
+                    out.append((p, a - off, c - off,
+                                self.exp_avgs[bi][a - s_lo:c - s_lo],
+                                self.exp_avg_sqs[bi][a - s_lo:c - s_lo]))
+        return out
+
     @torch.no_grad()
     def step(self, grads=None, grad_scale: float = 1.0):
         self.step_count += 1
+        from ..ops import fp8 as _fp8
+        _fp8.bump_epoch()  # invalidate fp8 quantized-weight caches
         pshards, gshards = self.m.shards()
         ops.fused_adamw(pshards, gshards, self.exp_avgs, self.exp_avg_sqs,
                         self.step_count, self.lr, self.beta1, self.beta2,

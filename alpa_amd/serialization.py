@@ -116,12 +116,31 @@ def save_checkpoint(ckpt_dir: str, state_tree: Any, step: int,
     if host == 0 or not is_distributed():
         with open(os.path.join(ckpt_dir, f"checkpoint_{step}"), "wb") as f:
             f.write(msgpack.packb(packed_tree, use_bin_type=True))
+    if is_distributed():
+        # a checkpoint is complete only when every rank's shards (and rank
+        # 0's tree) are on disk — ranks may restore immediately after
+        import torch.distributed as dist
+        dist.barrier()
 
 
 def _load_slice(tdir: str, want_index: Tuple[slice, ...],
                 global_shape, np_dtype) -> np.ndarray:
     """Assemble want_index of the global tensor from saved shards
     (resharding on load)."""
+    # layout conversion: when the SAVED global shape differs from the
+    # requested one but the element count matches (serial p-shaped Adam
+    # moments vs ZeRO's flat per-param ranges), assemble the full stored
+    # tensor and reindex through a reshape.
+    metas = sorted(f for f in os.listdir(tdir) if f.startswith("metadata_"))
+    if metas and global_shape is not None:
+        with open(os.path.join(tdir, metas[0]), "rb") as f:
+            stored_gs = tuple(pickle.load(f)["global_shape"])
+        if stored_gs != tuple(global_shape) and \
+                int(np.prod(stored_gs)) == int(np.prod(global_shape)):
+            full = _load_slice(tdir, tuple(slice(0, s) for s in stored_gs),
+                               stored_gs, np_dtype)
+            return np.ascontiguousarray(
+                full.reshape(tuple(global_shape))[tuple(want_index)])
     out_shape = tuple(s.stop - s.start for s in want_index)
     out = np.zeros(out_shape, dtype=np_dtype)
     filled = np.zeros(out_shape, dtype=bool) if out.size else None
@@ -272,21 +291,63 @@ def model_shard_specs(model: torch.nn.Module,
 
 
 def _train_state_tree_and_specs(state):
-    """(tree, specs) for a TrainState: params + AdamW moments (the moments
-    share the parameter sharding).  ZeRO-2's bucket-sharded moments are
-    topology-specific and handled by its own state_dict."""
+    """(tree, specs) for a TrainState: params + Adam moments.
+
+    Raw AdamW moments share the parameter sharding (2-D specs, TP-
+    reshardable).  ZeRO-2/3 moments are bucket/block flat shards: they are
+    saved per parameter as 1-D ranges of the param's own flat index space
+    (ShardSpec over (numel,)), which reshards correctly across dp sizes
+    and — via the layout-conversion fallback in ``_load_slice`` — to and
+    from the serial p-shaped layout.  TP-sharded params under ZeRO get a
+    ``@tp<offsets>`` key qualifier (per-tp-rank flat spaces are disjoint);
+    such moments restore only under the same tp topology.
+    """
     specs = model_shard_specs(state.model, getattr(state, "mesh", None))
-    names = [n for n, _ in state.model.named_parameters()]
     tree = {"params": dict(state.model.state_dict()),
             "step": state.step_count}
     full_specs = {f"params.{k}": v for k, v in specs.items()}
-    opt_sd = state.optimizer.state_dict()
-    if "exp_avgs" in opt_sd and len(opt_sd["exp_avgs"]) == len(names):
-        tree["opt_m"] = dict(zip(names, opt_sd["exp_avgs"]))
-        tree["opt_v"] = dict(zip(names, opt_sd["exp_avg_sqs"]))
-        for n in names:
+    name_of = {id(p): n for n, p in state.model.named_parameters()}
+    opt = state.optimizer
+
+    def add_dense(adam):
+        m_tree = tree.setdefault("opt_m", {})
+        v_tree = tree.setdefault("opt_v", {})
+        for p, m, v in zip(adam.params, adam.exp_avgs, adam.exp_avg_sqs):
+            n = name_of.get(id(p))
+            if n is None:
+                continue
+            m_tree[n], v_tree[n] = m, v
             full_specs[f"opt_m.{n}"] = specs[n]
             full_specs[f"opt_v.{n}"] = specs[n]
+
+    def add_sharded(z):
+        m_tree = tree.setdefault("opt_m", {})
+        v_tree = tree.setdefault("opt_v", {})
+        for (p, lo, hi, m, v) in z.moment_slices():
+            n = name_of.get(id(p))
+            if n is None:
+                continue
+            key = n
+            sp = specs.get(n)
+            if sp is not None and sp.index != tuple(
+                    slice(0, s) for s in sp.global_shape):
+                starts = "_".join(str(s.start) for s in sp.index)
+                key = f"{n}@tp{starts}"
+            m_tree[key], v_tree[key] = m, v
+            fsp = ShardSpec((p.numel(),), (slice(lo, hi),), True)
+            full_specs[f"opt_m.{key}"] = fsp
+            full_specs[f"opt_v.{key}"] = fsp
+
+    if opt is None:
+        pass
+    elif hasattr(opt, "moment_slices"):     # ZeroOptimizer (ZeRO-2)
+        add_sharded(opt)
+    elif hasattr(opt, "z3"):                # ZeRO-3 composite
+        add_sharded(opt.z3)
+        if opt.rest_opt is not None:
+            add_dense(opt.rest_opt)
+    elif hasattr(opt, "exp_avgs") and hasattr(opt, "params"):
+        add_dense(opt)                      # raw AdamW (incl. pipeline)
     return tree, full_specs
 
 
@@ -317,5 +378,12 @@ def restore_train_state(ckpt_dir: str, state, step: int) -> None:
         tree, full_specs = _train_state_tree_and_specs(state)
         loaded = restore_checkpoint(ckpt_dir, step, tree, full_specs)
     if isinstance(loaded.get("step"), int):
-        state.optimizer.step_count = loaded["step"]
-        state.step_count = loaded["step"]
+        st = loaded["step"]
+        state.optimizer.step_count = st
+        # ZeRO-3 composite: bias correction is driven by the nested
+        # optimizers' own counters
+        for sub in ("z3", "rest_opt"):
+            o = getattr(state.optimizer, sub, None)
+            if o is not None:
+                o.step_count = st
+        state.step_count = st

@@ -32,8 +32,12 @@ def main():
 
     aa.init()
     n = aa.world_size()
-    # experts sharded over the model axis (EP); dp=1 within this example
-    method = aa.ShardParallel(logical_mesh_shape=(1, n))
+    # (n, 1) mesh: data-parallel over axis 0 with the experts sharded over
+    # the SAME axis (ep_axis=0, DeepSpeed-MoE style EP-within-DP) — each
+    # rank feeds its own batch and the top-2 dispatch all-to-alls tokens
+    # to the expert owners; dense block grads all-reduce over dp while
+    # expert grads stay local (marked _expert_parallel).
+    method = aa.ShardParallel(logical_mesh_shape=(n, 1))
     dtype = torch.bfloat16 if torch.cuda.is_available() else torch.float32
     cfg = MoEConfig(hidden_size=args.hidden, num_layers=args.layers,
                     num_heads=args.heads, seq_len=args.seq,
@@ -46,8 +50,11 @@ def main():
     step = aa.parallelize(lambda m, b: m.loss(b["ids"], b["labels"]),
                           method=method)
 
-    per_rank = args.batch  # EP shards experts, not the batch
-    g = torch.Generator().manual_seed(7)
+    per_rank = args.batch
+    # distinct batch per dp rank — the printed tokens/s below counts the
+    # DISTINCT global batch (per_rank * n), which is only honest when the
+    # ranks' data actually differs
+    g = torch.Generator().manual_seed(7 + aa.rank())
 
     def make_batch():
         ids = torch.randint(0, cfg.vocab_size, (per_rank, cfg.seq_len),
@@ -66,10 +73,11 @@ def main():
     dt = (time.perf_counter() - t0) / args.steps
     if aa.rank() == 0:
         tokens = per_rank * cfg.seq_len * n
+        ep = min(n, cfg.num_experts)
         print(f"loss {float(loss):.4f}  {dt * 1e3:.1f} ms/step  "
               f"{tokens / dt / 1e6:.2f} Mtok/s  "
               f"({cfg.num_experts} experts, top-2, "
-              f"{'EP' + str(n) if n > 1 else 'serial'})")
+              f"{f'dp{n} EP{ep}' if n > 1 else 'serial'})")
     aa.shutdown()
 
 
