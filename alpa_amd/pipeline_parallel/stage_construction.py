@@ -120,8 +120,10 @@ def _memory_factors(db, cluster_key: str,
         sc = getattr(r, "scalars", {}) or {}
         k = "gpt_act_bytes_per_token_hidden" + ("_remat" if remat else "")
         if k in sc:
-            # scalar is bytes/token/hidden; boundary is 2 bytes/token/hidden
-            act_f = sc[k] / 2.0
+            # scalar is bytes/token/hidden; boundary is 2 bytes/token/hidden.
+            # Under remat the measured coefficient excludes the held
+            # block INPUT (it predates the block) — floor at 1 boundary.
+            act_f = max(sc[k] / 2.0, 1.0)
         if "gpt_state_bytes_per_param" in sc:
             state_f = sc["gpt_state_bytes_per_param"] / 2.0
         if act_f != 3.0 or state_f != 6.0:
