@@ -17,6 +17,7 @@ from .mesh import (DeviceMesh, VirtualMesh, device, full_mesh, full_virtual_mesh
                    get_device_mesh, init_distributed, local_rank, rank,
                    world_size)
 from .optim import AdamW
+from .shard_parallel import auto_shard, capture_graph  # noqa: F401
 from .parallel_method import (AutoShardingOption, CreateStateParallel,
                               DataParallel, FollowParallel, ParallelMethod,
                               PipeshardParallel, ShardParallel,

@@ -92,6 +92,55 @@ class TrainState:
                                             weight_decay=weight_decay)
         return state
 
+    @classmethod
+    def create_auto(cls, builder: Callable[[], torch.nn.Module],
+                    example_inputs, method: Optional["ShardParallel"] = None,
+                    lr: float = 1e-4, betas=(0.9, 0.95),
+                    weight_decay: float = 0.0) -> "TrainState":
+        """Automatic parallelization of an ARBITRARY plain torch module —
+        no zoo membership, no model_hint (the reference's headline
+        capability: trace any program and parallelize it,
+        compile_shard_executable.py:54).
+
+        ``builder()`` returns the plain model (seed inside the builder so
+        every rank constructs identical weights); ``example_inputs`` is a
+        tensor or tuple of tensors for one forward.  The model is traced
+        (shard_parallel/capture.py), the ILP solves per-node strategies,
+        and the plan is EXECUTED by converting the matched modules to
+        parallel layers with automatic resharding on mismatched edges
+        (shard_parallel/plan_apply.py).  The usual step machinery
+        (microbatching, overlapped grad sync, fused AdamW) is installed
+        on top, so `aa.parallelize(step_fn)` works unchanged.
+        """
+        from .shard_parallel import auto_shard
+        method = method or ShardParallel()
+        model = builder()
+        dtype = getattr(torch, global_config.compute_dtype) \
+            if torch.cuda.is_available() else torch.float32
+        model = model.to(device=device(), dtype=dtype)
+
+        def to_dev(t):
+            if torch.is_tensor(t):
+                t = t.to(device())
+                return t.to(dtype) if t.is_floating_point() else t
+            return t
+        if torch.is_tensor(example_inputs):
+            example_inputs = (example_inputs,)
+        example_inputs = tuple(to_dev(t) for t in example_inputs)
+        o = method.auto_sharding_option
+        model, plan, mesh = auto_shard(
+            model, example_inputs,
+            memory_budget=o.memory_budget_per_device,
+            force_data_parallel=o.force_data_parallel,
+            mesh_shape=method.logical_mesh_shape)
+        method.logical_mesh_shape = plan.mesh_shape
+        state = cls(model, None, method, mesh)
+        state.plan = plan
+        state._install_grad_sync()
+        state.optimizer = AdamW(model.parameters(), lr=lr, betas=betas,
+                                weight_decay=weight_decay)
+        return state
+
     def _install_zero3(self, lr, betas, weight_decay):
         """ZeRO-3: block params sharded + JIT-gathered (parallel/zero3.py);
         leftover params (e.g. positional embeddings) use plain DP sync."""

@@ -295,3 +295,62 @@ def vocab_parallel_cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
         return ops.softmax_cross_entropy(logits, targets)
     return _VocabParallelCrossEntropy.apply(logits, targets, mesh, axis,
                                             vocab_start)
+
+
+class _GatherFromParallel(torch.autograd.Function):
+    """All-gather the feature (last) dim forward; slice backward (the
+    Megatron gather/split conjugate pair — downstream computation is
+    replicated SPMD, so each rank's grad of the full tensor is identical
+    and backward just takes this rank's slice)."""
+
+    @staticmethod
+    def forward(ctx, x, mesh: DeviceMesh, axis: int):
+        ctx.mesh, ctx.axis = mesh, axis
+        tp = mesh.axis_size(axis)
+        x = x.contiguous()
+        parts = [torch.empty_like(x) for _ in range(tp)]
+        dist.all_gather(parts, x, group=mesh.axis_group(axis))
+        return torch.cat(parts, dim=-1)
+
+    @staticmethod
+    def backward(ctx, g):
+        tp = ctx.mesh.axis_size(ctx.axis)
+        idx = ctx.mesh.axis_index(ctx.axis)
+        n = g.shape[-1] // tp
+        return g.narrow(-1, idx * n, n).contiguous(), None, None
+
+
+class _ScatterToParallel(torch.autograd.Function):
+    """Slice this rank's feature shard forward; all-gather backward
+    (Megatron scatter/gather conjugate: the producer is replicated, its
+    grad is the concatenation of the per-shard consumer grads)."""
+
+    @staticmethod
+    def forward(ctx, x, mesh: DeviceMesh, axis: int):
+        ctx.mesh, ctx.axis = mesh, axis
+        tp = mesh.axis_size(axis)
+        idx = mesh.axis_index(axis)
+        n = x.shape[-1] // tp
+        return x.narrow(-1, idx * n, n).contiguous()
+
+    @staticmethod
+    def backward(ctx, g):
+        tp = ctx.mesh.axis_size(ctx.axis)
+        g = g.contiguous()
+        parts = [torch.empty_like(g) for _ in range(tp)]
+        dist.all_gather(parts, g, group=ctx.mesh.axis_group(ctx.axis))
+        return torch.cat(parts, dim=-1), None, None
+
+
+def gather_from_tp(x, mesh, axis):
+    """feature-sharded -> replicated (resharding edge (b,w) -> (b,None))."""
+    if mesh is None or mesh.axis_size(axis) == 1:
+        return x
+    return _GatherFromParallel.apply(x, mesh, axis)
+
+
+def scatter_to_tp(x, mesh, axis):
+    """replicated -> feature-sharded (resharding edge (b,None) -> (b,w))."""
+    if mesh is None or mesh.axis_size(axis) == 1:
+        return x
+    return _ScatterToParallel.apply(x, mesh, axis)

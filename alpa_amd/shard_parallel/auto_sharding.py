@@ -192,3 +192,157 @@ def solve_gpt_sharding(num_devices: int, hidden: int, layers: int,
         for k, v in list(best.choices.items())[:12]:
             print(f"  {k}: {v}")
     return best
+
+
+# ----------------------------------------------------------------------
+# Captured graphs: arbitrary user programs
+# ----------------------------------------------------------------------
+
+
+def build_captured_graph(cap, mesh: MeshModel, dtype_bytes: int = 2,
+                         train: bool = True,
+                         fixed_b: Optional[int] = None) -> Graph:
+    """Lower a capture.CapturedGraph (torch.fx trace of the USER's
+    program) to the ILP op graph: per-node strategies enumerated for the
+    traced shapes — the analog of the reference's per-instruction
+    strategy construction over real HLO
+    (playground/auto_sharding_solver/hlo.py:664-830), replacing the
+    model_hint template path.
+
+    ``fixed_b`` restricts every node's batch split to that mesh axis
+    (None = no batch split): the executor feeds per-rank batch shards,
+    so the batch axis must be uniform across the program.  Per-node
+    WEIGHT axes stay free — mixed plans (shard the MLP, replicate the
+    attention) remain expressible.
+    """
+    import math as _math
+    from .strategies import (input_strategies,
+                             replicated_feature_strategies)
+    g = Graph()
+    for d in cap.ops:
+        nelem = float(_math.prod(d.out_shape)) if d.out_shape else 0.0
+        out_bytes = nelem * dtype_bytes
+        if d.kind == "input":
+            strats = input_strategies(mesh, out_bytes)
+            g.add(Node(d.name, "input", [], out_bytes, strategies=strats))
+        elif d.kind == "matmul" or d.kind == "conv":
+            tokens = int(nelem // max(d.extra["n"], 1))
+            strats = matmul_strategies(mesh, tokens, d.extra["k"],
+                                       d.extra["n"], dtype_bytes,
+                                       train=train)
+            # only divisible shards are executable by the parallel layers
+            def _divisible(st):
+                w_ax = st.out_spec[1]
+                if w_ax is not None and \
+                        d.extra["n"] % mesh.shape[w_ax] != 0:
+                    return False
+                in_ax = st.in_specs[0][1] if st.in_specs else None
+                if in_ax is not None and \
+                        d.extra["k"] % mesh.shape[in_ax] != 0:
+                    return False
+                return True
+            strats = [st for st in strats if _divisible(st)]
+            g.add(Node(d.name, "matmul", list(d.inputs), out_bytes,
+                       strategies=strats))
+        elif d.kind == "embedding":
+            tokens = int(nelem // max(d.extra["h"], 1))
+            strats = embedding_strategies(mesh, tokens, d.extra["vocab"],
+                                          d.extra["h"], dtype_bytes)
+            strats = [st for st in strats
+                      if "vocab" not in st.name or
+                      d.extra["vocab"] % mesh.shape[
+                          int(st.name.split("vocab")[1])] == 0]
+            g.add(Node(d.name, "embedding", [], out_bytes,
+                       strategies=strats))
+        elif d.kind == "elemwise" and d.inputs:
+            g.add(elemwise_follow_node(d.name, d.inputs[0], out_bytes))
+        elif d.kind in ("norm", "opaque", "output") or not d.inputs:
+            strats = replicated_feature_strategies(
+                mesh, out_bytes, n_inputs=max(len(d.inputs), 1),
+                name=d.kind)
+            g.add(Node(d.name, d.kind, list(d.inputs), out_bytes,
+                       strategies=strats))
+        else:
+            g.add(elemwise_follow_node(d.name, d.inputs[0], out_bytes))
+    # uniform batch axis across the program (the executor feeds fixed
+    # per-rank batch shards); weight axes stay per-node free
+    for node in g.nodes:
+        if node.follow is not None or not node.strategies:
+            continue
+        kept = [st for st in node.strategies
+                if st.out_spec[0] == fixed_b]
+        if kept:
+            node.strategies = kept
+    return g
+
+
+@dataclass
+class CapturedPlan:
+    """Solved sharding for a captured program: logical mesh + one
+    strategy per op (follow nodes resolved), ready for
+    plan_apply.apply_captured_plan."""
+    mesh_shape: Tuple[int, int]            # (dp, tp): axis 0 batch, 1 tp
+    objective: float
+    #: op index -> strategy name (owners only)
+    choices: Dict[int, str]
+    #: op index -> (in_specs, out_spec) of the chosen strategy
+    specs: Dict[int, Tuple[tuple, tuple]]
+
+
+def solve_captured(cap, num_devices: int,
+                   memory_budget: Optional[float] = None,
+                   force_data_parallel: bool = False,
+                   time_limit: Optional[float] = None,
+                   dtype_bytes: int = 2,
+                   train: bool = True,
+                   mesh_shape: Optional[Tuple[int, int]] = None
+                   ) -> CapturedPlan:
+    """Outer loop over (dp, tp) factorizations x inner ILP over the
+    captured graph (reference compile_shard_executable logical-shape
+    loop + _call_solver_serialized_args)."""
+    from .mesh_search import factorizations
+    if time_limit is None:
+        time_limit = min(30.0, global_config.solver_timeout / 8)
+    best = None
+    for (dp, tp) in factorizations(num_devices):
+        if force_data_parallel and tp != 1:
+            continue
+        if mesh_shape is not None and (dp, tp) != tuple(mesh_shape):
+            continue
+        mesh = MeshModel((dp, tp), alpha=global_config.mesh_alpha,
+                         beta=global_config.mesh_beta)
+        fixed_b = 0 if dp > 1 else None
+        g = build_captured_graph(cap, mesh, dtype_bytes, train, fixed_b)
+        res = solve(g, mesh, memory_budget=memory_budget,
+                    time_limit=time_limit,
+                    edge_cost_factor=2.0 if train else 1.0)
+        if not res.feasible:
+            continue
+        choices, specs = {}, {}
+        for i, r in res.choices.items():
+            node = g.nodes[i]
+            if node.follow is not None:
+                # resolve to owner's strategy for spec bookkeeping
+                owner = i
+                seen = set()
+                while g.nodes[owner].follow is not None and \
+                        owner not in seen:
+                    seen.add(owner)
+                    owner = g.nodes[owner].follow
+                st = g.nodes[owner].strategies[res.choices[owner]]
+                specs[i] = ([st.out_spec] * max(len(node.inputs), 1),
+                            st.out_spec)
+                continue
+            st = node.strategies[r]
+            choices[i] = st.name
+            specs[i] = (list(st.in_specs), st.out_spec)
+        plan = CapturedPlan((dp, tp), res.objective, choices, specs)
+        if best is None or plan.objective < best.objective:
+            best = plan
+    assert best is not None, "no feasible sharding found for captured graph"
+    if global_config.print_strategy:
+        print(f"[auto_sharding/captured] mesh {best.mesh_shape} "
+              f"objective {best.objective:.4f}")
+        for i, name in list(best.choices.items())[:16]:
+            print(f"  {cap.ops[i].name}: {name}")
+    return best

@@ -182,3 +182,35 @@ def loss_strategies(mesh: MeshModel, tokens: int, vocab: int,
                                 out_spec=(b_ax, None), compute_cost=0.0,
                                 comm_cost=small, memory=0.0))
     return out
+
+
+def input_strategies(mesh: MeshModel, out_bytes: float) -> List[Strategy]:
+    """Graph inputs (placeholders): arrive pre-sharded on the batch dim
+    for free (the SPMD data loader feeds per-rank shards)."""
+    out = [Strategy(name="repl", in_specs=[], out_spec=REPLICATED,
+                    compute_cost=0.0, comm_cost=0.0, memory=out_bytes)]
+    for b_ax in (0, 1):
+        if mesh.axis_size(b_ax) > 1:
+            out.append(Strategy(
+                name=f"b{b_ax}", in_specs=[], out_spec=(b_ax, None),
+                compute_cost=0.0, comm_cost=0.0,
+                memory=out_bytes / mesh.axis_size(b_ax)))
+    return out
+
+
+def replicated_feature_strategies(mesh: MeshModel, out_bytes: float,
+                                  n_inputs: int = 1,
+                                  name: str = "norm") -> List[Strategy]:
+    """Ops that need the FULL feature dim locally (LayerNorm, softmax
+    over features, raw bmm/reshape mixing features, unknown modules):
+    batch split passes through, feature split must be gathered first —
+    the resharding cost lands on the incoming edge."""
+    out = []
+    axes = [None] + [a for a in (0, 1) if mesh.shape[a] > 1]
+    for b_ax in axes:
+        d = mesh.axis_size(b_ax)
+        out.append(Strategy(
+            name=f"b{b_ax}", in_specs=[(b_ax, None)] * n_inputs,
+            out_spec=(b_ax, None), compute_cost=0.0, comm_cost=0.0,
+            memory=out_bytes / d))
+    return out
