@@ -43,6 +43,11 @@ hipError_t launch_bias_gelu_bwd(const void*, const void*, const void*, void*,
                                 hipStream_t);
 hipError_t launch_adamw(const void*, const void*, int, float, float, float,
                         float, float, float, int, hipStream_t);
+hipError_t launch_fp8_quantize(const void*, void*, const float*, float*,
+                               int64_t, int, hipStream_t);
+hipError_t launch_fp8_quantize_dual(const void*, void*, void*, const float*,
+                                    float*, int64_t, int64_t, int,
+                                    hipStream_t);
 hipError_t launch_ce_fwd(const void*, const int64_t*, float*, float*, int64_t,
                          int64_t, hipStream_t);
 hipError_t launch_ce_bwd(const float*, const void*, const int64_t*,
@@ -586,6 +591,39 @@ at::Tensor mfma_probe32(const at::Tensor& a, const at::Tensor& b) {
 
 }  // namespace
 
+// fp8 quantize: one-pass amax + delayed-scale cast (optional transposed
+// twin for the dW GEMM).  x: bf16 contiguous; scale: fp32 scalar tensor
+// on device (previous-step scale); returns (q, qt | empty, amax).
+std::vector<at::Tensor> fp8_quantize(const at::Tensor& x,
+                                     const at::Tensor& scale, bool dual,
+                                     bool e5m2) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == at::kBFloat16, "x must be bf16 contiguous");
+  TORCH_CHECK(scale.is_cuda() && scale.scalar_type() == at::kFloat &&
+              scale.numel() == 1, "scale must be a device fp32 scalar");
+  auto qtype = e5m2 ? at::kFloat8_e5m2 : at::kFloat8_e4m3fn;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+  auto amax = at::zeros({1}, x.options().dtype(at::kFloat));
+  auto q = at::empty(x.sizes(), x.options().dtype(qtype));
+  if (!dual) {
+    TORCH_CHECK(x.numel() % 8 == 0, "numel must be divisible by 8");
+    HIP_OK(launch_fp8_quantize(x.const_data_ptr(), q.data_ptr(),
+                               scale.const_data_ptr<float>(),
+                               amax.data_ptr<float>(), x.numel(),
+                               e5m2 ? 1 : 0, stream));
+    return {q, at::Tensor(), amax};
+  }
+  TORCH_CHECK(x.dim() == 2, "dual quantize needs a 2-D tensor");
+  int64_t M = x.size(0), N = x.size(1);
+  auto qt = at::empty({N, M}, x.options().dtype(qtype));
+  HIP_OK(launch_fp8_quantize_dual(x.const_data_ptr(), q.data_ptr(),
+                                  qt.data_ptr(),
+                                  scale.const_data_ptr<float>(),
+                                  amax.data_ptr<float>(), M, N,
+                                  e5m2 ? 1 : 0, stream));
+  return {q, qt, amax};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // Bump when the Python<->extension call surface changes; checked by
   // alpa_amd/version.py (reference check_alpa_jaxlib_version, version.py:10)
@@ -599,6 +637,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "bias+GeLU forward (gfx950)");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "bias+GeLU backward (gfx950)");
   m.def("adamw_step_raw", &adamw_step_raw, "multi-tensor AdamW (gfx950)");
+  m.def("fp8_quantize", &fp8_quantize,
+        "one-pass fp8 amax+cast, optional transposed twin (gfx950)");
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused CE fwd (gfx950)");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused CE bwd (gfx950)");
   m.def("attn_fwd", &attn_fwd, "flash attention fwd (gfx950 MFMA)",

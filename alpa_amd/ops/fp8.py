@@ -1,17 +1,29 @@
-"""fp8 (e4m3) forward GEMMs — EXPERIMENTAL, off by default.
+"""fp8 (OCP e4m3) GEMM recipe for gfx950 — forward AND backward.
 
-MI355X's matrix cores run fp8 at 2x the bf16 rate (≈5 PF dense); the
-GEMM-bound ~55% of the training step is the target.  This module
-implements the standard "fp8 forward, bf16 backward" recipe: the forward
-GEMM quantizes X and W to float8_e4m3fn with per-tensor amax scaling and
-runs `torch._scaled_mm` (hipBLASLt fp8 under ROCm — probed working on
-this stack, ~3.6% rel err); the backward keeps exact bf16 GEMMs, so
-gradients match the bf16 path up to the forward quantization error.
+MI355X matrix cores run fp8 at ~2x the bf16 rate; measured at the
+GPT-2.6B step shapes (profiles/fp8_bench): 2.6-3.1 PF vs 1.35-1.55 PF
+bf16.  Round 1 lost the win to cast overhead (torch-op amax+cast up to
+7.5 ms/tensor); round 2 fixes that with one-pass HIP quantize kernels
+(ops/csrc/fp8_quant.hip) and quantizes all three GEMMs:
 
-NOT used by the flagship benchmark: BASELINE comparisons require bf16
-compute.  Enable per-layer via global_config.fp8_gemm
-(`ALPA_AMD_FP8=1`) for experiments; a delayed-scaling recipe and fp8
-backward are round-2 work.
+  fwd : y  = Xq  @ Wq^T         (X e4m3 via fused one-pass quantize)
+  bwd : dX = dYq @ Wq           (dY e4m3, W from the per-step cache)
+        dW = dYq^T @ Xq         (transposed twins from DUAL quantize —
+                                 no extra HBM pass for either operand)
+
+Scaling is DELAYED (Transformer-Engine style): each tensor role keeps a
+running amax on device (`r = max(amax_now, 0.999 r)`) and the NEXT
+quantize uses scale = r / 448.  Everything is device-side in-place, so
+the whole recipe is hipGraph-capturable: on replay the captured
+quantize kernels re-read the updated weights/running-amax in place.
+
+Numerics: e4m3 everywhere with per-tensor delayed scaling (the
+DeepSeek-V3 style choice); the loss-curve check against bf16 lives in
+tools/fp8_losscheck.py with results in docs/BENCHMARK.md.
+
+Enable via global_config.fp8_gemm (ALPA_AMD_FP8=1).  The flagship
+BASELINE bench stays bf16 by default; `bench.py --fp8` reports the fp8
+number separately.
 """
 from __future__ import annotations
 
@@ -19,12 +31,16 @@ from typing import Optional
 
 import torch
 
-E4M3_MAX = 448.0
+from ._backend import hip_ops
 
-# epoch counter: optimizers bump it after each parameter update so the
-# per-module quantized-weight caches invalidate (the fused AdamW kernel
-# writes through raw pointers and does not touch torch's version
-# counters)
+E4M3_MAX = 448.0
+AMAX_DECAY = 0.999
+
+# epoch counter: optimizers bump it after each parameter update so
+# EAGER-mode per-module quantized-weight caches invalidate.  Under
+# hipGraph capture the quantize kernels are captured in-step (the cache
+# misses during capture) and re-run on every replay, reading the updated
+# weights in place — no epoch check executes at replay time.
 _EPOCH = 0
 
 
@@ -33,68 +49,119 @@ def bump_epoch() -> None:
     _EPOCH += 1
 
 
-def quantize_weight_cached(module, weight: torch.Tensor):
-    """(wq, ws) for `module.weight`, re-quantized once per optimizer
-    epoch instead of per forward call — the weight-side half of the
-    naive recipe's per-GEMM amax+cast overhead disappears."""
-    cached = getattr(module, "_fp8_cache", None)
-    if cached is not None and cached[0] == _EPOCH:
-        return cached[1], cached[2]
-    wq, ws = _quantize(weight)
-    module._fp8_cache = (_EPOCH, wq, ws)
-    return wq, ws
-
-
-def _quantize(t: torch.Tensor):
-    """Per-tensor symmetric scaling into float8_e4m3fn; returns
-    (fp8 tensor, fp32 scale such that t ≈ t_fp8 * scale)."""
-    amax = t.abs().amax().clamp_min(1e-12).float()
-    scale = amax / E4M3_MAX
-    q = (t.float() / scale).clamp(-E4M3_MAX, E4M3_MAX).to(
-        torch.float8_e4m3fn)
-    return q, scale
-
-
 def fp8_available(x: torch.Tensor) -> bool:
     return (x.is_cuda and hasattr(torch, "float8_e4m3fn")
             and hasattr(torch, "_scaled_mm"))
 
 
+class _RoleState:
+    """Delayed-scaling state for one tensor role (module activation /
+    grad): persistent device tensors, updated in place (graph-safe)."""
+
+    __slots__ = ("running", "scale")
+
+    def __init__(self):
+        self.running = None
+        self.scale = None
+
+    def ensure(self, ref: torch.Tensor):
+        if self.scale is None:
+            with torch.no_grad():
+                amax = ref.abs().amax().float().clamp_min(1e-12).reshape(1)
+                self.running = amax.clone()
+                self.scale = (amax / E4M3_MAX).contiguous()
+
+    @torch.no_grad()
+    def update(self, amax: torch.Tensor):
+        # r <- max(amax, decay * r); scale <- r / 448, all in place
+        torch.maximum(amax, self.running * AMAX_DECAY, out=self.running)
+        self.scale.copy_(self.running.clamp_min(1e-12) / E4M3_MAX)
+
+
+def _role(module, name: str) -> _RoleState:
+    st = getattr(module, f"_fp8_{name}", None)
+    if st is None:
+        st = _RoleState()
+        setattr(module, f"_fp8_{name}", st)
+    return st
+
+
+def quantize(x2: torch.Tensor, state: _RoleState, dual: bool):
+    """One-pass delayed-scale quantize (+ transposed twin when dual).
+    Returns (q, qt|None, used_scale); updates the role's running amax.
+    `used_scale` is a snapshot — state.scale mutates in place right
+    after, and the dequant scale must match the cast actually done."""
+    state.ensure(x2)
+    used = state.scale.clone()
+    q, qt, amax = hip_ops().fp8_quantize(x2, used, dual, False)
+    state.update(amax)
+    return q, qt, used
+
+
+def quantize_weight_cached(module, weight: torch.Tensor):
+    """(wq [n,k], wqt [k,n], ws) re-quantized once per optimizer epoch in
+    eager mode; captured in-step under hipGraphs (see module doc)."""
+    cached = getattr(module, "_fp8_cache", None)
+    graphing = torch.cuda.is_current_stream_capturing() \
+        if weight.is_cuda else False
+    if cached is not None and cached[0] == _EPOCH and not graphing:
+        return cached[1], cached[2], cached[3]
+    st = _role(module, "w")
+    with torch.no_grad():
+        # exact per-epoch weight scale (weights are fully known)
+        st.ensure(weight)
+        amax = weight.abs().amax().float().clamp_min(1e-12).reshape(1)
+        st.running.copy_(amax)
+        st.scale.copy_(amax / E4M3_MAX)
+        wq, wqt, _ = hip_ops().fp8_quantize(weight.contiguous(),
+                                             st.scale, True, False)
+    module._fp8_cache = (_EPOCH, wq, wqt, st.scale)
+    return wq, wqt, st.scale
+
+
 class _Fp8Linear(torch.autograd.Function):
-    """y = x @ w.T (+ bias): fp8 forward GEMM, exact bf16 backward."""
+    """y = x @ w.T (+ bias) with fp8 fwd AND bwd GEMMs (see module doc).
+    Saves the fp8 transposed activation instead of the bf16 input —
+    halves the saved-activation bytes as a side effect."""
 
     @staticmethod
-    def forward(ctx, x, w, bias, wq=None, ws=None):
+    def forward(ctx, x, w, bias, module):
         shape = x.shape
-        x2 = x.reshape(-1, shape[-1])
-        xq, xs = _quantize(x2)
-        if wq is None:
-            wq, ws = _quantize(w)
-        # _scaled_mm wants B column-major: w.T with w row-major qualifies
-        y = torch._scaled_mm(xq, wq.t(), scale_a=xs, scale_b=ws,
+        x2 = x.reshape(-1, shape[-1]).contiguous()
+        sx = _role(module, "x")
+        xq, xqt, x_scale = quantize(x2, sx, dual=True)
+        wq, wqt, ws = quantize_weight_cached(module, w)
+        y = torch._scaled_mm(xq, wq.t(), scale_a=x_scale, scale_b=ws,
                              bias=bias, out_dtype=x.dtype)
-        ctx.save_for_backward(x2, w)
+        ctx.module = module
+        ctx.save_for_backward(xqt, wqt, x_scale, ws)
         ctx.has_bias = bias is not None
         return y.reshape(*shape[:-1], w.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        x2, w = ctx.saved_tensors
-        dy2 = dy.reshape(-1, dy.shape[-1])
-        dx = (dy2 @ w).reshape(*dy.shape[:-1], w.shape[1])
-        dw = dy2.t() @ x2
+        xqt, wqt, sx, ws = ctx.saved_tensors
+        module = ctx.module
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        sg = _role(module, "g")
+        dyq, dyqt, g_scale = quantize(dy2, sg, dual=True)
+        # dX [M,k] = dY [M,n] @ W [n,k]; B col-major = wqt.t()
+        dx = torch._scaled_mm(dyq, wqt.t(), scale_a=g_scale, scale_b=ws,
+                              out_dtype=dy.dtype)
+        # dW [n,k] = dY^T [n,M] @ X [M,k]; A row-major = dyqt,
+        # B col-major = xqt.t()
+        dw = torch._scaled_mm(dyqt, xqt.t(), scale_a=g_scale, scale_b=sx,
+                              out_dtype=dy.dtype)
         db = dy2.sum(0) if ctx.has_bias else None
-        return dx, dw, db, None, None
+        return (dx.reshape(*dy.shape[:-1], wqt.shape[0]), dw, db, None)
 
 
 def fp8_linear(x: torch.Tensor, weight: torch.Tensor,
                bias: Optional[torch.Tensor] = None,
                module=None) -> torch.Tensor:
-    """Drop-in for F.linear with an fp8 forward GEMM.  Requires CUDA and
-    dims divisible by 16 (hipBLASLt fp8 tile constraint); callers gate on
-    `fp8_available` and fall back to bf16 matmul otherwise.  Passing the
-    owning `module` enables the per-epoch quantized-weight cache."""
-    if module is not None:
-        wq, ws = quantize_weight_cached(module, weight)
-        return _Fp8Linear.apply(x, weight, bias, wq, ws)
-    return _Fp8Linear.apply(x, weight, bias)
+    """Drop-in for F.linear with fp8 fwd+bwd GEMMs.  Requires CUDA and
+    dims divisible by 16 (hipBLASLt fp8 tiles); callers gate on
+    `fp8_available`.  `module` anchors the per-role delayed-scaling
+    state and the per-epoch weight cache."""
+    assert module is not None, "fp8_linear needs the owning module"
+    return _Fp8Linear.apply(x, weight, bias, module)

@@ -61,6 +61,11 @@ def parse_args():
                         "accounting then uses the reference's factor 96)")
     p.add_argument("--no-hipgraph", action="store_true",
                    help="disable hipGraph step capture (single-GPU only)")
+    p.add_argument("--fp8", action="store_true",
+                   help="fp8 (e4m3) GEMMs with delayed scaling for all "
+                        "linear fwd+bwd (ops/fp8.py).  NOT the headline "
+                        "metric: BASELINE requires bf16 compute; this "
+                        "reports dtype=bf16+fp8gemm for comparison")
     args = p.parse_args()
     if args.nmb == 0:
         args.nmb = 1
@@ -82,6 +87,9 @@ def main():
     args = parse_args()
     if torch.cuda.is_available():
         _enable_tunableop()
+    if args.fp8:
+        from alpa_amd.global_env import global_config as _gc0
+        _gc0.fp8_gemm = True
     aa.init()
     on_gpu = torch.cuda.is_available()
     n = aa.world_size()
@@ -217,7 +225,8 @@ def main():
             "scaling": "weak",
             "vs_baseline": round(value / baseline_aggregate_tflops, 3)
                            if on_gpu else None,
-            "dtype": "bf16" if on_gpu else "fp32",
+            "dtype": ("bf16+fp8gemm" if args.fp8 else "bf16")
+                     if on_gpu else "fp32",
             "data": "synthetic",
             "config": {
                 "model": f"GPT-{model_name}" if on_gpu else "GPT-tiny-cpu",
