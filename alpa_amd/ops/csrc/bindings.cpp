@@ -608,8 +608,8 @@ std::vector<at::Tensor> fp8_quantize(const at::Tensor& x,
   if (!dual) {
     TORCH_CHECK(x.numel() % 8 == 0, "numel must be divisible by 8");
     HIP_OK(launch_fp8_quantize(x.const_data_ptr(), q.data_ptr(),
-                               scale.const_data_ptr<float>(),
-                               amax.data_ptr<float>(), x.numel(),
+                               (const float*)scale.const_data_ptr(),
+                               (float*)amax.data_ptr(), x.numel(),
                                e5m2 ? 1 : 0, stream));
     return {q, at::Tensor(), amax};
   }
@@ -618,8 +618,8 @@ std::vector<at::Tensor> fp8_quantize(const at::Tensor& x,
   auto qt = at::empty({N, M}, x.options().dtype(qtype));
   HIP_OK(launch_fp8_quantize_dual(x.const_data_ptr(), q.data_ptr(),
                                   qt.data_ptr(),
-                                  scale.const_data_ptr<float>(),
-                                  amax.data_ptr<float>(), M, N,
+                                  (const float*)scale.const_data_ptr(),
+                                  (float*)amax.data_ptr(), M, N,
                                   e5m2 ? 1 : 0, stream));
   return {q, qt, amax};
 }
