@@ -233,6 +233,7 @@ class GPTModel(nn.Module):
                                             dtype=dtype, device=device,
                                             init_seed=init_seed,
                                             init_tag="lm_head")
+        self.lm_head._fp8_exclude = True  # logits GEMM stays bf16
         if cfg.tie_embeddings:
             self.lm_head.weight = self.wte.weight
 
@@ -309,6 +310,7 @@ class GPTStage(nn.Module):
                 dtype=dtype, device=device, init_seed=init_seed,
                 init_tag="wte" if cfg.tie_embeddings else "lm_head",
                 init_std=0.02 if cfg.tie_embeddings else None)
+            self.lm_head._fp8_exclude = True  # logits GEMM stays bf16
 
     def forward(self, x, microbatch):
         ids, labels = microbatch["ids"], microbatch.get("labels")

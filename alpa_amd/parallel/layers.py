@@ -102,11 +102,16 @@ def _sharded_normal_(w: torch.Tensor, full_shape, shard_dim: int,
     del full
 
 
-def _fp8_ok(x, weight) -> bool:
-    """Gate for the EXPERIMENTAL fp8-forward GEMM path (ops/fp8.py):
-    opt-in via global_config.fp8_gemm, CUDA only, dims 16-aligned."""
+def _fp8_ok(x, weight, module=None) -> bool:
+    """Gate for the fp8 GEMM path (ops/fp8.py): opt-in via
+    global_config.fp8_gemm, CUDA only, dims 16-aligned.  Modules marked
+    ``_fp8_exclude`` (the LM head — standard mixed-fp8 practice keeps
+    first/last layers in high precision; measured to close most of the
+    fp8 loss-curve gap) stay bf16."""
     from ..global_env import global_config
     if not global_config.fp8_gemm or not x.is_cuda:
+        return False
+    if module is not None and getattr(module, "_fp8_exclude", False):
         return False
     from ..ops.fp8 import fp8_available
     return (fp8_available(x) and x.shape[-1] % 16 == 0 and
@@ -165,7 +170,7 @@ class ColumnParallelLinear(nn.Module):
         return y
 
     def _use_fp8(self, x):
-        return _fp8_ok(x, self.weight)
+        return _fp8_ok(x, self.weight, self)
 
 
 class RowParallelLinear(nn.Module):
@@ -197,7 +202,7 @@ class RowParallelLinear(nn.Module):
             self.register_parameter("bias", None)
 
     def forward(self, x):
-        if _fp8_ok(x, self.weight):
+        if _fp8_ok(x, self.weight, self):
             from ..ops.fp8 import fp8_linear
             y = fp8_linear(x, self.weight, module=self)
         else:
