@@ -29,11 +29,16 @@ class GlobalConfig:
     # ---------- compute ----------
     #: default compute dtype for models/benchmarks
     compute_dtype: str = "bfloat16"
-    #: EXPERIMENTAL: fp8(e4m3) forward GEMMs in the parallel linear
-    #: layers (bf16 backward; ops/fp8.py).  NOT valid for BASELINE
-    #: comparisons — env ALPA_AMD_FP8=1
+    #: fp8(e4m3) GEMMs in the parallel linear layers with delayed
+    #: scaling (fwd + backward; ops/fp8.py).  NOT valid for BASELINE
+    #: comparisons (bf16 contract) — env ALPA_AMD_FP8=1
     fp8_gemm: bool = field(
         default_factory=lambda: os.environ.get("ALPA_AMD_FP8", "0") == "1")
+    #: compute dW in fp8 too (default); ALPA_AMD_FP8_WGRAD=0 keeps the
+    #: weight-grad GEMM in bf16 (tighter numerics, ~1/3 less fp8 win)
+    fp8_wgrad: bool = field(
+        default_factory=lambda: os.environ.get("ALPA_AMD_FP8_WGRAD",
+                                               "1") == "1")
     #: use hand-written HIP kernels when their extension is available
     use_hip_kernels: bool = True
     #: fail loudly if running on GPU without the HIP extension (anti-silent-fallback)

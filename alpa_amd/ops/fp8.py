@@ -126,32 +126,41 @@ class _Fp8Linear(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, w, bias, module):
+        from ..global_env import global_config
         shape = x.shape
         x2 = x.reshape(-1, shape[-1]).contiguous()
         sx = _role(module, "x")
-        xq, xqt, x_scale = quantize(x2, sx, dual=True)
+        wgrad_fp8 = global_config.fp8_wgrad
+        xq, xqt, x_scale = quantize(x2, sx, dual=wgrad_fp8)
         wq, wqt, ws = quantize_weight_cached(module, w)
         y = torch._scaled_mm(xq, wq.t(), scale_a=x_scale, scale_b=ws,
                              bias=bias, out_dtype=x.dtype)
         ctx.module = module
-        ctx.save_for_backward(xqt, wqt, x_scale, ws)
+        ctx.wgrad_fp8 = wgrad_fp8
+        if wgrad_fp8:
+            ctx.save_for_backward(xqt, wqt, x_scale, ws)
+        else:
+            ctx.save_for_backward(x2, wqt, x_scale, ws)
         ctx.has_bias = bias is not None
         return y.reshape(*shape[:-1], w.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        xqt, wqt, sx, ws = ctx.saved_tensors
+        xsaved, wqt, sx, ws = ctx.saved_tensors
         module = ctx.module
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         sg = _role(module, "g")
-        dyq, dyqt, g_scale = quantize(dy2, sg, dual=True)
+        dyq, dyqt, g_scale = quantize(dy2, sg, dual=ctx.wgrad_fp8)
         # dX [M,k] = dY [M,n] @ W [n,k]; B col-major = wqt.t()
         dx = torch._scaled_mm(dyq, wqt.t(), scale_a=g_scale, scale_b=ws,
                               out_dtype=dy.dtype)
-        # dW [n,k] = dY^T [n,M] @ X [M,k]; A row-major = dyqt,
-        # B col-major = xqt.t()
-        dw = torch._scaled_mm(dyqt, xqt.t(), scale_a=g_scale, scale_b=sx,
-                              out_dtype=dy.dtype)
+        if ctx.wgrad_fp8:
+            # dW [n,k] = dY^T [n,M] @ X [M,k]; A row-major = dyqt,
+            # B col-major = xqt.t()
+            dw = torch._scaled_mm(dyqt, xsaved.t(), scale_a=g_scale,
+                                  scale_b=sx, out_dtype=dy.dtype)
+        else:
+            dw = dy2.t() @ xsaved     # exact bf16 weight grad
         db = dy2.sum(0) if ctx.has_bias else None
         return (dx.reshape(*dy.shape[:-1], wqt.shape[0]), dw, db, None)
 
