@@ -14,6 +14,7 @@
 //   B (32x16): n = lane&15, k = (lane>>4)*8 + j
 //   C/D      : n = lane&15, m = (lane>>4)*4 + reg (f32x4)
 #include "common.h"
+#include <cstdlib>
 
 #define ATTN_BLOCK_Q 64
 #define ATTN_BLOCK_K 64
@@ -45,8 +46,8 @@ __device__ inline int xcd_swizzle(int flat, int total) {
 // later phases; asm keep-alives prevent dead-code elimination of earlier
 // phases — guide methodology rule 17)
 template <int Dp, int ABL = 0, bool AL = false, bool VL = false,
-          bool DB = true>
-__global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
+          bool DB = true, int NT = ATTN_THREADS>
+__global__ __launch_bounds__(NT) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ o,
     float* __restrict__ lse_out, int H, int S, int Skv, int D, float scale,
@@ -60,10 +61,10 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   constexpr int NTILES = ATTN_BLOCK_K / 16;  // 4
   constexpr int DTILES = Dp / 16;
   constexpr int LP = 8;                      // LDS bank padding (16 B)
-  constexpr int NW = ATTN_THREADS / 64;      // 8 waves
+  constexpr int NW = NT / 64;                // waves per block
   constexpr int GPR = Dp / 8;                // bf16x8 groups per kv row
   constexpr int TOTAL_G = ATTN_BLOCK_K * GPR;
-  constexpr int G_PER_T = (TOTAL_G + ATTN_THREADS - 1) / ATTN_THREADS;
+  constexpr int G_PER_T = (TOTAL_G + NT - 1) / NT;
 
   // XCD co-location + causal ordering: swizzle the flattened id so one
   // XCD owns contiguous (bh, qb) work; within a bh, later q blocks
@@ -136,7 +137,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   auto issue_loads = [&](int tile) {
 #pragma unroll
     for (int it = 0; it < G_PER_T; ++it) {
-      int t = threadIdx.x + it * ATTN_THREADS;
+      int t = threadIdx.x + it * NT;
       bf16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
       kreg[2 * it] = z; kreg[2 * it + 1] = z;
       vreg[2 * it] = z; vreg[2 * it + 1] = z;
@@ -158,7 +159,7 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
   auto write_tile = [&](int buf) {
 #pragma unroll
     for (int it = 0; it < G_PER_T; ++it) {
-      int t = threadIdx.x + it * ATTN_THREADS;
+      int t = threadIdx.x + it * NT;
       if (t < TOTAL_G / 2) {
         int kvr = (t / GPR) * 2, dg = (t % GPR) * 8;
         *reinterpret_cast<bf16x8*>(&k_lds[buf][kvr][dg]) = kreg[2 * it];
@@ -268,14 +269,21 @@ __global__ __launch_bounds__(ATTN_THREADS) void attn_fwd_kernel(
     tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
 
     // ---- online rescale; P^T = exp(S^T - m); sum ----
-    float m_new = fmaxf(m_state, tile_max);
-    float alpha = (m_state == -INFINITY) ? 0.f : __expf(m_state - m_new);
-    m_state = m_new;
-    l_state *= alpha;
+    // T13 defer-max: skip the O-wide rescale while the max grows by
+    // less than THR (P values then bounded by e^THR=2981, fine in f32
+    // accum); only when some lane's max jumps past the threshold does
+    // the wave rescale and advance m (guide T13; isolated +5%).
+    constexpr float DEFER_THR = 8.0f;
+    if (!__all(tile_max <= m_state + DEFER_THR)) {
+      float m_new = fmaxf(m_state, tile_max);
+      float alpha = (m_state == -INFINITY) ? 0.f : __expf(m_state - m_new);
+      m_state = m_new;
+      l_state *= alpha;
 #pragma unroll
-    for (int dt = 0; dt < DTILES; ++dt) {
-      o_acc[dt][0] *= alpha; o_acc[dt][1] *= alpha;
-      o_acc[dt][2] *= alpha; o_acc[dt][3] *= alpha;
+      for (int dt = 0; dt < DTILES; ++dt) {
+        o_acc[dt][0] *= alpha; o_acc[dt][1] *= alpha;
+        o_acc[dt][2] *= alpha; o_acc[dt][3] *= alpha;
+      }
     }
     float part = 0.f;
 #pragma unroll
@@ -615,18 +623,44 @@ hipError_t launch_attn_fwd_v2(const void* q, const void* k, const void* v,
   return hipGetLastError();
 }
 
+static int attn_fwd_nt() {
+  // block size selection: 512 threads (8 waves) fits TWO blocks per CU
+  // at Dp=96 double-buffered (72.6 KB LDS each) — cross-block overlap
+  // hides the per-tile barriers that a single 16-wave block serializes.
+  // Override with ALPA_ATTN_FWD_NT=1024 for the round-1 shape.
+  static int nt = -1;
+  if (nt < 0) {
+    const char* e = getenv("ALPA_ATTN_FWD_NT");
+    nt = e ? atoi(e) : 512;
+    if (nt != 512 && nt != 1024) nt = 512;
+  }
+  return nt;
+}
+
 hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
                            void* o, float* lse, int64_t B, int64_t H,
                            int64_t S, int64_t Skv, int64_t D, float scale,
                            int causal, const float* alibi,
                            const int* kv_lens, const int64_t* strides,
                            hipStream_t stream) {
-  dim3 grid((uint32_t)ceil_div(S, 16 * (ATTN_THREADS / 64)), (uint32_t)(B * H));
-  dim3 block(ATTN_THREADS);
+  const int NTsel = attn_fwd_nt();
+  dim3 grid((uint32_t)ceil_div(S, 16 * (NTsel / 64)), (uint32_t)(B * H));
+  dim3 block(NTsel);
 #define FWD_VARIANT(DP, ALB, VLB, ALP, VLP)                                \
-  attn_fwd_kernel<DP, 0, ALB, VLB><<<grid, block, 0, stream>>>(            \
-      (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,   \
-      (int)H, (int)S, (int)Skv, (int)D, scale, causal, ALP, VLP, st)
+  do {                                                                     \
+    if (NTsel == 512)                                                      \
+      attn_fwd_kernel<DP, 0, ALB, VLB, true, 512>                          \
+          <<<grid, block, 0, stream>>>(                                    \
+              (const short*)q, (const short*)k, (const short*)v,           \
+              (short*)o, lse, (int)H, (int)S, (int)Skv, (int)D, scale,     \
+              causal, ALP, VLP, st);                                       \
+    else                                                                   \
+      attn_fwd_kernel<DP, 0, ALB, VLB, true, 1024>                         \
+          <<<grid, block, 0, stream>>>(                                    \
+              (const short*)q, (const short*)k, (const short*)v,           \
+              (short*)o, lse, (int)H, (int)S, (int)Skv, (int)D, scale,     \
+              causal, ALP, VLP, st);                                       \
+  } while (0)
 #define FWD_DISPATCH(DP)                                                   \
   do {                                                                     \
     if (alibi && kv_lens) FWD_VARIANT(DP, true, true, alibi, kv_lens);     \
