@@ -213,6 +213,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       continue;
     }
 
+__builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
     // ---- S^T = K Q^T (A = K tile from LDS, B = Q fragments) ----
     f32x4 s_acc[NTILES];
 #pragma unroll
@@ -227,6 +228,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       }
       s_acc[nt] = acc;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     if (ABL >= 3) {  // QK^T only
 #pragma unroll
@@ -309,6 +311,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
       continue;
     }
 
+__builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
     // ---- O^T += V^T P^T  (A = V^T from vt_lds, B = P^T from p_lds) ----
 #pragma unroll
     for (int ks = 0; ks < ATTN_BLOCK_K / 32; ++ks) {
@@ -322,6 +325,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
             __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, o_acc[dt], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     advance(ti, cur);
   }
@@ -963,6 +967,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
     }
 
     // S and dP tiles (C: row = q (hi*4+r), col = kv (nt*16+lo))
+__builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
     f32x4 s_acc[NT], dp_acc[NT];
 #pragma unroll
     for (int nt = 0; nt < NT; ++nt) {
@@ -981,6 +986,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
       s_acc[nt] = sa;
       dp_acc[nt] = da;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // dS = P * (dP - delta) * scale -> p_lds as the dS A-operand
 #pragma unroll
@@ -999,6 +1005,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
       }
     }
 
+__builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
     // dQ += dS @ K  (A = dS from p_lds, B = Kt)
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -1012,6 +1019,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
             a, b, dq_acc[dt], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     __syncthreads();
     if (ti + 1 < n_tiles) {
@@ -1196,6 +1204,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       if (ti + 1 < n_tiles) { write_tile(); __syncthreads(); }
       continue;
     }
+__builtin_amdgcn_s_setprio(1);  // T5
     // S^T and dP^T tiles (C: row = kv (hi*4+r), col = q (nt*16+lo))
     f32x4 st_acc[NT], dpt_acc[NT];
 #pragma unroll
@@ -1215,6 +1224,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       st_acc[nt] = sa;
       dpt_acc[nt] = da;
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // P^T written to p_lds immediately; dS^T retained PACKED as bf16
     // (8 VGPRs, not 16 floats) — keeps the kernel under the 128-VGPR
@@ -1245,6 +1255,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       if (ti + 1 < n_tiles) { write_tile(); __syncthreads(); }
       continue;
     }
+__builtin_amdgcn_s_setprio(1);  // T5
     // dV += P^T @ dO (A = P^T via p_lds, B = dOt)
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -1258,6 +1269,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
             a, b, dv_acc[dt], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     if (ABL >= 1) {  // skip the dK group
       __syncthreads();
@@ -1270,6 +1282,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r)
         p_lds[wave][hi * 4 + r][nt * 16 + lo] = dst_pk[nt][r];
+__builtin_amdgcn_s_setprio(1);  // T5
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
@@ -1282,6 +1295,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
             a, b, dk_acc[dt], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
 
     __syncthreads();
     if (ti + 1 < n_tiles) {
