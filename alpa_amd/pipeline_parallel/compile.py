@@ -166,13 +166,14 @@ def build_pipeline_state(spec: PipelineModelSpec, method, lr: float,
         act_shape=None,  # resolved lazily from the first microbatch
         act_dtype=dtype, schedule=method.schedule, grad_sync=gs)
     engine._act_shape_fn = spec.act_shape
-    if hetero:
-        # heterogeneous boundaries: activations cross via the tile
-        # resharding exchange (batch-dim placements per stage); the
+    if hetero or spec.boundary_parts is not None:
+        # heterogeneous boundaries (or non-default activation layouts):
+        # activations cross via the tile resharding exchange; the
         # engine slices GLOBAL microbatches by its stage's dp
         engine.hetero = True
         engine.stage_meshes = stage_meshes
         engine.stage_shapes = stage_shapes
+        engine.boundary_parts_fn = spec.boundary_parts
     # rank that holds the authoritative loss (first rank of last stage)
     engine.loss_src_rank = int(stage_meshes[P - 1].ranks[0])
     # grad scale divides by the LAST stage's dp (the loss-definition dp);

@@ -59,6 +59,15 @@ class PipelineEngine:
         self.hetero = False
         self.stage_meshes = None
         self.stage_shapes = None
+        #: optional per-stage activation layout override:
+        #: fn(stage_idx, (dp, tp), act_rank) -> dim-partition tuple.
+        #: Default is batch-dim sharding over dp with tp replicas; a
+        #: stage whose boundary activation is FEATURE-sharded (e.g. a
+        #: col-parallel last layer without gather, or sequence parallel)
+        #: declares it here and the tile exchange reshards it
+        #: (reference scatter-allgather rewrite,
+        #: cross_mesh_resharding.py:995).
+        self.boundary_parts_fn = None
         self._specs = None  # (prev_fwd, prev_bwd, next_fwd, next_bwd)
 
     # ------------------------- p2p primitives -------------------------
@@ -70,12 +79,37 @@ class PipelineEngine:
             w.wait()
 
     # heterogeneous boundaries: tile-resharding exchange specs
+    def _stage_parts(self, stage: int):
+        dp, tp = self.stage_shapes[stage]
+        if self.boundary_parts_fn is not None:
+            p = self.boundary_parts_fn(stage, (dp, tp),
+                                       len(self.act_shape))
+            if p is not None:
+                return tuple(p)
+        return (dp,) + (1,) * (len(self.act_shape) - 1)
+
     def _act_placement(self, stage: int):
+        import numpy as np
         from ..parallel.resharding import Placement
         dp, tp = self.stage_shapes[stage]
         grid = self.stage_meshes[stage].grid
-        ranks = tuple(int(r) for r in grid.flatten(order="F"))
-        parts = (dp,) + (1,) * (len(self.act_shape) - 1)
+        parts = self._stage_parts(stage)
+        n_tiles = int(np.prod(parts))
+        if parts[0] == dp and n_tiles == dp:
+            # batch-only sharding, tp replicas: replica-major layout
+            # (tile b replica t = grid[b, t])
+            ranks = tuple(int(r) for r in grid.flatten(order="F"))
+        elif n_tiles == dp * tp:
+            # batch x feature tiling: row-major tiles (tile (b, t) =
+            # grid[b, t])
+            ranks = tuple(int(r) for r in grid.flatten(order="C"))
+        elif n_tiles == tp:
+            # feature-only sharding, dp replicas (tile t replica b =
+            # grid[b, t])
+            ranks = tuple(int(r) for r in grid.flatten(order="C"))
+        else:
+            raise NotImplementedError(
+                f"boundary parts {parts} on stage mesh ({dp},{tp})")
         return Placement(tuple(self.act_shape), parts, ranks)
 
     def _build_specs(self):
@@ -92,8 +126,8 @@ class PipelineEngine:
         self._specs = (pf, pb, nf, nb)
 
     def _local_tile_shape(self) -> tuple:
-        dp = self.stage_shapes[self.s][0]
-        return (self.act_shape[0] // dp,) + tuple(self.act_shape[1:])
+        parts = self._stage_parts(self.s)
+        return tuple(d // p for d, p in zip(self.act_shape, parts))
 
     def _hetero_exchange(self, send=None, recv_prev=False, recv_next=False,
                          send_prev=False, send_next=False):

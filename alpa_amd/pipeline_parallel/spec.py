@@ -44,3 +44,9 @@ class PipelineModelSpec:
     layer_flops: Optional[Sequence[float]] = None
     boundary_act_bytes: float = 0.0
     layer_param_bytes: Optional[Sequence[float]] = None
+    #: optional activation layout per stage for the boundary exchange:
+    #: fn(stage_idx, (dp, tp), act_rank) -> dim-partition tuple (None =
+    #: the default batch-over-dp layout).  Declaring a feature-sharded
+    #: boundary makes the engine reshard tiles instead of replicating
+    #: (runtime.PipelineEngine.boundary_parts_fn)
+    boundary_parts: Optional[Callable] = None

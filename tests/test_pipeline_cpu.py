@@ -553,3 +553,92 @@ def test_pipeline_with_remat_matches_serial():
     for r in results:
         for a, b in zip(r, serial):
             assert abs(a - b) < 2e-4, (r, serial)
+
+
+# ---------------------------------------------------------------------------
+# Feature-sharded boundary activations: stage 0 on a
+# (1,2) mesh emits a COLUMN-SHARDED activation (no gather), stage 1 on
+# (2,1) consumes batch shards — the tile exchange reshards feature
+# shards into batch shards (reference scatter-allgather rewrite,
+# cross_mesh_resharding.py:995).
+# ---------------------------------------------------------------------------
+
+FB, FS, FF = 4, 4, 32  # batch, seq, feature
+
+
+def _feat_stage_builder(layer_range, is_first, is_last, mesh, axis, dtype,
+                        device):
+    import torch.nn as nn
+    from alpa_amd.parallel.layers import ColumnParallelLinear
+
+    class Stage0(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = ColumnParallelLinear(FF, FF, mesh, axis, bias=False,
+                                           dtype=dtype, device=device,
+                                           init_seed=5, init_tag="s0.fc")
+
+        def forward(self, x, microbatch):
+            return torch.tanh(self.fc(microbatch["x"]))  # [B,S,FF/tp]
+
+    class Stage1(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = ColumnParallelLinear(FF, FF, None, 1, bias=False,
+                                           dtype=dtype, device=device,
+                                           init_seed=5, init_tag="s1.fc")
+
+        def forward(self, x, microbatch):
+            return self.fc(x).float().pow(2).mean()
+
+    return Stage0() if is_first else Stage1()
+
+
+def _feat_batch(i):
+    g = torch.Generator().manual_seed(40 + i)
+    return {"x": torch.randn(FB, FS, FF, generator=g)}
+
+
+def _feat_serial(steps):
+    from alpa_amd.parallel.layers import ColumnParallelLinear
+    torch.manual_seed(0)
+    fc0 = ColumnParallelLinear(FF, FF, None, 1, bias=False,
+                               init_seed=5, init_tag="s0.fc")
+    fc1 = ColumnParallelLinear(FF, FF, None, 1, bias=False,
+                               init_seed=5, init_tag="s1.fc")
+    opt = aa.AdamW(list(fc0.parameters()) + list(fc1.parameters()),
+                   lr=1e-3, weight_decay=0.0)
+    out = []
+    for i in range(steps):
+        x = _feat_batch(i)["x"]
+        loss = fc1(torch.tanh(fc0(x))).float().pow(2).mean()
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        out.append(float(loss))
+    return out
+
+
+def _feat_worker(rank, world_size):
+    from alpa_amd.pipeline_parallel.spec import PipelineModelSpec
+    method = aa.PipeshardParallel(num_micro_batches=1,
+                                  stage_mesh_shapes=[(1, 2), (2, 1)])
+    spec = PipelineModelSpec(
+        num_layers=2,
+        build_stage=_feat_stage_builder,
+        act_shape=lambda mb: (mb["x"].shape[0], FS, FF),
+        boundary_parts=lambda s, shape, rank_: (1, 1, 2) if s == 0
+        else None)
+    state = aa.TrainState.create(spec, method, lr=1e-3, weight_decay=0.0)
+    step = aa.parallelize(lambda m, b: None, method=method)
+    return [float(step(state, _feat_batch(i))) for i in range(3)]
+
+
+def test_feature_sharded_boundary_resharding():
+    """A (1,2) column-sharded boundary reshards into (2,1) batch shards
+    through the tile exchange and the 3-step trajectory matches serial."""
+    serial = _feat_serial(3)
+    results = run_distributed(_feat_worker, world_size=4, timeout=300)
+    for r in results:
+        for a, b in zip(r, serial):
+            assert abs(a - b) < 3e-4, (r, serial)
