@@ -18,6 +18,10 @@ from .mesh import (DeviceMesh, VirtualMesh, device, full_mesh, full_virtual_mesh
                    world_size)
 from .optim import AdamW
 from .shard_parallel import auto_shard, capture_graph  # noqa: F401
+from .pipeline_parallel.boundary import (automatic_remat,  # noqa: F401
+                                         manual_remat,
+                                         mark_pipeline_boundary,
+                                         spec_from_module)
 from .parallel_method import (AutoShardingOption, CreateStateParallel,
                               DataParallel, FollowParallel, ParallelMethod,
                               PipeshardParallel, ShardParallel,

@@ -208,7 +208,11 @@ def capture_graph(model: nn.Module, example_inputs) -> CapturedGraph:
                            module_path=path), node.name)
         elif node.op == "call_function":
             fn = node.target
-            if fn in _NORM_FNS:
+            if getattr(fn, "__name__", "") == "_boundary_identity":
+                # pipeline marker: sharding-transparent identity
+                add(OpDesc(node.name, "elemwise", prods[:1], shape),
+                    node.name)
+            elif fn in _NORM_FNS:
                 add(OpDesc(node.name, "norm", prods[:1], shape), node.name)
             elif fn in _ELEMWISE_FNS:
                 add(OpDesc(node.name, "elemwise", prods, shape), node.name)
