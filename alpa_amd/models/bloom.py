@@ -173,9 +173,26 @@ class BloomModel(nn.Module, GenerationMixin):
         self.dtype = dtype
         self.device_ = device
 
-    def new_cache(self, batch: int) -> KVCache:
+    def new_cache(self, batch: int, max_len: Optional[int] = None
+                  ) -> KVCache:
         return KVCache(self.cfg, self.cfg.num_layers, batch,
-                       self.heads_per_rank, self.dtype, self.device_)
+                       self.heads_per_rank, self.dtype, self.device_,
+                       max_len=max_len)
+
+    @torch.no_grad()
+    def forward_prefill(self, ids: torch.Tensor, lens: torch.Tensor,
+                        cache: KVCache) -> torch.Tensor:
+        """Batched variable-length prefill (see OPTModel.forward_prefill);
+        ALiBi biases depend only on relative positions, so right-padded
+        causal prefill stays exact for rows < len."""
+        B, S = ids.shape
+        x = self.ln_emb(self.wte(ids))
+        for i, blk in enumerate(self.blocks):
+            x = blk(x, cache.k[i], cache.v[i], 0, self.slopes)
+        cache.length = S
+        idx = (lens.to(ids.device) - 1).clamp(min=0)
+        x_last = x[torch.arange(B, device=ids.device), idx].unsqueeze(1)
+        return self.lm_head(self.ln_f(x_last))[:, 0]
 
     def forward_step(self, ids: torch.Tensor, cache: KVCache
                      ) -> torch.Tensor:

@@ -67,12 +67,14 @@ class KVCache:
     wrapper.py:356-371)."""
 
     def __init__(self, cfg: OPTConfig, num_layers: int, batch: int,
-                 heads_per_rank: int, dtype, device):
+                 heads_per_rank: int, dtype, device,
+                 max_len: Optional[int] = None):
         d = cfg.head_dim
-        self.k = [torch.zeros(batch, heads_per_rank, cfg.max_seq_len, d,
+        max_len = max_len or cfg.max_seq_len
+        self.k = [torch.zeros(batch, heads_per_rank, max_len, d,
                               dtype=dtype, device=device)
                   for _ in range(num_layers)]
-        self.v = [torch.zeros(batch, heads_per_rank, cfg.max_seq_len, d,
+        self.v = [torch.zeros(batch, heads_per_rank, max_len, d,
                               dtype=dtype, device=device)
                   for _ in range(num_layers)]
         self.length = 0
@@ -217,9 +219,30 @@ class OPTModel(nn.Module, GenerationMixin):
         self.dtype = dtype
         self.device_ = device
 
-    def new_cache(self, batch: int) -> KVCache:
+    def new_cache(self, batch: int, max_len: Optional[int] = None
+                  ) -> KVCache:
         return KVCache(self.cfg, self.cfg.num_layers, batch,
-                       self.heads_per_rank, self.dtype, self.device_)
+                       self.heads_per_rank, self.dtype, self.device_,
+                       max_len=max_len)
+
+    @torch.no_grad()
+    def forward_prefill(self, ids: torch.Tensor, lens: torch.Tensor,
+                        cache: KVCache) -> torch.Tensor:
+        """Batched variable-length prefill: ids [n, Smax] right-padded
+        prompts, lens [n] true lengths.  ONE causal pass prefills every
+        slot (padding rows only contaminate padding rows under the
+        causal mask); returns each slot's last-real-position logits
+        [n, vocab/tp].  The continuous batcher admits all pending
+        requests through this instead of per-slot prefills
+        (reference 1-D batching, opt_model_1d.py)."""
+        B, S = ids.shape
+        x = self.wte(ids) + self.wpe[2:2 + S]
+        for i, blk in enumerate(self.blocks):
+            x = blk(x, cache.k[i], cache.v[i], 0)
+        cache.length = S
+        idx = (lens.to(ids.device) - 1).clamp(min=0)
+        x_last = x[torch.arange(B, device=ids.device), idx].unsqueeze(1)
+        return self.lm_head(self.ln_f(x_last))[:, 0]
 
     def forward_step(self, ids: torch.Tensor, cache: KVCache
                      ) -> torch.Tensor:

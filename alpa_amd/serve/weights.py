@@ -1,0 +1,166 @@
+"""HuggingFace checkpoint loading for the serving model zoo (VERDICT r1
+item 9; reference ``examples/llm_serving/model/wrapper.py:501`` loads
+real OPT/BLOOM weights and ``scripts/step_*.py`` convert formats).
+
+``load_opt_hf`` / ``load_bloom_hf`` take an HF model instance or a
+``from_pretrained``-able path and return our TP-sharded serving model
+with this rank's weight shards copied in:
+
+- column layers (qkv/fc1/lm_head/embedding) take ROW slices of the HF
+  weight; row layers (out/fc2) take COLUMN slices; biases follow the
+  layer kind (column bias sharded, row bias replicated).
+- OPT's separate q/k/v projections are re-packed into our per-head
+  ``[head, {q,k,v}, head_dim]`` fused layout (head-aligned, so TP row
+  slices stay consistent); BLOOM's fused query_key_value already uses
+  exactly that layout.
+
+No network is needed: pass an in-memory HF model (tests build random
+ones) or a local save_pretrained directory.
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from ..mesh import DeviceMesh
+from ..models.bloom import BloomConfig, BloomModel
+from ..models.opt import OPTConfig, OPTModel
+
+
+def _tp(mesh: Optional[DeviceMesh], axis: int):
+    if mesh is None:
+        return 1, 0
+    return mesh.axis_size(axis), max(mesh.axis_index(axis), 0)
+
+
+def _rows(w: torch.Tensor, tp: int, idx: int) -> torch.Tensor:
+    per = w.shape[0] // tp
+    return w[idx * per:(idx + 1) * per]
+
+
+def _cols(w: torch.Tensor, tp: int, idx: int) -> torch.Tensor:
+    per = w.shape[1] // tp
+    return w[:, idx * per:(idx + 1) * per]
+
+
+def _pack_qkv(qw, kw, vw, heads: int, d: int):
+    """(3H, in) per-head-interleaved [head, {q,k,v}, d] fused weight."""
+    def per_head(w):
+        return w.reshape(heads, d, -1) if w.dim() == 2 else \
+            w.reshape(heads, d)
+    stk = torch.stack([per_head(qw), per_head(kw), per_head(vw)], dim=1)
+    return stk.reshape(heads * 3 * d, -1).squeeze(-1) if qw.dim() == 1 \
+        else stk.reshape(heads * 3 * d, -1)
+
+
+@torch.no_grad()
+def _set(param: torch.nn.Parameter, value: torch.Tensor):
+    param.copy_(value.to(dtype=param.dtype, device=param.device))
+
+
+@torch.no_grad()
+def load_opt_hf(src, mesh: Optional[DeviceMesh] = None, axis: int = 1,
+                dtype=torch.float32, device=None) -> OPTModel:
+    """HF OPTForCausalLM (instance or path) -> TP-sharded OPTModel."""
+    if isinstance(src, str):
+        from transformers import OPTForCausalLM
+        src = OPTForCausalLM.from_pretrained(src,
+                                             torch_dtype=torch.float32)
+    c = src.config
+    assert c.word_embed_proj_dim == c.hidden_size, \
+        "word_embed projection (OPT-350m style) unsupported"
+    assert getattr(c, "do_layer_norm_before", True), \
+        "post-LN OPT variants unsupported"
+    cfg = OPTConfig(hidden_size=c.hidden_size,
+                    num_layers=c.num_hidden_layers,
+                    num_heads=c.num_attention_heads,
+                    vocab_size=c.vocab_size,
+                    ffn_mult=c.ffn_dim // c.hidden_size,
+                    max_seq_len=c.max_position_embeddings)
+    model = OPTModel(cfg, mesh, axis, dtype, device)
+    tp, idx = _tp(mesh, axis)
+    sd = {k: v.float() for k, v in src.state_dict().items()}
+    dec = "model.decoder."
+    heads, d = cfg.num_heads, cfg.head_dim
+
+    _set(model.wte.weight, _rows(sd[dec + "embed_tokens.weight"], tp, idx))
+    _set(model.wpe, sd[dec + "embed_positions.weight"])
+    _set(model.ln_f.weight, sd[dec + "final_layer_norm.weight"])
+    _set(model.ln_f.bias, sd[dec + "final_layer_norm.bias"])
+    _set(model.lm_head.weight, _rows(sd["lm_head.weight"], tp, idx))
+    for i, blk in enumerate(model.blocks):
+        p = f"{dec}layers.{i}."
+        _set(blk.ln1.weight, sd[p + "self_attn_layer_norm.weight"])
+        _set(blk.ln1.bias, sd[p + "self_attn_layer_norm.bias"])
+        wq = _pack_qkv(sd[p + "self_attn.q_proj.weight"],
+                       sd[p + "self_attn.k_proj.weight"],
+                       sd[p + "self_attn.v_proj.weight"], heads, d)
+        bq = _pack_qkv(sd[p + "self_attn.q_proj.bias"],
+                       sd[p + "self_attn.k_proj.bias"],
+                       sd[p + "self_attn.v_proj.bias"], heads, d).reshape(-1)
+        _set(blk.qkv.weight, _rows(wq, tp, idx))
+        _set(blk.qkv.bias, _rows(bq.unsqueeze(-1), tp, idx).squeeze(-1))
+        _set(blk.out.weight, _cols(sd[p + "self_attn.out_proj.weight"],
+                                   tp, idx))
+        _set(blk.out.bias, sd[p + "self_attn.out_proj.bias"])
+        _set(blk.ln2.weight, sd[p + "final_layer_norm.weight"])
+        _set(blk.ln2.bias, sd[p + "final_layer_norm.bias"])
+        _set(blk.fc1.weight, _rows(sd[p + "fc1.weight"], tp, idx))
+        _set(blk.fc1.bias, _rows(sd[p + "fc1.bias"].unsqueeze(-1),
+                                 tp, idx).squeeze(-1))
+        _set(blk.fc2.weight, _cols(sd[p + "fc2.weight"], tp, idx))
+        _set(blk.fc2.bias, sd[p + "fc2.bias"])
+    return model
+
+
+@torch.no_grad()
+def load_bloom_hf(src, mesh: Optional[DeviceMesh] = None, axis: int = 1,
+                  dtype=torch.float32, device=None) -> BloomModel:
+    """HF BloomForCausalLM (instance or path) -> TP-sharded BloomModel.
+    BLOOM's fused query_key_value is already per-head [h, {q,k,v}, d]."""
+    if isinstance(src, str):
+        from transformers import BloomForCausalLM
+        src = BloomForCausalLM.from_pretrained(src,
+                                               torch_dtype=torch.float32)
+    c = src.config
+    cfg = BloomConfig(hidden_size=c.hidden_size,
+                      num_layers=c.n_layer,
+                      num_heads=c.n_head,
+                      vocab_size=c.vocab_size)
+    model = BloomModel(cfg, mesh, axis, dtype, device)
+    tp, idx = _tp(mesh, axis)
+    sd = {k: v.float() for k, v in src.state_dict().items()}
+    t = "transformer."
+
+    _set(model.wte.weight, _rows(sd[t + "word_embeddings.weight"], tp, idx))
+    _set(model.ln_emb.weight,
+         sd[t + "word_embeddings_layernorm.weight"])
+    _set(model.ln_emb.bias, sd[t + "word_embeddings_layernorm.bias"])
+    _set(model.ln_f.weight, sd[t + "ln_f.weight"])
+    _set(model.ln_f.bias, sd[t + "ln_f.bias"])
+    _set(model.lm_head.weight, _rows(sd[t + "word_embeddings.weight"],
+                                     tp, idx))
+    for i, blk in enumerate(model.blocks):
+        p = f"{t}h.{i}."
+        _set(blk.ln1.weight, sd[p + "input_layernorm.weight"])
+        _set(blk.ln1.bias, sd[p + "input_layernorm.bias"])
+        _set(blk.qkv.weight,
+             _rows(sd[p + "self_attention.query_key_value.weight"],
+                   tp, idx))
+        _set(blk.qkv.bias,
+             _rows(sd[p + "self_attention.query_key_value.bias"]
+                   .unsqueeze(-1), tp, idx).squeeze(-1))
+        _set(blk.out.weight, _cols(sd[p + "self_attention.dense.weight"],
+                                   tp, idx))
+        _set(blk.out.bias, sd[p + "self_attention.dense.bias"])
+        _set(blk.ln2.weight, sd[p + "post_attention_layernorm.weight"])
+        _set(blk.ln2.bias, sd[p + "post_attention_layernorm.bias"])
+        _set(blk.fc1.weight, _rows(sd[p + "mlp.dense_h_to_4h.weight"],
+                                   tp, idx))
+        _set(blk.fc1.bias, _rows(sd[p + "mlp.dense_h_to_4h.bias"]
+                                 .unsqueeze(-1), tp, idx).squeeze(-1))
+        _set(blk.fc2.weight, _cols(sd[p + "mlp.dense_4h_to_h.weight"],
+                                   tp, idx))
+        _set(blk.fc2.bias, sd[p + "mlp.dense_4h_to_h.bias"])
+    return model
