@@ -573,8 +573,9 @@ def test_codegen_6b_heads_on_gpu():
 
 
 def test_fp8_linear_numerics():
-    """EXPERIMENTAL fp8-forward GEMM (ops/fp8.py): forward within fp8
-    quantization error of bf16; backward exactly the bf16 GEMMs."""
+    """fp8 GEMM recipe (ops/fp8.py, round 2: fwd AND bwd fp8 with
+    delayed scaling): forward within fp8 quantization error of bf16;
+    backward grads within the e4m3 tolerance."""
     from alpa_amd.ops.fp8 import fp8_available, fp8_linear
     torch.manual_seed(19)
     x = torch.randn(512, 1024, device="cuda", dtype=torch.bfloat16,
@@ -584,15 +585,20 @@ def test_fp8_linear_numerics():
     w.retain_grad()
     assert fp8_available(x)
     b = torch.randn(2048, device="cuda", dtype=torch.bfloat16)
-    y = fp8_linear(x, w, b)
+
+    class Anchor(torch.nn.Module):
+        pass
+
+    y = fp8_linear(x, w, b, module=Anchor().cuda())
     ref = torch.nn.functional.linear(x, w, b)
     rel = (y.float() - ref.float()).abs().mean() / ref.float().abs().mean()
     assert rel < 0.06, float(rel)
     dy = torch.randn_like(y)
     y.backward(dy)
     dx_ref = dy @ w
-    torch.testing.assert_close(x.grad.float(), dx_ref.float(), rtol=1e-2,
-                               atol=1e-2)
+    rel = (x.grad.float() - dx_ref.float()).abs().mean() / \
+        dx_ref.float().abs().mean()
+    assert rel < 0.06, float(rel)
 
 
 def test_fp8_gpt_step():
