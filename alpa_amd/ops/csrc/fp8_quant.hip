@@ -33,6 +33,7 @@ __device__ __forceinline__ void atomic_max_f32(float* addr, float v) {
 }
 
 typedef unsigned char u8x8 __attribute__((ext_vector_type(8)));
+typedef unsigned char u8x16 __attribute__((ext_vector_type(16)));
 
 // -------------------- single layout --------------------
 // src bf16 [total], q fp8 [total]; q = src / *scale; amax_out = max|src|
@@ -80,6 +81,11 @@ __global__ void fp8_quantize_dual_kernel(const short* __restrict__ src,
                                          const float* __restrict__ scale,
                                          float* __restrict__ amax_out,
                                          int64_t M, int64_t N) {
+  // Phase 1 converts a 64x64 tile into LDS; phases 2/3 drain it with
+  // 16-BYTE stores per lane (row-major q and transposed qt) so both
+  // global write streams are fully coalesced — the first cut used
+  // 4-byte scattered stores and ran 4x off the bandwidth roofline
+  // (profiles/prof_fp8 r2).
   const float inv = 1.0f / fmaxf(*scale, 1e-12f);
   __shared__ unsigned char tile[QT][QT + 4];  // +4 bytes: depivot banks
   const int64_t row0 = (int64_t)blockIdx.y * QT;
@@ -87,6 +93,8 @@ __global__ void fp8_quantize_dual_kernel(const short* __restrict__ src,
   const int t = threadIdx.x;           // 256 threads
   const int tc = (t & 15) * 4;         // 4 consecutive cols
   const int tr = t >> 4;               // 16 rows per pass
+  const bool interior =
+      (row0 + QT <= M) && (col0 + QT <= N);
   float amax = 0.f;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
@@ -117,30 +125,37 @@ __global__ void fp8_quantize_dual_kernel(const short* __restrict__ src,
       amax = fmaxf(amax, fabsf(f[j]));
       tile[r][tc + j] = b[j];
     }
-    if (gr < M && col0 + tc + 3 < N) {
-      *reinterpret_cast<uchar4*>(q + gr * N + col0 + tc) =
-          make_uchar4(b[0], b[1], b[2], b[3]);
-    } else if (gr < M) {
-      for (int j = 0; j < 4; ++j)
-        if (col0 + tc + j < N) q[gr * N + col0 + tc + j] = b[j];
-    }
   }
   __syncthreads();
-  // transposed store: qT row = original col, qT col = original row
+  // phase 2: row-major q, 16 B/lane (4 lanes cover one 64-B tile row)
+  {
+    const int r = t >> 2;               // 0..63 tile row
+    const int c16 = (t & 3) * 16;
+    const int64_t gr = row0 + r;
+    if (interior) {
+      u8x16 v;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    const int r = tr + i * 16;           // qT row offset (original col)
-    const int64_t gq = col0 + r;         // global qT row
-    if (gq >= N) continue;
-    unsigned char b[4];
+      for (int j = 0; j < 16; ++j) v[j] = tile[r][c16 + j];
+      *reinterpret_cast<u8x16*>(q + gr * N + col0 + c16) = v;
+    } else if (gr < M) {
+      for (int j = 0; j < 16; ++j)
+        if (col0 + c16 + j < N) q[gr * N + col0 + c16 + j] = tile[r][c16 + j];
+    }
+  }
+  // phase 3: transposed qt, 16 B/lane along the original row dim
+  {
+    const int r = t >> 2;               // 0..63 = original col = qt row
+    const int c16 = (t & 3) * 16;       // original rows = qt cols
+    const int64_t gq = col0 + r;
+    if (interior) {
+      u8x16 v;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) b[j] = tile[tc + j][r];
-    if (row0 + tc + 3 < M) {
-      *reinterpret_cast<uchar4*>(qt + gq * M + row0 + tc) =
-          make_uchar4(b[0], b[1], b[2], b[3]);
-    } else {
-      for (int j = 0; j < 4; ++j)
-        if (row0 + tc + j < M) qt[gq * M + row0 + tc + j] = b[j];
+      for (int j = 0; j < 16; ++j) v[j] = tile[c16 + j][r];
+      *reinterpret_cast<u8x16*>(qt + gq * M + row0 + c16) = v;
+    } else if (gq < N) {
+      for (int j = 0; j < 16; ++j)
+        if (row0 + c16 + j < M)
+          qt[gq * M + row0 + c16 + j] = tile[c16 + j][r];
     }
   }
   __shared__ float red[16];

@@ -100,7 +100,11 @@ def quantize(x2: torch.Tensor, state: _RoleState, dual: bool):
 
 def quantize_weight_cached(module, weight: torch.Tensor):
     """(wq [n,k], wqt [k,n], ws) re-quantized once per optimizer epoch in
-    eager mode; captured in-step under hipGraphs (see module doc)."""
+    eager mode; captured in-step under hipGraphs (see module doc).
+    Weight scaling is delayed like activations (running amax from the
+    kernel's own one-pass reduction — weights drift slowly per step, and
+    the separate torch abs+amax passes cost ~2% of the step in the
+    r2 profile)."""
     cached = getattr(module, "_fp8_cache", None)
     graphing = torch.cuda.is_current_stream_capturing() \
         if weight.is_cuda else False
@@ -108,15 +112,9 @@ def quantize_weight_cached(module, weight: torch.Tensor):
         return cached[1], cached[2], cached[3]
     st = _role(module, "w")
     with torch.no_grad():
-        # exact per-epoch weight scale (weights are fully known)
-        st.ensure(weight)
-        amax = weight.abs().amax().float().clamp_min(1e-12).reshape(1)
-        st.running.copy_(amax)
-        st.scale.copy_(amax / E4M3_MAX)
-        wq, wqt, _ = hip_ops().fp8_quantize(weight.contiguous(),
-                                             st.scale, True, False)
-    module._fp8_cache = (_EPOCH, wq, wqt, st.scale)
-    return wq, wqt, st.scale
+        wq, wqt, ws = quantize(weight.contiguous(), st, dual=True)
+    module._fp8_cache = (_EPOCH, wq, wqt, ws)
+    return wq, wqt, ws
 
 
 class _Fp8Linear(torch.autograd.Function):
