@@ -28,8 +28,14 @@ __device__ __forceinline__ unsigned short cvt2_fp8(float a, float b) {
 }
 
 __device__ __forceinline__ void atomic_max_f32(float* addr, float v) {
-  // positive floats compare correctly as uints
-  atomicMax(reinterpret_cast<unsigned int*>(addr), __float_as_uint(v));
+  // positive floats compare correctly as uints.  Guard with a plain
+  // read first: with ~80k blocks all reducing into ONE scalar, the
+  // serialized same-address atomics would otherwise dominate; after
+  // the first few blocks the running max is usually already >= v and
+  // the atomic is skipped (monotonic, so the race is benign).
+  volatile unsigned int* a = reinterpret_cast<volatile unsigned int*>(addr);
+  if (__float_as_uint(v) > *a)
+    atomicMax(reinterpret_cast<unsigned int*>(addr), __float_as_uint(v));
 }
 
 typedef unsigned char u8x8 __attribute__((ext_vector_type(8)));
