@@ -359,3 +359,119 @@ def scatter_to_tp(x, mesh, axis):
     if mesh is None or mesh.axis_size(axis) == 1:
         return x
     return _ScatterToParallel.apply(x, mesh, axis)
+
+
+class _GatherFromParallelDim(torch.autograd.Function):
+    """All-gather along an arbitrary dim forward; slice backward (the
+    channel-dim variant of the Megatron gather/split pair — conv
+    activations shard dim 1)."""
+
+    @staticmethod
+    def forward(ctx, x, mesh: DeviceMesh, axis: int, dim: int):
+        ctx.mesh, ctx.axis, ctx.dim = mesh, axis, dim
+        tp = mesh.axis_size(axis)
+        x = x.contiguous()
+        parts = [torch.empty_like(x) for _ in range(tp)]
+        dist.all_gather(parts, x, group=mesh.axis_group(axis))
+        return torch.cat(parts, dim=dim)
+
+    @staticmethod
+    def backward(ctx, g):
+        tp = ctx.mesh.axis_size(ctx.axis)
+        idx = ctx.mesh.axis_index(ctx.axis)
+        n = g.shape[ctx.dim] // tp
+        return (g.narrow(ctx.dim, idx * n, n).contiguous(), None, None,
+                None)
+
+
+class _ScatterToParallelDim(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, x, mesh: DeviceMesh, axis: int, dim: int):
+        ctx.mesh, ctx.axis, ctx.dim = mesh, axis, dim
+        tp = mesh.axis_size(axis)
+        idx = mesh.axis_index(axis)
+        n = x.shape[dim] // tp
+        return x.narrow(dim, idx * n, n).contiguous()
+
+    @staticmethod
+    def backward(ctx, g):
+        tp = ctx.mesh.axis_size(ctx.axis)
+        g = g.contiguous()
+        parts = [torch.empty_like(g) for _ in range(tp)]
+        dist.all_gather(parts, g, group=ctx.mesh.axis_group(ctx.axis))
+        return torch.cat(parts, dim=ctx.dim), None, None, None
+
+
+def gather_from_tp_dim(x, mesh, axis, dim):
+    if mesh is None or mesh.axis_size(axis) == 1:
+        return x
+    return _GatherFromParallelDim.apply(x, mesh, axis, dim)
+
+
+def scatter_to_tp_dim(x, mesh, axis, dim):
+    if mesh is None or mesh.axis_size(axis) == 1:
+        return x
+    return _ScatterToParallelDim.apply(x, mesh, axis, dim)
+
+
+class ColumnParallelConv2d(nn.Module):
+    """Conv2d with OUT channels sharded over the tp axis (the conv
+    analog of column-parallel: dW strategies for conv, reference
+    auto_sharding_dot_handler conv handling).  Output stays
+    channel-sharded (dim 1); per-channel consumers (BN/ReLU/pool)
+    operate on the shard unchanged."""
+
+    def __init__(self, conv: nn.Conv2d, mesh: DeviceMesh, axis: int = 1):
+        super().__init__()
+        tp = mesh.axis_size(axis)
+        idx = max(mesh.axis_index(axis), 0) if mesh.is_member else 0
+        assert conv.out_channels % tp == 0 and conv.groups == 1
+        self.mesh, self.axis = mesh, axis
+        self.stride, self.padding = conv.stride, conv.padding
+        self.dilation = conv.dilation
+        per = conv.out_channels // tp
+        with torch.no_grad():
+            self.weight = nn.Parameter(
+                conv.weight[idx * per:(idx + 1) * per].clone())
+            self.bias = nn.Parameter(
+                conv.bias[idx * per:(idx + 1) * per].clone()) \
+                if conv.bias is not None else None
+        self.out_channels, self.in_channels = conv.out_channels, \
+            conv.in_channels
+
+    def forward(self, x):
+        x = copy_to_tp(x, self.mesh, self.axis)
+        return nn.functional.conv2d(x, self.weight, self.bias,
+                                    self.stride, self.padding,
+                                    self.dilation)
+
+
+class RowParallelConv2d(nn.Module):
+    """Conv2d with IN channels sharded; input arrives channel-sharded,
+    output partial sums all-reduce over the tp axis."""
+
+    def __init__(self, conv: nn.Conv2d, mesh: DeviceMesh, axis: int = 1):
+        super().__init__()
+        tp = mesh.axis_size(axis)
+        idx = max(mesh.axis_index(axis), 0) if mesh.is_member else 0
+        assert conv.in_channels % tp == 0 and conv.groups == 1
+        self.mesh, self.axis = mesh, axis
+        self.stride, self.padding = conv.stride, conv.padding
+        self.dilation = conv.dilation
+        per = conv.in_channels // tp
+        with torch.no_grad():
+            self.weight = nn.Parameter(
+                conv.weight[:, idx * per:(idx + 1) * per].clone())
+            self.bias = nn.Parameter(conv.bias.clone()) \
+                if conv.bias is not None else None
+        self.out_channels, self.in_channels = conv.out_channels, \
+            conv.in_channels
+
+    def forward(self, x):
+        y = nn.functional.conv2d(x, self.weight, None, self.stride,
+                                 self.padding, self.dilation)
+        y = reduce_from_tp(y, self.mesh, self.axis)
+        if self.bias is not None:
+            y = y + self.bias.view(1, -1, 1, 1)
+        return y

@@ -231,14 +231,20 @@ def build_captured_graph(cap, mesh: MeshModel, dtype_bytes: int = 2,
                                        d.extra["n"], dtype_bytes,
                                        train=train)
             # only divisible shards are executable by the parallel layers
+            # (conv row-sharding splits IN CHANNELS, not k=cin*kh*kw;
+            # grouped convs stay replicated)
             def _divisible(st):
+                if d.kind == "conv" and d.extra.get("groups", 1) != 1:
+                    return st.out_spec[1] is None and (
+                        not st.in_specs or st.in_specs[0][1] is None)
                 w_ax = st.out_spec[1]
                 if w_ax is not None and \
                         d.extra["n"] % mesh.shape[w_ax] != 0:
                     return False
                 in_ax = st.in_specs[0][1] if st.in_specs else None
+                in_div = d.extra.get("cin", d.extra["k"])
                 if in_ax is not None and \
-                        d.extra["k"] % mesh.shape[in_ax] != 0:
+                        in_div % mesh.shape[in_ax] != 0:
                     return False
                 return True
             strats = [st for st in strats if _divisible(st)]

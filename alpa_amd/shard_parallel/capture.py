@@ -82,7 +82,11 @@ _ELEMWISE_METHODS = {
     "clone", "detach", "masked_fill", "add_", "mul_",
 }
 _ELEMWISE_MODULES = (nn.GELU, nn.ReLU, nn.SiLU, nn.Sigmoid, nn.Tanh,
-                     nn.Dropout, nn.Identity, nn.LeakyReLU)
+                     nn.Dropout, nn.Identity, nn.LeakyReLU,
+                     # channel-preserving CNN ops: per-channel math, so
+                     # they FOLLOW a channel-sharded conv producer
+                     nn.BatchNorm2d, nn.MaxPool2d, nn.AvgPool2d,
+                     nn.AdaptiveAvgPool2d)
 # feature-normalizing: need the full feature dim present
 _NORM_FNS = {F.layer_norm, F.softmax, F.log_softmax, F.cross_entropy,
              F.rms_norm} if hasattr(F, "rms_norm") else {
@@ -136,7 +140,15 @@ def capture_graph(model: nn.Module, example_inputs) -> CapturedGraph:
     gm = symbolic_trace(model)
     if not isinstance(example_inputs, (list, tuple)):
         example_inputs = (example_inputs,)
-    ShapeProp(gm).propagate(*example_inputs)
+    # shape propagation EXECUTES the graph; run it in eval + no_grad so
+    # stateful modules (BatchNorm running stats) are not mutated by the
+    # capture pass (gm shares buffers with the user's model)
+    was_training = gm.training
+    gm.eval()
+    with torch.no_grad():
+        ShapeProp(gm).propagate(*example_inputs)
+    if was_training:
+        gm.train()
 
     ops: List[OpDesc] = []
     idx_of: Dict[str, int] = {}
@@ -193,7 +205,9 @@ def capture_graph(model: nn.Module, example_inputs) -> CapturedGraph:
                     mod.kernel_size[1] // mod.groups
                 i = add(OpDesc(node.name, "conv", prods[:1], shape,
                                module_path=path,
-                               extra={"k": k, "n": mod.out_channels}),
+                               extra={"k": k, "n": mod.out_channels,
+                                      "cin": mod.in_channels,
+                                      "groups": mod.groups}),
                         node.name)
                 cap.by_module[path] = i
             elif isinstance(mod, _NORM_MODULES):

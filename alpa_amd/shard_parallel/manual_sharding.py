@@ -77,6 +77,14 @@ def _convert(mod: nn.Module, kind: str, mesh: DeviceMesh, axis: int):
                                          dtype=dtype, device=device)
             new.weight.copy_(_shard_rows(mod.weight, tp, idx))
             return new
+        if kind == "conv_column":
+            from ..parallel.layers import ColumnParallelConv2d
+            assert isinstance(mod, nn.Conv2d), type(mod)
+            return ColumnParallelConv2d(mod, mesh, axis)
+        if kind == "conv_row":
+            from ..parallel.layers import RowParallelConv2d
+            assert isinstance(mod, nn.Conv2d), type(mod)
+            return RowParallelConv2d(mod, mesh, axis)
     raise ValueError(f"unknown partition kind {kind!r}")
 
 

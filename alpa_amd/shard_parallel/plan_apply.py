@@ -104,7 +104,8 @@ def apply_captured_plan(model: nn.Module, cap: CapturedGraph,
 
     convertible = {
         i for i, d in enumerate(ops)
-        if d.module_path is not None and d.kind in ("matmul", "embedding")
+        if d.module_path is not None and
+        d.kind in ("matmul", "embedding", "conv")
     }
 
     # 1) decide post-gathers: a feature-sharded module output is kept
@@ -123,8 +124,8 @@ def apply_captured_plan(model: nn.Module, cap: CapturedGraph,
         for (j, slot) in real_consumers(i):
             want = specs[j][0][slot] if slot < len(specs[j][0]) \
                 else specs[j][1]
-            if j in convertible and ops[j].kind == "matmul" and \
-                    want == out_spec:
+            if j in convertible and ops[j].kind in ("matmul", "conv") \
+                    and want == out_spec:
                 continue
             keep_sharded = False
             break
@@ -135,10 +136,16 @@ def apply_captured_plan(model: nn.Module, cap: CapturedGraph,
     def effective_of(i: int) -> tuple:
         return effective[root(i)]
 
-    # 2) convert + wrap
+    # 2) convert + wrap (conv tensors shard CHANNEL dim 1; everything
+    # else the last dim)
+    from ..parallel.layers import (gather_from_tp_dim, scatter_to_tp_dim)
     from .manual_sharding import _convert
     if mesh is None or mesh.axis_size(axis) == 1:
         return model  # tp degenerates: nothing to convert
+
+    def shard_dim_of(i: int) -> int:
+        return 1 if ops[root(i)].kind == "conv" else -1
+
     for i in sorted(convertible):
         d = ops[i]
         name = plan.choices.get(i)
@@ -147,9 +154,9 @@ def apply_captured_plan(model: nn.Module, cap: CapturedGraph,
         # strategy name -> partition kind (strategies.py naming)
         kind = None
         if "_col" in name and not name.endswith("colNone"):
-            kind = "column"
+            kind = "conv_column" if d.kind == "conv" else "column"
         elif "_row" in name:
-            kind = "row"
+            kind = "conv_row" if d.kind == "conv" else "row"
         elif "_vocab" in name:
             kind = "vocab"
         mod = model.get_submodule(d.module_path)
@@ -157,19 +164,52 @@ def apply_captured_plan(model: nn.Module, cap: CapturedGraph,
         if d.inputs:
             eff = effective_of(d.inputs[0])
             want = specs[i][0][0] if specs[i][0] else eff
+            pdim = shard_dim_of(d.inputs[0])
             if eff != want:
                 if eff[1] is None and want[1] is not None:
-                    pre = (lambda x, m=mesh, a=axis: scatter_to_tp(x, m, a))
+                    pre = (lambda x, m=mesh, a=axis, dd=pdim:
+                           scatter_to_tp_dim(x, m, a, dd))
                 elif eff[1] is not None and want[1] is None:
-                    pre = (lambda x, m=mesh, a=axis: gather_from_tp(x, m, a))
+                    pre = (lambda x, m=mesh, a=axis, dd=pdim:
+                           gather_from_tp_dim(x, m, a, dd))
         post = None
         if i in post_gather:
-            post = (lambda y, m=mesh, a=axis: gather_from_tp(y, m, a))
+            mdim = 1 if d.kind == "conv" else -1
+            post = (lambda y, m=mesh, a=axis, dd=mdim:
+                    gather_from_tp_dim(y, m, a, dd))
         new = _convert(mod, kind, mesh, axis) if kind else mod
         if pre is not None or post is not None:
             new = ReshardWrap(new, pre=pre, post=post)
         if new is not mod:
             _set_submodule(model, d.module_path, new)
+
+    # 3) per-channel follow modules living on a KEPT-SHARDED conv chain
+    # (BatchNorm between a col-conv and a row-conv) shard their own
+    # channel params/stats to the local slice
+    tp = mesh.axis_size(axis)
+    idx = max(mesh.axis_index(axis), 0) if mesh.is_member else 0
+    for j, d in enumerate(ops):
+        if d.kind != "elemwise" or d.module_path is None or not d.inputs:
+            continue
+        r = root(j)
+        if ops[r].kind != "conv" or effective[r][1] is None:
+            continue
+        mod = model.get_submodule(d.module_path)
+        import torch.nn as _nn
+        if isinstance(mod, _nn.BatchNorm2d) and mod.num_features % tp == 0:
+            per = mod.num_features // tp
+            with torch.no_grad():
+                for attr in ("weight", "bias", "running_mean",
+                             "running_var"):
+                    t = getattr(mod, attr, None)
+                    if t is not None and t.numel() == mod.num_features:
+                        src = t[idx * per:(idx + 1) * per].clone()
+                        if isinstance(t, torch.nn.Parameter):
+                            setattr(mod, attr,
+                                    torch.nn.Parameter(src))
+                        else:
+                            setattr(mod, attr, src)
+            mod.num_features = per
     return model
 
 

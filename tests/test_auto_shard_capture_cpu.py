@@ -299,3 +299,91 @@ def _train_worker(rank, world_size):
 def test_create_auto_trains():
     results = run_distributed(_train_worker, world_size=2, timeout=300)
     assert results[0] == results[1], results
+
+
+# ---------------------------------------------------------------------------
+# Conv (CNN family) auto-sharding: strategies enumerated for traced
+# nn.Conv2d AND executed via channel-parallel conv layers (closing the
+# r1 gap "WResNet/UNet get no TP"; reference conv strategies in
+# auto_sharding_dot_handler.cc).
+# ---------------------------------------------------------------------------
+
+
+class SmallCNN(nn.Module):
+    def __init__(self):
+        super().__init__()
+        torch.manual_seed(8)
+        self.conv1 = nn.Conv2d(3, 64, 3, padding=1)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.relu1 = nn.ReLU()
+        self.conv2 = nn.Conv2d(64, 256, 3, padding=1)
+        self.bn2 = nn.BatchNorm2d(256)
+        self.relu2 = nn.ReLU()
+        self.conv3 = nn.Conv2d(256, 256, 3, padding=1)
+        self.relu3 = nn.ReLU()
+        self.pool = nn.AdaptiveAvgPool2d(1)
+        self.fc = nn.Linear(256, 1000)
+
+    def forward(self, x):
+        x = self.relu1(self.bn1(self.conv1(x)))
+        x = self.relu2(self.bn2(self.conv2(x)))
+        x = self.relu3(self.conv3(x))
+        x = self.pool(x).flatten(1)
+        return self.fc(x)
+
+
+CNN_BUDGET = 20e6  # full replication (12 MB state + acts) infeasible
+
+
+def _cnn_x():
+    g = torch.Generator().manual_seed(12)
+    return torch.randn(8, 3, 32, 32, generator=g)
+
+
+def test_cnn_plan_shards_convs():
+    cap = capture_graph(SmallCNN(), (_cnn_x(),))
+    kinds = {d.name: d.kind for d in cap.ops}
+    assert kinds["conv2"] == "conv" and kinds["bn2"] == "elemwise"
+    plan = solve_captured(cap, 2, memory_budget=CNN_BUDGET,
+                          time_limit=20, mesh_shape=(1, 2))
+    picks = {cap.ops[i].name: s for i, s in plan.choices.items()
+             if cap.ops[i].kind in ("conv", "matmul")}
+    sharded_convs = [k for k, v in picks.items()
+                     if k.startswith("conv") and not v.endswith("colNone")]
+    assert sharded_convs, picks
+
+
+def _cnn_worker(rank, world_size):
+    from alpa_amd.shard_parallel import apply_captured_plan
+    model = SmallCNN()
+    serial = SmallCNN()
+    x = _cnn_x()
+    cap = capture_graph(model, (x,))
+    plan = solve_captured(cap, world_size, memory_budget=CNN_BUDGET,
+                          time_limit=20, mesh_shape=(1, world_size))
+    mesh = aa.DeviceMesh(list(range(world_size)), plan.mesh_shape)
+    model = apply_captured_plan(model, cap, plan, mesh)
+    model.eval()
+    serial.eval()  # BN in eval so running stats do not drift mid-check
+    out = model(x)
+    ref = serial(x)
+    torch.testing.assert_close(out.float(), ref.float(), rtol=1e-4,
+                               atol=1e-4)
+    # backward through the sharded convs
+    model.train()
+    serial.train()
+    loss = model(x).float().pow(2).mean()
+    loss.backward()
+    sloss = serial(x).float().pow(2).mean()
+    sloss.backward()
+    assert abs(float(loss) - float(sloss)) < 1e-5
+    # conv1 stays replicated in every feasible plan at this budget
+    torch.testing.assert_close(model.conv1.weight.grad,
+                               serial.conv1.weight.grad, rtol=1e-4,
+                               atol=1e-4)
+    return float(loss)
+
+
+def test_cnn_auto_shard_two_rank_parity():
+    losses = run_distributed(_cnn_worker, world_size=2, timeout=300)
+    assert abs(losses[0] - losses[1]) < 1e-7
