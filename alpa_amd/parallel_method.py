@@ -171,15 +171,54 @@ class FollowParallel(ParallelMethod):
         return self.train_method.resolve_mesh()
 
 
-def parallelize_inference(fn, state):
-    """FollowParallel execution: run fn(model, batch) under no_grad on the
-    state's mesh/placement (degenerate sharding-propagation-only compile,
-    reference follow_parallel.py:25)."""
+def parallelize_inference(fn, state, num_micro_batches: Optional[int] = None):
+    """FollowParallel execution (reference follow_parallel.py:25 /
+    parallelize(method=FollowParallel(...))): compile-free reuse of the
+    TRAINING state's placement for an eval/inference step.
+
+    - shard-parallel states: fn(model, microbatch) under no_grad with
+      the training method's microbatch split; scalar results average
+      over microbatches, tensor results concatenate on the batch dim.
+    - pipeline states: the engine's INFERENCE schedule (fill-only, no
+      backward) drives the stages; last-stage outputs are combined the
+      same way and broadcast from the loss rank so every rank returns
+      the same value (the reference's output-placement contract).
+    """
     import torch as _torch
 
+    nmb = num_micro_batches or getattr(state.method, "num_micro_batches", 1)
+
+    def _combine(outs):
+        outs = [o for o in outs if o is not None]
+        if not outs:
+            return None
+        if outs[0].dim() == 0:
+            return _torch.stack(outs).mean()
+        return _torch.cat(outs, dim=0)
+
     def run(batch):
-        with _torch.no_grad():
-            return fn(state.model, batch)
+        from .api import _split_microbatches
+        was_training = state.model.training
+        state.model.eval()
+        try:
+            micro = _split_microbatches(batch, nmb)
+            with _torch.no_grad():
+                if state.engine is not None:
+                    outs = state.engine.inference_step(micro)
+                    out = _combine(outs)
+                    import torch.distributed as dist
+                    from .mesh import is_distributed
+                    if is_distributed():
+                        if out is None or out.dim() == 0:
+                            t = (out if out is not None else
+                                 _torch.zeros(())).reshape(1).contiguous()
+                            dist.broadcast(t, src=state.engine.loss_src_rank)
+                            out = t.reshape(())
+                    return out
+                return _combine([fn(state.model, mb) for mb in micro])
+        finally:
+            if was_training:
+                state.model.train()
 
     return run
 

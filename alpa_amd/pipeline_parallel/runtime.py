@@ -408,7 +408,7 @@ class PipelineEngine:
         if self.act_shape is None:
             self.act_shape = tuple(self._act_shape_fn(microbatches[0]))
         with torch.no_grad():
-            for i in range(self.M):
+            for i in range(len(microbatches)):
                 x = None if self.is_first else self._recv_forward()
                 out = self.stage(x, microbatches[i])
                 if not self.is_last:
