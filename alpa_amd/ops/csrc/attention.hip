@@ -1256,7 +1256,19 @@ __builtin_amdgcn_s_setprio(1);  // T5
       continue;
     }
 __builtin_amdgcn_s_setprio(1);  // T5
-    // dV += P^T @ dO (A = P^T via p_lds, B = dOt)
+    // dV += P^T @ dO (A = P^T via p_lds, B = dOt).  The dS^T spill for
+    // the dK group is INTERLEAVED into this loop: after dV's ks-th read
+    // drains q-columns [32ks, 32ks+32), those p_lds columns are dead and
+    // dS^T for tiles nt = 2ks, 2ks+1 overwrites them (same-wave LDS ops
+    // complete in issue order, so no barrier is needed).  Measured
+    // NEUTRAL on the bench shape — the ablation's "+206 us dK phase" is
+    // tail LATENCY (any last MFMA group absorbs the drain into the
+    // barrier), not a write->read stall — kept because it frees the
+    // dst_pk registers before the dK group.  The remaining bwd levers
+    // are structural: 3 blocks/CU via T10 hardware-transpose reads of
+    // q/do (dropping qt/dot tiles, -27.6 KB LDS), or FA2-style dq
+    // accumulation by atomics inside this kernel (kills the separate dq
+    // kernel's S/dP recompute at the cost of fp32 atomics).
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8 a = *reinterpret_cast<const bf16x8*>(
@@ -1268,6 +1280,13 @@ __builtin_amdgcn_s_setprio(1);  // T5
         dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a, b, dv_acc[dt], 0, 0, 0);
       }
+      if (ABL == 0) {
+#pragma unroll
+        for (int nt = 2 * ks; nt < 2 * ks + 2; ++nt)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            p_lds[wave][hi * 4 + r][nt * 16 + lo] = dst_pk[nt][r];
+      }
     }
     __builtin_amdgcn_s_setprio(0);
 
@@ -1277,11 +1296,6 @@ __builtin_amdgcn_s_setprio(1);  // T5
       continue;
     }
     // dK += dS^T @ Q (A = dS^T via p_lds, B = Qt)
-#pragma unroll
-    for (int nt = 0; nt < NT; ++nt)
-#pragma unroll
-      for (int r = 0; r < 4; ++r)
-        p_lds[wave][hi * 4 + r][nt * 16 + lo] = dst_pk[nt][r];
 __builtin_amdgcn_s_setprio(1);  // T5
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
