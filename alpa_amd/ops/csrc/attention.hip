@@ -1050,7 +1050,7 @@ __builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
 //   dP^T = V dO^T      (A=V regs, B=dO_lds row-major)
 //   dS^T = P^T*(dP^T - delta[q])*scale
 //   dK += dS^T Q       (A=dS^T via p_lds, B=Qt_lds)
-template <int Dp, int ABL = 0, bool AL = false>
+template <int Dp, int ABL = 0, bool AL = false, bool LEAN = false>
 __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -1090,9 +1090,12 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
   const float al_slope = AL ? alibi[bh % H] : 0.f;
 
   __shared__ short q_lds[64][Dp + LP];
-  __shared__ short qt_lds[Dp][64 + LP];
   __shared__ short do_lds[64][Dp + LP];
-  __shared__ short dot_lds[Dp][64 + LP];
+  // LEAN: drop the transposed Q/dO tiles (-27.6 KB LDS => 3 blocks/CU)
+  // and gather the dV/dK B-fragments from the row-major tiles with
+  // strided b16 reads instead (occupancy-vs-LDS-op-count experiment)
+  __shared__ short qt_lds[LEAN ? 1 : Dp][64 + LP];
+  __shared__ short dot_lds[LEAN ? 1 : Dp][64 + LP];
   __shared__ short p_lds[NW][16][64 + LP];
   __shared__ float lse_lds[64];
   __shared__ float delta_lds[64];
@@ -1162,15 +1165,17 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
       *reinterpret_cast<bf16x8*>(&q_lds[qr + 1][dg]) = qreg[1];
       *reinterpret_cast<bf16x8*>(&do_lds[qr][dg]) = doreg[0];
       *reinterpret_cast<bf16x8*>(&do_lds[qr + 1][dg]) = doreg[1];
+      if (!LEAN) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        short2 pq, pd;
-        pq.x = qreg[0][j];
-        pq.y = qreg[1][j];
-        pd.x = doreg[0][j];
-        pd.y = doreg[1][j];
-        *reinterpret_cast<short2*>(&qt_lds[dg + j][qr]) = pq;
-        *reinterpret_cast<short2*>(&dot_lds[dg + j][qr]) = pd;
+        for (int j = 0; j < 8; ++j) {
+          short2 pq, pd;
+          pq.x = qreg[0][j];
+          pq.y = qreg[1][j];
+          pd.x = doreg[0][j];
+          pd.y = doreg[1][j];
+          *reinterpret_cast<short2*>(&qt_lds[dg + j][qr]) = pq;
+          *reinterpret_cast<short2*>(&dot_lds[dg + j][qr]) = pd;
+        }
       }
     }
     if (t < 64) {
@@ -1197,8 +1202,8 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
     }
 
     if (ABL >= 3) {  // staging only
-      float keep = bf2f(q_lds[lane & 63][0]) + bf2f(dot_lds[0][lane & 63]) +
-                   bf2f(qt_lds[0][lane & 63]) + bf2f(do_lds[lane & 63][0]);
+      float keep = bf2f(q_lds[lane & 63][0]) + bf2f(do_lds[0][lane & 63]) +
+                   bf2f(q_lds[0][lane & 63]) + bf2f(do_lds[lane & 63][0]);
       asm volatile("" ::"v"(keep));
       __syncthreads();
       if (ti + 1 < n_tiles) { write_tile(); __syncthreads(); }
@@ -1275,8 +1280,15 @@ __builtin_amdgcn_s_setprio(1);  // T5
           &p_lds[wave][lo][ks * 32 + hi * 8]);
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            &dot_lds[dt * 16 + lo][ks * 32 + hi * 8]);
+        bf16x8 b;
+        if (LEAN) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            b[j] = do_lds[ks * 32 + hi * 8 + j][dt * 16 + lo];
+        } else {
+          b = *reinterpret_cast<const bf16x8*>(
+              &dot_lds[dt * 16 + lo][ks * 32 + hi * 8]);
+        }
         dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a, b, dv_acc[dt], 0, 0, 0);
       }
@@ -1303,8 +1315,15 @@ __builtin_amdgcn_s_setprio(1);  // T5
           &p_lds[wave][lo][ks * 32 + hi * 8]);
 #pragma unroll
       for (int dt = 0; dt < DT; ++dt) {
-        bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            &qt_lds[dt * 16 + lo][ks * 32 + hi * 8]);
+        bf16x8 b;
+        if (LEAN) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            b[j] = q_lds[ks * 32 + hi * 8 + j][dt * 16 + lo];
+        } else {
+          b = *reinterpret_cast<const bf16x8*>(
+              &qt_lds[dt * 16 + lo][ks * 32 + hi * 8]);
+        }
         dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             a, b, dk_acc[dt], 0, 0, 0);
       }
@@ -1415,7 +1434,15 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
         (const short*)q, (const short*)k, (const short*)v,                   \
         (const short*)dout, lse, delta_ws, (short*)dq, (int)H, (int)S,       \
         (int)Skv, (int)D, scale, causal, ALP, st);                           \
-    attn_bwd_dkv_kernel<DP, 0, ALB><<<grid_kv, block, 0, stream>>>(          \
+    if (getenv("ALPA_ATTN_DKV_LEAN") &&                                      \
+        getenv("ALPA_ATTN_DKV_LEAN")[0] == '1')                              \
+      attn_bwd_dkv_kernel<DP, 0, ALB, true>                                  \
+          <<<grid_kv, block, 0, stream>>>(                                   \
+              (const short*)q, (const short*)k, (const short*)v,             \
+              (const short*)dout, lse, delta_ws, (short*)dk, (short*)dv,     \
+              (int)H, (int)S, (int)Skv, (int)D, scale, causal, ALP, st);     \
+    else                                                                     \
+      attn_bwd_dkv_kernel<DP, 0, ALB><<<grid_kv, block, 0, stream>>>(        \
         (const short*)q, (const short*)k, (const short*)v,                   \
         (const short*)dout, lse, delta_ws, (short*)dk, (short*)dv, (int)H,   \
         (int)S, (int)Skv, (int)D, scale, causal, ALP, st);                   \
