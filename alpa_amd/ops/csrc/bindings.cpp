@@ -627,7 +627,7 @@ std::vector<at::Tensor> fp8_quantize(const at::Tensor& x,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // Bump when the Python<->extension call surface changes; checked by
   // alpa_amd/version.py (reference check_alpa_jaxlib_version, version.py:10)
-  m.attr("ABI_VERSION") = 1;
+  m.attr("ABI_VERSION") = 2;  // r2: fp8_quantize, NT-templated attention
   m.def("layer_norm_fwd", &layer_norm_fwd, "LayerNorm forward (gfx950)");
   m.def("layer_norm_bwd", &layer_norm_bwd, "LayerNorm backward (gfx950)");
   m.def("add_layer_norm_fwd", &add_layer_norm_fwd,

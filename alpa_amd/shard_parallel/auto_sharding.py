@@ -286,13 +286,41 @@ def build_captured_graph(cap, mesh: MeshModel, dtype_bytes: int = 2,
 class CapturedPlan:
     """Solved sharding for a captured program: logical mesh + one
     strategy per op (follow nodes resolved), ready for
-    plan_apply.apply_captured_plan."""
+    plan_apply.apply_captured_plan.  JSON-serializable for the
+    solve-once/replay workflow (reference ParallelPlan +
+    LoadSolutionParallelArgs, benchmark_parallel_utils.py:39)."""
     mesh_shape: Tuple[int, int]            # (dp, tp): axis 0 batch, 1 tp
     objective: float
     #: op index -> strategy name (owners only)
     choices: Dict[int, str]
     #: op index -> (in_specs, out_spec) of the chosen strategy
     specs: Dict[int, Tuple[tuple, tuple]]
+
+    def save(self, path: str):
+        import json
+        d = {"mesh_shape": list(self.mesh_shape),
+             "objective": self.objective,
+             "choices": {str(i): n for i, n in self.choices.items()},
+             "specs": {str(i): [[list(sp) for sp in ins], list(out)]
+                       for i, (ins, out) in self.specs.items()}}
+        with open(path, "w") as f:
+            json.dump(d, f, indent=2)
+
+    @staticmethod
+    def load(path: str) -> "CapturedPlan":
+        import json
+        with open(path) as f:
+            d = json.load(f)
+
+        def tup(x):
+            return tuple(None if v is None else int(v) for v in x)
+
+        return CapturedPlan(
+            mesh_shape=tuple(d["mesh_shape"]),
+            objective=d["objective"],
+            choices={int(i): n for i, n in d["choices"].items()},
+            specs={int(i): ([tup(sp) for sp in ins], tup(out))
+                   for i, (ins, out) in d["specs"].items()})
 
 
 def solve_captured(cap, num_devices: int,

@@ -219,7 +219,8 @@ def auto_shard(model: nn.Module, example_inputs,
                memory_budget: Optional[float] = None,
                force_data_parallel: bool = False,
                train: bool = True,
-               mesh_shape=None):
+               mesh_shape=None,
+               plan: Optional[CapturedPlan] = None):
     """Capture -> solve -> apply: automatic parallelization of an
     ARBITRARY plain torch module, no model_hint, no zoo membership
     (the reference's headline capability, @parallelize of any program —
@@ -236,8 +237,8 @@ def auto_shard(model: nn.Module, example_inputs,
     cap = capture_graph(model, example_inputs)
     n = num_devices or (mesh.num_devices() if mesh is not None
                         else world_size())
-    plan = None
-    if not is_distributed() or torch.distributed.get_rank() == 0:
+    if plan is None and (not is_distributed()
+                         or torch.distributed.get_rank() == 0):
         plan = solve_captured(cap, n, memory_budget=memory_budget,
                               force_data_parallel=force_data_parallel,
                               train=train, mesh_shape=mesh_shape)

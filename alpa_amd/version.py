@@ -12,7 +12,7 @@ from __future__ import annotations
 __version__ = "0.1.0"
 
 #: minimum extension ABI this Python tree can drive
-MIN_HIP_OPS_ABI = 1
+MIN_HIP_OPS_ABI = 2
 
 
 def check_hip_ops_version() -> int:

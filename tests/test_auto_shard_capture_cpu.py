@@ -387,3 +387,17 @@ def _cnn_worker(rank, world_size):
 def test_cnn_auto_shard_two_rank_parity():
     losses = run_distributed(_cnn_worker, world_size=2, timeout=300)
     assert abs(losses[0] - losses[1]) < 1e-7
+
+
+def test_captured_plan_save_load_replay(tmp_path):
+    """Solve once, save the plan, replay it without the solver — the
+    reference's LoadSolutionParallelArgs workflow
+    (benchmark_parallel_utils.py:39)."""
+    cap, plan = _plan_for(2)
+    p = str(tmp_path / "plan.json")
+    plan.save(p)
+    from alpa_amd.shard_parallel.auto_sharding import CapturedPlan
+    loaded = CapturedPlan.load(p)
+    assert loaded.mesh_shape == plan.mesh_shape
+    assert loaded.choices == plan.choices
+    assert loaded.specs == plan.specs
