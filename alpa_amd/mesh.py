@@ -291,6 +291,22 @@ class DeviceMesh:
         if is_distributed() and self.group is not None:
             dist.barrier(group=self.group)
 
+    def check_alive(self, timeout_s: float = 30.0) -> bool:
+        """Liveness probe over the mesh (reference MeshHostWorker.check_alive + driver polling, device_mesh.py:616 / pipeshard_executable.py:417): a tiny async all-reduce that every member must answer within the timeout.  Returns False instead of raising when a peer is dead/hung; detection-only, recovery = restart + restore_checkpoint, the same policy as the reference."""
+        if not is_distributed() or self.group is None:
+            return True
+        t = torch.ones(1, device=device()
+                       if torch.cuda.is_available() else "cpu")
+        try:
+            work = dist.all_reduce(t, group=self.group, async_op=True)
+            import datetime
+            ok = work.wait(datetime.timedelta(seconds=timeout_s))
+            if ok is False:
+                return False
+            return bool(abs(float(t.item()) - len(self.ranks)) < 0.5)
+        except Exception:
+            return False
+
     # -------------------- p2p (cross-mesh / pipeline) --------------------
     @staticmethod
     def send(t: torch.Tensor, dst_rank: int, tag: int = 0):

@@ -62,3 +62,25 @@ def test_virtual_mesh_slicing():
     assert r.shape == (2, 2) and r.ranks == (0, 1, 2, 3)
     assert r.rank_grid().tolist() == [[0, 1], [2, 3]]
     assert full_virtual_mesh(8).num_devices == 8
+
+
+def _alive_worker(rank, world_size):
+    import alpa_amd as aa
+    mesh = aa.DeviceMesh(list(range(world_size)), (world_size, 1))
+    return mesh.check_alive(timeout_s=30.0)
+
+
+def test_check_alive_probe():
+    """Liveness probe (reference MeshHostWorker.check_alive +
+    pipeline_check_alive polling): healthy mesh answers True on every
+    rank within the timeout."""
+    from dist_utils import run_distributed
+    results = run_distributed(_alive_worker, world_size=2, timeout=120)
+    assert all(results), results
+
+
+def test_check_alive_single_process():
+    import alpa_amd as aa
+    from alpa_amd.mesh import VirtualMesh  # noqa: F401
+    mesh = aa.DeviceMesh([0], (1, 1))
+    assert mesh.check_alive()
