@@ -60,7 +60,7 @@ __global__ __launch_bounds__(NT) void attn_fwd_kernel(
   constexpr int KSTEPS_QK = Dp / 32;
   constexpr int NTILES = ATTN_BLOCK_K / 16;  // 4
   constexpr int DTILES = Dp / 16;
-  constexpr int LP = 8;                      // LDS bank padding (16 B)
+  constexpr int LP = 4;  // padding: (Dp+LP)*2B stride -> gcd(words,32)=2, 16 distinct banks (LP=8 gave gcd 4 => 2-way conflicts)
   constexpr int NW = NT / 64;                // waves per block
   constexpr int GPR = Dp / 8;                // bf16x8 groups per kv row
   constexpr int TOTAL_G = ATTN_BLOCK_K * GPR;
@@ -372,7 +372,7 @@ __global__ __launch_bounds__(ATTN32_THREADS) void attn_fwd_kernel32(
   constexpr int KS_QK = Dp / 16;       // k-steps over head dim (K=16)
   constexpr int MT = 2;                // 2 kv m-tiles of 32 per 64-kv tile
   constexpr int DT = Dp / 32;          // d tiles of 32 (O^T rows)
-  constexpr int LP = 8;
+  constexpr int LP = 4;
   constexpr int NW = ATTN32_THREADS / 64;  // 16 waves
   constexpr int GPR = Dp / 8;
   constexpr int TOTAL_G = 64 * GPR;
@@ -851,7 +851,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dq_kernel(
   constexpr int KD = Dp / 32;
   constexpr int NT = 4;                 // 64 kv per tile
   constexpr int DT = Dp / 16;
-  constexpr int LP = 8;
+  constexpr int LP = 4;
   constexpr int NW = ATTN_BWD_THREADS / 64;
   constexpr int GPR = Dp / 8;
   constexpr int PAIRS = 32 * GPR;       // (64 rows / 2) * groups
@@ -1065,7 +1065,7 @@ __global__ __launch_bounds__(ATTN_BWD_THREADS) void attn_bwd_dkv_kernel(
   constexpr int KD = Dp / 32;
   constexpr int NT = 4;   // 64 q per tile
   constexpr int DT = Dp / 16;
-  constexpr int LP = 8;
+  constexpr int LP = 4;
   constexpr int NW = ATTN_BWD_THREADS / 64;
   constexpr int GPR = Dp / 8;
   constexpr int PAIRS = 32 * GPR;
