@@ -706,16 +706,19 @@ hipError_t launch_attn_fwd_sbuf(const void* q, const void* k, const void* v,
                                 float scale, int causal,
                                 const int64_t* strides,
                                 hipStream_t stream) {
-  dim3 grid((uint32_t)ceil_div(S, 16 * (ATTN_THREADS / 64)),
-            (uint32_t)(B * H));
-  dim3 block(ATTN_THREADS);
+  // 512 threads + single-buffered K/V = 43.25 KB LDS at Dp=96 ->
+  // THREE blocks per CU (24 waves); cross-block overlap replaces the
+  // intra-block prefetch
+  dim3 grid((uint32_t)ceil_div(S, 16 * (512 / 64)), (uint32_t)(B * H));
+  dim3 block(512);
   AttnStrides st;
   st.qb = strides[0]; st.qh = strides[1]; st.qs = strides[2];
   st.kb = strides[3]; st.kh = strides[4]; st.ks = strides[5];
   st.vb = strides[6]; st.vh = strides[7]; st.vs = strides[8];
   st.ob = strides[9]; st.oh = strides[10]; st.os = strides[11];
 #define SBUF_CASE(DP)                                                      \
-  attn_fwd_kernel<DP, 0, false, false, false><<<grid, block, 0, stream>>>( \
+  attn_fwd_kernel<DP, 0, false, false, false, 512>                         \
+      <<<grid, block, 0, stream>>>(                                        \
       (const short*)q, (const short*)k, (const short*)v, (short*)o, lse,   \
       (int)H, (int)S, (int)Skv, (int)D, scale, causal, nullptr, nullptr,   \
       st)
