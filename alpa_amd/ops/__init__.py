@@ -133,11 +133,20 @@ class _FlashAttention(torch.autograd.Function):
             scale = 1.0 / math.sqrt(q.shape[-1])
         if use_hip(q):
             if q.shape[-1] > 128:
-                # fused kernel tiles head_dim in registers up to 128;
-                # bigger heads (CodeGen-6B/16B: 256) take the blocked
-                # hipBLASLt path (no alibi there — BLOOM heads are <=128)
+                # big heads (CodeGen-6B/16B: 256): the fused Dp=256
+                # instantiation (single-buffered K/V, 512 threads) runs
+                # PREFILL 3.3-5x faster than the blocked hipBLASLt path
+                # (241 vs 70 TF at S=2048, tools/d256_probe.py); the
+                # blocked path keeps single-token DECODE (q rows << the
+                # kernel's 128-row block).  No alibi at D>128 (BLOOM
+                # heads are <=128).
                 assert alibi is None, "alibi requires head_dim <= 128"
-                o, lse = hip_ops().attn_fwd_blocked(q, k, v, causal, scale)
+                if q.shape[2] >= 128 and q.shape[-1] <= 256:
+                    o, lse = hip_ops().attn_fwd(q, k, v, causal, scale,
+                                                None)
+                else:
+                    o, lse = hip_ops().attn_fwd_blocked(q, k, v, causal,
+                                                        scale)
             else:
                 o, lse = hip_ops().attn_fwd(q, k, v, causal, scale, alibi)
         else:

@@ -691,6 +691,17 @@ hipError_t launch_attn_fwd(const void* q, const void* k, const void* v,
     FWD_DISPATCH(96);
   } else if (D <= 128) {
     FWD_DISPATCH(128);
+  } else if (D <= 256 && !alibi && !kv_lens) {
+    // split-D experiment for CodeGen-class heads: single-buffered K/V
+    // (85.5 KB LDS at Dp=256), 512 threads; register tiling carries 16
+    // o_acc f32x4 — expect low occupancy, measured against the blocked
+    // hipBLASLt path before becoming the default (ROUND2 item 4)
+    dim3 grid2((uint32_t)ceil_div(S, 16 * 8), (uint32_t)(B * H));
+    attn_fwd_kernel<256, 0, false, false, false, 512>
+        <<<grid2, dim3(512), 0, stream>>>(
+            (const short*)q, (const short*)k, (const short*)v, (short*)o,
+            lse, (int)H, (int)S, (int)Skv, (int)D, scale, causal,
+            nullptr, nullptr, st);
   } else {
     return hipErrorInvalidValue;
   }

@@ -336,7 +336,7 @@ at::Tensor attn_fwd_out(const at::Tensor& q, const at::Tensor& k,
   check_bf16_strided4(o, "o");
   int64_t B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   int64_t Skv = k.size(2);
-  TORCH_CHECK(D % 16 == 0 && D <= 128, "head_dim must be <=128, mult of 16");
+  TORCH_CHECK(D % 16 == 0 && D <= 256, "head_dim must be <=256, mult of 16");
   TORCH_CHECK(lse.is_contiguous() && lse.numel() == B * H * S);
   int64_t strides[12] = {q.stride(0), q.stride(1), q.stride(2),
                          k.stride(0), k.stride(1), k.stride(2),
