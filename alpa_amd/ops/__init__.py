@@ -203,6 +203,13 @@ def flash_attention_varlen(q: torch.Tensor, k: torch.Tensor,
         scale = 1.0 / math.sqrt(q.shape[-1])
     lens = kv_lens.to(dtype=torch.int32)
     if use_hip(q):
+        if q.shape[-1] > 128:
+            # the varlen (continuous-batching) template is instantiated
+            # for head_dim <= 128; CodeGen-class heads use per-request
+            # decode (unchanged from r1 — the Dp=256 instantiation
+            # covers the plain prefill/decode path only)
+            raise NotImplementedError(
+                "varlen attention requires head_dim <= 128")
         o, _ = hip_ops().attn_fwd(q.contiguous(), k.contiguous(),
                                   v.contiguous(), False, scale,
                                   alibi_slopes, lens.contiguous())
