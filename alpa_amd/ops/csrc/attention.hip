@@ -290,16 +290,18 @@ __builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
     float part = 0.f;
 #pragma unroll
     for (int nt = 0; nt < NTILES; ++nt) {
+      bf16x4 pk;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float pval = (s_acc[nt][r] == -INFINITY)
                          ? 0.f
                          : __expf(s_acc[nt][r] - m_state);
         part += pval;
-        // P^T transposed store: p_lds[q][kv] so the PV B-fragment reads
-        // are contiguous
-        p_lds[wave][lo][nt * 16 + hi * 4 + r] = f2bf(pval);
+        pk[r] = f2bf(pval);
       }
+      // P^T transposed store: p_lds[q][kv], 4 consecutive kv per lane
+      // packed into ONE ds_write_b64 (was 4 b16 writes)
+      *reinterpret_cast<bf16x4*>(&p_lds[wave][lo][nt * 16 + hi * 4]) = pk;
     }
     part += __shfl_xor(part, 16, 64);
     part += __shfl_xor(part, 32, 64);
