@@ -39,6 +39,10 @@ class GlobalConfig:
     fp8_wgrad: bool = field(
         default_factory=lambda: os.environ.get("ALPA_AMD_FP8_WGRAD",
                                                "1") == "1")
+    #: scatter-allgather resharding rewrite at replicated stage
+    #: boundaries (ship 1/R per replica + intra-group all-gather;
+    #: reference use_local_allgather, global_env.py:72)
+    use_local_allgather: bool = True
     #: use hand-written HIP kernels when their extension is available
     use_hip_kernels: bool = True
     #: fail loudly if running on GPU without the HIP extension (anti-silent-fallback)

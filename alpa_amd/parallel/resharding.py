@@ -99,6 +99,28 @@ class TileTransfer:
     dst_offset: Index    # region relative to the dst tile
 
 
+@dataclass(frozen=True)
+class AllGatherFix:
+    """Post-exchange intra-group all-gather for one replicated dst tile
+    (the scatter-allgather rewrite, reference _rewrite_allgather_spec,
+    cross_mesh_resharding.py:995): the p2p phase delivers each replica
+    only 1/R of the tile; the replicas then all-gather locally."""
+    owners: Tuple[int, ...]
+    split_dim: int
+
+
+def _replica_group_available(owners: Tuple[int, ...]) -> bool:
+    """scatter-allgather needs a pre-created process group over the
+    replica set; DeviceMesh construction creates every row/column group
+    on every rank, so boundary replica sets are always cached.  Unknown
+    sets fall back to full-replica sends (creating a group here would be
+    a collective only a subset reaches)."""
+    from ..mesh import _GROUP_CACHE, is_distributed as _isd
+    if not _isd():
+        return False
+    return tuple(owners) in _GROUP_CACHE
+
+
 @dataclass
 class ReshardingTaskSpec:
     """All transfers needed to convert src placement -> dst placement
@@ -106,13 +128,53 @@ class ReshardingTaskSpec:
     src: Placement
     dst: Placement
     transfers: List[TileTransfer]
+    ag_fixes: List[AllGatherFix] = None
 
     @staticmethod
-    def build(src: Placement, dst: Placement) -> "ReshardingTaskSpec":
+    def build(src: Placement, dst: Placement,
+              scatter_allgather: bool = False) -> "ReshardingTaskSpec":
+        """scatter_allgather: for dst tiles with R replicas, ship 1/R of
+        the tile to each replica and all-gather inside the replica group
+        afterwards — cross-placement traffic drops by R (reference local
+        all-gather optimization, use_local_allgather)."""
         assert src.global_shape == dst.global_shape
         transfers = []
+        ag_fixes = []
         src_tiles = src.tiles()
         for dst_idx, dst_owners in dst.tiles():
+            R = len(dst_owners)
+            sa_dim = None
+            if scatter_allgather and R > 1 and \
+                    _replica_group_available(tuple(dst_owners)):
+                dims = sorted(range(len(dst_idx)),
+                              key=lambda d: -(dst_idx[d][1] - dst_idx[d][0]))
+                for d in dims:
+                    if (dst_idx[d][1] - dst_idx[d][0]) % R == 0 and \
+                            (dst_idx[d][1] - dst_idx[d][0]) >= R:
+                        sa_dim = d
+                        break
+            if sa_dim is not None:
+                lo0, hi0 = dst_idx[sa_dim]
+                per = (hi0 - lo0) // R
+                for r_i, d_own in enumerate(dst_owners):
+                    sub = list(dst_idx)
+                    sub[sa_dim] = (lo0 + r_i * per, lo0 + (r_i + 1) * per)
+                    sub = tuple(sub)
+                    for src_idx, src_owners in src_tiles:
+                        inter = _intersect(sub, src_idx)
+                        if inter is None:
+                            continue
+                        s_own = src_owners[d_own % len(src_owners)]
+                        transfers.append(TileTransfer(
+                            src_rank=s_own, dst_rank=d_own, region=inter,
+                            src_offset=tuple(
+                                (lo - s0, hi - s0) for (lo, hi),
+                                (s0, _) in zip(inter, src_idx)),
+                            dst_offset=tuple(
+                                (lo - d0, hi - d0) for (lo, hi),
+                                (d0, _) in zip(inter, dst_idx))))
+                ag_fixes.append(AllGatherFix(tuple(dst_owners), sa_dim))
+                continue
             for d_own in dst_owners:
                 for src_idx, src_owners in src_tiles:
                     inter = _intersect(dst_idx, src_idx)
@@ -129,7 +191,7 @@ class ReshardingTaskSpec:
                                          (d0, _) in zip(inter, dst_idx))))
         # deterministic global order => deadlock-free paired exchange
         transfers.sort(key=lambda t: (t.src_rank, t.dst_rank, t.region))
-        return ReshardingTaskSpec(src, dst, transfers)
+        return ReshardingTaskSpec(src, dst, transfers, ag_fixes)
 
     def total_bytes(self, elem_size: int = 2) -> int:
         n = 0
@@ -182,6 +244,29 @@ def apply_fixups(fixups):
             _slice(dst_buf, t.dst_offset).copy_(buf)
 
 
+def apply_allgather_fixes(spec: ReshardingTaskSpec,
+                          dst_buf: Optional[torch.Tensor]):
+    """Scatter-allgather phase 2: replicas of each dst tile exchange
+    their 1/R stripes (runs after the p2p fixups)."""
+    if not spec.ag_fixes or dst_buf is None or not is_distributed():
+        return
+    from ..mesh import _GROUP_CACHE
+    me = rank()
+    for f in spec.ag_fixes:
+        if me not in f.owners:
+            continue
+        g = _GROUP_CACHE.get(f.owners)
+        R = len(f.owners)
+        r = f.owners.index(me)
+        n = dst_buf.shape[f.split_dim] // R
+        mine = dst_buf.narrow(f.split_dim, r * n, n).contiguous()
+        parts = [torch.empty_like(mine) for _ in range(R)]
+        dist.all_gather(parts, mine, group=g)
+        for j in range(R):
+            if j != r:
+                dst_buf.narrow(f.split_dim, j * n, n).copy_(parts[j])
+
+
 def execute_resharding(spec: ReshardingTaskSpec,
                        local_src: Optional[torch.Tensor],
                        dst_buf: Optional[torch.Tensor]
@@ -192,4 +277,5 @@ def execute_resharding(spec: ReshardingTaskSpec,
         for w in dist.batch_isend_irecv(ops):
             w.wait()
     apply_fixups(fixups)
+    apply_allgather_fixes(spec, dst_buf)
     return dst_buf

@@ -113,16 +113,18 @@ class PipelineEngine:
         return Placement(tuple(self.act_shape), parts, ranks)
 
     def _build_specs(self):
+        from ..global_env import global_config
         from ..parallel.resharding import ReshardingTaskSpec
+        sa = global_config.use_local_allgather
         pf = pb = nf = nb = None
         if self.s > 0:
             a, b = self._act_placement(self.s - 1), self._act_placement(self.s)
-            pf = ReshardingTaskSpec.build(a, b)
-            pb = ReshardingTaskSpec.build(b, a)
+            pf = ReshardingTaskSpec.build(a, b, scatter_allgather=sa)
+            pb = ReshardingTaskSpec.build(b, a, scatter_allgather=sa)
         if self.s < self.P - 1:
             a, b = self._act_placement(self.s), self._act_placement(self.s + 1)
-            nf = ReshardingTaskSpec.build(a, b)
-            nb = ReshardingTaskSpec.build(b, a)
+            nf = ReshardingTaskSpec.build(a, b, scatter_allgather=sa)
+            nb = ReshardingTaskSpec.build(b, a, scatter_allgather=sa)
         self._specs = (pf, pb, nf, nb)
 
     def _local_tile_shape(self) -> tuple:
@@ -158,6 +160,9 @@ class PipelineEngine:
             ops += o; fixups += f
         self._p2p(ops)
         apply_fixups(fixups)
+        if buf is not None:
+            from ..parallel.resharding import apply_allgather_fixes
+            apply_allgather_fixes(pf if recv_prev else nb, buf)
         return buf
 
     def _recv_forward(self) -> torch.Tensor:

@@ -642,3 +642,26 @@ def test_feature_sharded_boundary_resharding():
     for r in results:
         for a, b in zip(r, serial):
             assert abs(a - b) < 3e-4, (r, serial)
+
+
+def _tp_stage_hetero_worker(rank, world_size):
+    """(1,1) -> (1,2) stage boundary: the second stage's tp REPLICAS
+    receive the activation via the scatter-allgather rewrite (half-tile
+    p2p + intra-pair all-gather) — trajectory must still match serial."""
+    from alpa_amd.models.gpt import GPTStage, gpt_pipeline_spec
+    method = aa.PipeshardParallel(num_micro_batches=1,
+                                  stage_mesh_shapes=[(1, 1), (1, 2)])
+    spec = gpt_pipeline_spec(CFG)
+    spec.build_stage = _stage_builder
+    state = aa.TrainState.create(spec, method, lr=1e-3)
+    step = aa.parallelize(lambda m, b: None, method=method)
+    return [float(step(state, make_batch(i))) for i in range(STEPS)]
+
+
+def test_tp_stage_scatter_allgather_matches_serial():
+    serial = run_serial(2)
+    results = run_distributed(_tp_stage_hetero_worker, world_size=3,
+                              timeout=300)
+    for r in results:
+        for a, b in zip(r, serial):
+            assert abs(a - b) < 3e-4, (r, serial)
