@@ -80,6 +80,20 @@ class Placement:
         return None
 
 
+def _pick_replica(owners: Sequence[int], region: Index,
+                  load: Dict[int, int]) -> int:
+    """Greedy min-load source-replica choice: assign this transfer to
+    the replica with the fewest bytes scheduled so far (ties broken by
+    rank for determinism).  Balances per-link xGMI traffic the way the
+    reference's randomized-greedy task solver does
+    (cross_mesh_resharding.py:1615), but deterministically — every rank
+    must compute the identical spec."""
+    n = int(np.prod([hi - lo for lo, hi in region]))
+    best = min(owners, key=lambda r: (load.get(r, 0), r))
+    load[best] = load.get(best, 0) + n
+    return best
+
+
 def _intersect(a: Index, b: Index) -> Optional[Index]:
     out = []
     for (a0, a1), (b0, b1) in zip(a, b):
@@ -140,6 +154,7 @@ class ReshardingTaskSpec:
         assert src.global_shape == dst.global_shape
         transfers = []
         ag_fixes = []
+        load: Dict[int, int] = {}
         src_tiles = src.tiles()
         for dst_idx, dst_owners in dst.tiles():
             R = len(dst_owners)
@@ -164,7 +179,7 @@ class ReshardingTaskSpec:
                         inter = _intersect(sub, src_idx)
                         if inter is None:
                             continue
-                        s_own = src_owners[d_own % len(src_owners)]
+                        s_own = _pick_replica(src_owners, inter, load)
                         transfers.append(TileTransfer(
                             src_rank=s_own, dst_rank=d_own, region=inter,
                             src_offset=tuple(
@@ -180,9 +195,11 @@ class ReshardingTaskSpec:
                     inter = _intersect(dst_idx, src_idx)
                     if inter is None:
                         continue
-                    # pick the source replica deterministically (balance by
-                    # dst rank, reference load-balancing is fancier)
-                    s_own = src_owners[d_own % len(src_owners)]
+                    # greedy load-balanced replica choice: the replica
+                    # with the least bytes assigned so far sends
+                    # (deterministic; reference's randomized-greedy /
+                    # DFS task solvers, cross_mesh_resharding.py:1615)
+                    s_own = _pick_replica(src_owners, inter, load)
                     transfers.append(TileTransfer(
                         src_rank=s_own, dst_rank=d_own, region=inter,
                         src_offset=tuple((lo - s0, hi - s0) for (lo, hi),

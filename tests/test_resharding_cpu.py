@@ -119,3 +119,22 @@ def _sa_worker(rank, world_size):
 def test_scatter_allgather_rewrite():
     results = run_distributed(_sa_worker, world_size=4, timeout=300)
     assert all(results)
+
+
+def test_replica_choice_load_balanced():
+    """Source replicas split the outgoing volume ~evenly (greedy
+    min-load choice; reference's load-balancing task solvers)."""
+    from collections import Counter
+    from alpa_amd.parallel.resharding import Placement, ReshardingTaskSpec
+    # src: one tile replicated on 2 ranks; dst: 8 batch shards on 8 ranks
+    src = Placement((64, 4), (1, 1), (0, 1))
+    dst = Placement((64, 4), (8, 1), tuple(range(2, 10)))
+    spec = ReshardingTaskSpec.build(src, dst)
+    bytes_per_src = Counter()
+    for t in spec.transfers:
+        import numpy as np
+        n = int(np.prod([hi - lo for lo, hi in t.region]))
+        bytes_per_src[t.src_rank] += n
+    assert set(bytes_per_src) == {0, 1}
+    a, b = bytes_per_src[0], bytes_per_src[1]
+    assert abs(a - b) <= max(a, b) * 0.34, bytes_per_src  # ~even split
