@@ -53,7 +53,9 @@ def resolve_stage_layout(method, n: Optional[int] = None,
             got = training_dp_search(
                 n, method.num_micro_batches, spec.layer_flops,
                 spec.boundary_act_bytes, spec.layer_param_bytes,
-                memory_budget=budget)
+                memory_budget=budget,
+                stage_cost_curve=spec.stage_cost_curve,
+                microbatch_tokens=spec.microbatch_tokens)
             if got is None:  # infeasible under the budget: uniform search
                 got = profiled_stage_search(
                     n, method.num_micro_batches, spec.layer_flops,

@@ -44,6 +44,12 @@ class PipelineModelSpec:
     layer_flops: Optional[Sequence[float]] = None
     boundary_act_bytes: float = 0.0
     layer_param_bytes: Optional[Sequence[float]] = None
+    #: optional MEASURED stage-cost curve name in the profiling DB
+    #: (tools/profile_stages.py, e.g. "gpt_stage_cost_h2560") and this
+    #: job's tokens per microbatch — the training DP then uses measured
+    #: per-stage times instead of the flops/curve model
+    stage_cost_curve: Optional[str] = None
+    microbatch_tokens: Optional[float] = None
     #: optional activation layout per stage for the boundary exchange:
     #: fn(stage_idx, (dp, tp), act_rank) -> dim-partition tuple (None =
     #: the default batch-over-dp layout).  Declaring a feature-sharded

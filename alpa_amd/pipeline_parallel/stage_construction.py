@@ -102,6 +102,24 @@ def _matmul_curve(db, cluster_key: str):
     return None
 
 
+def _stage_cost_curve(db, cluster_key: str, name: Optional[str]):
+    """Measured per-stage cost curve (tools/profile_stages.py writes
+    "gpt_stage_cost_h<hidden>": x = layer count, y = fwd+bwd seconds per
+    microbatch at a recorded token count) — the 1-GPU analog of the
+    reference's ProfileWorker stage measurements
+    (stage_profiling.py:310-400).  Returns (curve, ref_tokens)|None."""
+    if name is None:
+        return None
+    for (key, _shape), r in db.data.items():
+        if key != cluster_key:
+            continue
+        if name in r.op_curves:
+            tok = (getattr(r, "scalars", {}) or {}).get(
+                name + "_batch")
+            return r.op_curves[name], tok
+    return None
+
+
 def _memory_factors(db, cluster_key: str,
                     remat: bool = False) -> Tuple[float, float]:
     """(act_factor, state_factor) for the memory feasibility test:
@@ -244,7 +262,9 @@ def training_dp_search(num_devices: int, num_microbatches: int,
                        boundary_act_bytes: float = 0.0,
                        layer_param_bytes: Optional[Sequence[float]] = None,
                        db=None, cluster_key: str = "mi355x",
-                       memory_budget: Optional[float] = None
+                       memory_budget: Optional[float] = None,
+                       stage_cost_curve: Optional[str] = None,
+                       microbatch_tokens: Optional[float] = None
                        ) -> Optional[Tuple[int, List[Tuple[int, int]],
                                            List[Tuple[int, int]], float]]:
     """The reference's inter-op training DP (training_dp_impl,
@@ -297,9 +317,20 @@ def training_dp_search(num_devices: int, num_microbatches: int,
             acc += layer_flops[k]
             pre[i][k + 1] = acc
 
+    measured = _stage_cost_curve(db, cluster_key, stage_cost_curve)
+
     def stage_cost(i, k, dp, tp):
-        f = pre[i][k]
-        t = matmul_time(f / (dp * tp))
+        if measured is not None:
+            # MEASURED per-stage time at dp=tp=1, interpolated by layer
+            # count; token and tp scaling applied on top (the tp
+            # all-reduce term below still covers the comm side)
+            curve, ref_tok = measured
+            t = curve.estimate(float(k - i)) / (dp * tp)
+            if ref_tok and microbatch_tokens:
+                t *= microbatch_tokens / ref_tok
+        else:
+            f = pre[i][k]
+            t = matmul_time(f / (dp * tp))
         if tp > 1:
             nb = 4 * (k - i) * boundary_act_bytes / max(dp, 1)
             t += coll_time("all_reduce", (dp, tp), 1, nb)

@@ -95,3 +95,38 @@ def test_ilp_plan_wellformed(n, hidden, tokens, force_dp):
         for name, strat in plan.choices.items():
             assert "col0" not in strat and "col1" not in strat and \
                 "row0" not in strat and "row1" not in strat, (name, strat)
+
+
+def test_training_dp_uses_measured_stage_costs():
+    """With a measured stage-cost curve in the DB (the 1-GPU analog of
+    the reference's ProfileWorker measurements), the training DP's
+    choice follows the MEASURED nonlinearity instead of flops-linear:
+    a curve where deep stages are super-linearly expensive pushes the
+    optimum toward more, shallower stages."""
+    from alpa_amd.mesh_profiling import (CostCurve, MeshProfilingResult,
+                                         ProfilingResultDatabase)
+    from alpa_amd.pipeline_parallel.stage_construction import \
+        training_dp_search
+    db = ProfilingResultDatabase()
+    db.insert_dummy_mesh_result("mi355x", (1, 1))
+    r = db.query("mi355x", (1, 1))
+    c = CostCurve()
+    # superlinear: 8 layers cost 4x of 4+4 split
+    for L, t in ((1.0, 1e-3), (2.0, 2e-3), (4.0, 5e-3), (8.0, 40e-3)):
+        c.add(L, t)
+    r.op_curves["gpt_stage_cost_hX"] = c
+    r.scalars["gpt_stage_cost_hX_batch"] = 8192.0
+    flops = [1e12] * 8
+    with_meas = training_dp_search(
+        4, 8, flops, boundary_act_bytes=1e6,
+        layer_param_bytes=[1e8] * 8, db=db,
+        stage_cost_curve="gpt_stage_cost_hX", microbatch_tokens=8192.0)
+    without = training_dp_search(
+        4, 8, flops, boundary_act_bytes=1e6,
+        layer_param_bytes=[1e8] * 8, db=db)
+    assert with_meas is not None and without is not None
+    P_m = with_meas[0]
+    # the superlinear measured curve forces multi-stage slicing
+    assert P_m >= 2, (with_meas, without)
+    # and the reported cost comes from the measured curve's scale
+    assert with_meas[3] < 1.0
