@@ -368,4 +368,8 @@ def gpt_pipeline_spec(cfg: GPTConfig, microbatch_tokens: int = 0):
         spec.layer_flops = [per_layer] * cfg.num_layers
         spec.boundary_act_bytes = 2.0 * T * H
         spec.layer_param_bytes = [2.0 * 12 * H * H] * cfg.num_layers
+        # MEASURED stage costs (tools/profile_stages.py) anchor the
+        # training DP when the profiling DB carries this curve
+        spec.stage_cost_curve = f"gpt_stage_cost_h{H}"
+        spec.microbatch_tokens = float(T)
     return spec
