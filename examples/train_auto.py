@@ -85,10 +85,13 @@ def main():
                                  args.vocab, args.seq),
         x, method, lr=1e-4)
     if aa.rank() == 0:
-        picks = {k: v for k, v in state.plan.choices.items()}
+        def is_sharded(v):
+            return (("_col" in v and not v.endswith("colNone"))
+                    or "_row" in v or "_vocab" in v)
+        picks = state.plan.choices
         print(f"plan: mesh {state.plan.mesh_shape}, "
-              f"{sum(1 for v in picks.values() if not v.endswith('colNone'))}"
-              f" sharded ops / {len(picks)}")
+              f"{sum(1 for v in picks.values() if is_sharded(v))}"
+              f" sharded weight ops / {len(picks)} solved nodes")
 
     step = aa.parallelize(
         lambda m, b: F.cross_entropy(
