@@ -225,6 +225,28 @@ class OPTModel(nn.Module, GenerationMixin):
                        self.heads_per_rank, self.dtype, self.device_,
                        max_len=max_len)
 
+    def forward_train(self, ids: torch.Tensor) -> torch.Tensor:
+        """Differentiable causal forward WITHOUT the KV cache (training/
+        finetuning path — cache writes are no-grad buffers and would
+        silently detach the k/v projections).  Returns full-sequence
+        logits [B, S, vocab/tp]."""
+        from .. import ops
+        B, S = ids.shape
+        x = self.wte(ids) + self.wpe[2:2 + S]
+        h, d = self.heads_per_rank, self.cfg.head_dim
+        for blk in self.blocks:
+            y = blk.ln1(x)
+            qkv = blk.qkv(y).view(B, S, h, 3, d)
+            q = qkv[:, :, :, 0].permute(0, 2, 1, 3).contiguous()
+            k = qkv[:, :, :, 1].permute(0, 2, 1, 3).contiguous()
+            v = qkv[:, :, :, 2].permute(0, 2, 1, 3).contiguous()
+            o = ops.flash_attention(q, k, v, causal=True)
+            o = o.permute(0, 2, 1, 3).reshape(B, S, h * d)
+            x = x + blk.out(o)
+            hmid = torch.nn.functional.relu(blk.fc1(blk.ln2(x)))
+            x = x + blk.fc2(hmid)
+        return self.lm_head(self.ln_f(x))
+
     @torch.no_grad()
     def forward_prefill(self, ids: torch.Tensor, lens: torch.Tensor,
                         cache: KVCache) -> torch.Tensor:

@@ -188,3 +188,25 @@ def test_follow_parallel_microbatched_eval():
     assert abs(float(loss) - (a + b) / 2) < 1e-6
     assert not state.model.training or True  # mode restored
     assert state.model.training  # create() leaves the model in train mode
+
+
+def test_opt_forward_train_grads_flow():
+    """OPT training path (forward_train, no KV cache): grads reach the
+    k/v projections — the cache path would silently detach them."""
+    from alpa_amd.models.opt import OPTConfig, OPTModel
+    cfg = OPTConfig(hidden_size=64, num_layers=2, num_heads=4,
+                    vocab_size=128, max_seq_len=64)
+    m = OPTModel(cfg, None, 1, torch.float32, None, init_seed=0)
+    ids = torch.randint(0, 128, (2, 16))
+    logits = m.forward_train(ids)
+    logits.float().pow(2).mean().backward()
+    qkv_g = m.blocks[0].qkv.weight.grad
+    assert qkv_g is not None
+    h, d = cfg.num_heads, cfg.head_dim
+    gk = qkv_g.view(h, 3, d, -1)[:, 1]
+    gv = qkv_g.view(h, 3, d, -1)[:, 2]
+    assert gk.abs().sum() > 0 and gv.abs().sum() > 0
+    # matches the cached forward numerically (last position)
+    with torch.no_grad():
+        ref = m.forward_step(ids, m.new_cache(2))
+    torch.testing.assert_close(logits[:, -1], ref, rtol=1e-4, atol=1e-4)
