@@ -829,22 +829,28 @@ __global__ void attn_bwd_preprocess_kernel(const short* __restrict__ dout,
                                            int64_t dob, int64_t doh,
                                            int64_t dos, int64_t ob,
                                            int64_t oh, int64_t os) {
-  // one wave per row
-  int64_t row = (int64_t)blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  // 16 lanes x bf16x8 per row, 4 rows per wave (the one-wave-per-row
+  // version left 44/64 lanes idle at D=80 and ran 6.7x off roofline)
+  int64_t row = (int64_t)blockIdx.x * (blockDim.x >> 4) + (threadIdx.x >> 4);
   if (row >= rows) return;
-  int lane = threadIdx.x & 63;
+  int lane16 = threadIdx.x & 15;
   int64_t sidx = row % S, h = (row / S) % H, b = row / ((int64_t)S * H);
   const short* dr = dout + b * dob + h * doh + sidx * dos;
   const short* orow = o + b * ob + h * oh + sidx * os;
   float s = 0.f;
-  for (int i = lane * 4; i < D; i += 64 * 4) {
-    bf16x4 dv = *reinterpret_cast<const bf16x4*>(dr + i);
-    bf16x4 ov = *reinterpret_cast<const bf16x4*>(orow + i);
+  for (int i = lane16 * 8; i + 8 <= D; i += 16 * 8) {
+    bf16x8 dv = *reinterpret_cast<const bf16x8*>(dr + i);
+    bf16x8 ov = *reinterpret_cast<const bf16x8*>(orow + i);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) s += bf2f(dv[j]) * bf2f(ov[j]);
+    for (int j = 0; j < 8; ++j) s += bf2f(dv[j]) * bf2f(ov[j]);
   }
-  s = wave_reduce_sum(s);
-  if (lane == 0) delta[row] = s;
+  // D not a multiple of 8: scalar tail on lane 0
+  if (lane16 == 0)
+    for (int i = (D / 8) * 8; i < D; ++i)
+      s += bf2f(dr[i]) * bf2f(orow[i]);
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) s += __shfl_xor(s, off, 64);
+  if (lane16 == 0) delta[row] = s;
 }
 
 // ---------------------------------------------------------------- dq kernel
@@ -1422,7 +1428,7 @@ hipError_t launch_attn_bwd(const void* q, const void* k, const void* v,
   int64_t rows = B * H * S;
   {
     dim3 block(256);
-    dim3 grid((uint32_t)ceil_div(rows, 4));
+    dim3 grid((uint32_t)ceil_div(rows, 16));  // 16 rows per 256-thr block
     attn_bwd_preprocess_kernel<<<grid, block, 0, stream>>>(
         (const short*)dout, (const short*)o, delta_ws, rows, (int)H, (int)S,
         (int)D, st.dob, st.doh, st.dos, p[21], p[22], p[23]);

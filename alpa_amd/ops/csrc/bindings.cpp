@@ -104,6 +104,14 @@ int stripes_for(int64_t N, int64_t cols) {
   return (int)std::min<int64_t>({p, N, 512});
 }
 
+// bias_gelu's dbias partial reads 8 cols/thread: fewer column chunks,
+// so more stripes keep >=2k blocks in flight
+int stripes_for_vec8(int64_t N, int64_t cols) {
+  int64_t chunks = std::max<int64_t>(1, cols / 2048);
+  int64_t p = std::max<int64_t>(64, 2048 / chunks);
+  return (int)std::min<int64_t>({p, N, 512});
+}
+
 // ------------------------------- LayerNorm -------------------------------
 
 std::vector<at::Tensor> layer_norm_fwd(const at::Tensor& x,
@@ -231,7 +239,7 @@ std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& dy,
   int64_t N = x.numel() / F;
   auto dx = at::empty_like(x);
   auto f32 = x.options().dtype(at::kFloat);
-  const int kStripes = stripes_for(N, F);
+  const int kStripes = stripes_for_vec8(N, F);
   auto db = at::empty({F}, f32);
   auto db_part = at::empty({kStripes, F}, f32);
   HIP_OK(launch_bias_gelu_bwd(dy.const_data_ptr(), x.const_data_ptr(),

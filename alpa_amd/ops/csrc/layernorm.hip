@@ -120,22 +120,44 @@ __global__ void layer_norm_bwd_dwdb_partial(const short* __restrict__ dy,
                                             float* __restrict__ dw_part,
                                             float* __restrict__ db_part,
                                             int N, int H) {
-  const int col = blockIdx.x * LNB_COLS + threadIdx.x;
+  // 4 columns per thread (bf16x4 loads — the 1-col scalar version was
+  // issue-bound well off the HBM roofline, cf. the dbias partial)
+  const int col4 = (blockIdx.x * LNB_COLS + threadIdx.x) * 4;
   const int p = blockIdx.y;  // stripe index
   const int P = gridDim.y;
-  if (col >= H) return;
   // contiguous row range per stripe (DRAM page locality)
   const int64_t r0 = (int64_t)p * N / P, r1 = (int64_t)(p + 1) * N / P;
-  float dw = 0.f, db = 0.f;
-  for (int64_t row = r0; row < r1; ++row) {
-    float mean = mean_in[row], rstd = rstd_in[row];
-    float d = bf2f(dy[row * H + col]);
-    float xv = bf2f(x[row * H + col]);
-    dw += d * (xv - mean) * rstd;
-    db += d;
+  if (col4 + 4 <= H) {
+    float dw[4] = {0.f, 0.f, 0.f, 0.f}, db[4] = {0.f, 0.f, 0.f, 0.f};
+    for (int64_t row = r0; row < r1; ++row) {
+      float mean = mean_in[row], rstd = rstd_in[row];
+      bf16x4 dv = *reinterpret_cast<const bf16x4*>(dy + row * H + col4);
+      bf16x4 xv = *reinterpret_cast<const bf16x4*>(x + row * H + col4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float d = bf2f(dv[j]);
+        dw[j] += d * (bf2f(xv[j]) - mean) * rstd;
+        db[j] += d;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      dw_part[(int64_t)p * H + col4 + j] = dw[j];
+      db_part[(int64_t)p * H + col4 + j] = db[j];
+    }
+  } else if (col4 < H) {
+    for (int c = col4; c < H; ++c) {
+      float dw = 0.f, db = 0.f;
+      for (int64_t row = r0; row < r1; ++row) {
+        float mean = mean_in[row], rstd = rstd_in[row];
+        float d = bf2f(dy[row * H + c]);
+        dw += d * (bf2f(x[row * H + c]) - mean) * rstd;
+        db += d;
+      }
+      dw_part[(int64_t)p * H + c] = dw;
+      db_part[(int64_t)p * H + c] = db;
+    }
   }
-  dw_part[(int64_t)p * H + col] = dw;
-  db_part[(int64_t)p * H + col] = db;
 }
 
 __global__ void column_sum_kernel(const float* __restrict__ part,
@@ -174,7 +196,7 @@ hipError_t launch_layer_norm_bwd_dwdb(const void* dy, const void* x,
                                       float* dw_part, float* db_part,
                                       float* /*dw*/, float* /*db*/, int64_t N,
                                       int64_t H, int P, hipStream_t stream) {
-  dim3 grid((uint32_t)ceil_div(H, LNB_COLS), P);
+  dim3 grid((uint32_t)ceil_div(H, LNB_COLS * 4), P);
   layer_norm_bwd_dwdb_partial<<<grid, dim3(LNB_COLS), 0, stream>>>(
       (const short*)dy, (const short*)x, mean, rstd, dw_part, db_part,
       (int)N, (int)H);
