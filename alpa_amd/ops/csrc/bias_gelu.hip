@@ -116,6 +116,14 @@ hipError_t launch_bias_gelu_fwd(const void* x, const void* bias, void* y,
   return hipGetLastError();
 }
 
+hipError_t launch_colsum_partial(const void* t, float* part, int64_t N,
+                                 int64_t F, int P, hipStream_t stream) {
+  dim3 grid((uint32_t)ceil_div(F, 256 * 8), P);
+  bias_grad_partial_kernel<<<grid, dim3(256), 0, stream>>>(
+      (const short*)t, part, N, (int)F);
+  return hipGetLastError();
+}
+
 hipError_t launch_bias_gelu_bwd(const void* dy, const void* x,
                                 const void* bias, void* dx, float* db_part,
                                 float* db, int64_t N, int64_t F, int P,

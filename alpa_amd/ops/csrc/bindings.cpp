@@ -41,6 +41,8 @@ hipError_t launch_bias_gelu_fwd(const void*, const void*, void*, int64_t,
 hipError_t launch_bias_gelu_bwd(const void*, const void*, const void*, void*,
                                 float*, float*, int64_t, int64_t, int,
                                 hipStream_t);
+hipError_t launch_colsum_partial(const void*, float*, int64_t, int64_t, int,
+                                 hipStream_t);
 hipError_t launch_adamw(const void*, const void*, int, float, float, float,
                         float, float, float, int, hipStream_t);
 hipError_t launch_fp8_quantize(const void*, void*, const float*, float*,
@@ -249,6 +251,24 @@ std::vector<at::Tensor> bias_gelu_bwd(const at::Tensor& dy,
                               cur_stream()));
   at::sum_out(db, db_part, {0});
   return {dx, db};
+}
+
+// column sum of a bf16 [N, F] tensor -> fp32 [F] (striped partials +
+// at::sum): the bias-gradient reduction for plain +bias layers (the
+// at::native reduce path cost ~8 ms/step in the r2 profile)
+at::Tensor colsum_bf16(const at::Tensor& t) {
+  check_bf16_contig(t, "t");
+  int64_t F = t.size(-1);
+  int64_t N = t.numel() / F;
+  auto f32 = t.options().dtype(at::kFloat);
+  const int kStripes = stripes_for_vec8(N, F);
+  auto part = at::empty({kStripes, F}, f32);
+  auto out = at::empty({F}, f32);
+  HIP_OK(launch_colsum_partial(t.const_data_ptr(),
+                               (float*)part.mutable_data_ptr(), N, F,
+                               kStripes, cur_stream()));
+  at::sum_out(out, part, {0});
+  return out;
 }
 
 // -------------------------------- AdamW ----------------------------------
@@ -644,6 +664,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused residual-add + LayerNorm bwd (gfx950)");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "bias+GeLU forward (gfx950)");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "bias+GeLU backward (gfx950)");
+  m.def("colsum_bf16", &colsum_bf16, "striped column sum (gfx950)");
   m.def("adamw_step_raw", &adamw_step_raw, "multi-tensor AdamW (gfx950)");
   m.def("fp8_quantize", &fp8_quantize,
         "one-pass fp8 amax+cast, optional transposed twin (gfx950)");

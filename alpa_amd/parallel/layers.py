@@ -166,7 +166,7 @@ class ColumnParallelLinear(nn.Module):
             assert self.bias is not None
             return ops.bias_gelu(y, self.bias)
         if self.bias is not None:
-            y = y + self.bias
+            y = ops.bias_add(y, self.bias)
         return y
 
     def _use_fp8(self, x):
@@ -209,7 +209,7 @@ class RowParallelLinear(nn.Module):
             y = torch.matmul(x, self.weight.t())
         y = reduce_from_tp(y, self.mesh, self.axis)
         if self.bias is not None:
-            y = y + self.bias
+            y = ops.bias_add(y, self.bias)
         return y
 
 

@@ -638,3 +638,30 @@ def test_moe_scale_regression():
     y.float().square().mean().backward()
     torch.cuda.synchronize()
     assert x.grad is not None and float(x.grad.abs().sum()) > 0
+
+
+def test_colsum_matches_reference():
+    """Striped column sum vs fp32 reference (bias-grad path)."""
+    torch.manual_seed(4)
+    g = (torch.randn(4096, 2560, device="cuda") * 2).to(torch.bfloat16)
+    from alpa_amd.ops._backend import hip_ops
+    got = hip_ops().colsum_bf16(g.contiguous())
+    ref = g.float().sum(0)
+    torch.testing.assert_close(got, ref, rtol=1e-3, atol=1e-1)
+
+
+def test_bias_add_backward_matches():
+    from alpa_amd import ops
+    x = torch.randn(512, 33, 768, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    b = torch.randn(768, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ops.bias_add(x, b)
+    gy = torch.randn_like(y)
+    y.backward(gy)
+    xr = x.detach().clone().requires_grad_(True)
+    br = b.detach().clone().requires_grad_(True)
+    (xr + br).backward(gy)
+    torch.testing.assert_close(x.grad, xr.grad)
+    torch.testing.assert_close(b.grad.float(), br.grad.float(),
+                               rtol=1e-2, atol=1e-1)
