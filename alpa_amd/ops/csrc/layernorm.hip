@@ -289,6 +289,10 @@ __global__ void add_layer_norm_bwd_dx_kernel(
   const float mean = mean_in[row], rstd = rstd_in[row];
   __shared__ float red[LN_BLOCK / 64];
 
+  // two passes over the row (L2-resident for these row sizes; a
+  // register-cached single-read variant measured NEUTRAL-to-worse —
+  // the extra VGPRs bought nothing).  dh (residual grad) read is
+  // vectorized.
   float c1 = 0.f, c2 = 0.f;
   for (int i = threadIdx.x * 8; i < H; i += LN_BLOCK * 8) {
     bf16x8 dv = *reinterpret_cast<const bf16x8*>(dyr + i);
@@ -309,13 +313,16 @@ __global__ void add_layer_norm_bwd_dx_kernel(
     bf16x8 dv = *reinterpret_cast<const bf16x8*>(dyr + i);
     bf16x8 xv = *reinterpret_cast<const bf16x8*>(xr + i);
     bf16x8 wv = *reinterpret_cast<const bf16x8*>(w + i);
+    bf16x8 hv;
+    if (dhr != nullptr)
+      hv = *reinterpret_cast<const bf16x8*>(dhr + i);
     bf16x8 o;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       float wdy = bf2f(dv[j]) * bf2f(wv[j]);
       float xhat = (bf2f(xv[j]) - mean) * rstd;
       float g = (wdy - c1 - xhat * c2) * rstd;
-      if (dhr != nullptr) g += bf2f(dhr[i + j]);
+      if (dhr != nullptr) g += bf2f(hv[j]);
       o[j] = f2bf(g);
     }
     *reinterpret_cast<bf16x8*>(dxr + i) = o;
