@@ -154,6 +154,41 @@ def bias_add(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
     return _BiasAdd.apply(x, bias)
 
 
+class _LinearBias(torch.autograd.Function):
+    """y = x @ w.T + bias via torch.addmm: hipBLASLt fuses the bias
+    epilogue into the GEMM (measured free — matmul+separate add cost
+    ~16 ms per GPT-2.6B step, tools/addmm_probe.py).  Backward runs the
+    standard dX/dW GEMMs and the HIP column-sum for db."""
+
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        shape = x.shape
+        x2 = x.reshape(-1, shape[-1])
+        y = torch.addmm(bias, x2, w.t())
+        ctx.save_for_backward(x2, w)
+        ctx.bias_dtype = bias.dtype
+        return y.reshape(*shape[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, g):
+        x2, w = ctx.saved_tensors
+        g2 = g.reshape(-1, g.shape[-1]).contiguous()
+        dx = (g2 @ w).reshape(*g.shape[:-1], w.shape[1])
+        dw = g2.t() @ x2
+        if use_hip(g2) and g2.dtype == torch.bfloat16:
+            db = hip_ops().colsum_bf16(g2).to(ctx.bias_dtype)
+        else:
+            db = g2.sum(0).to(ctx.bias_dtype)
+        return dx, dw, db
+
+
+def linear_bias(x: torch.Tensor, w: torch.Tensor,
+                bias: torch.Tensor) -> torch.Tensor:
+    """F.linear with the bias fused into the hipBLASLt epilogue and the
+    bias grad on the HIP column-sum."""
+    return _LinearBias.apply(x, w, bias)
+
+
 class _FlashAttention(torch.autograd.Function):
 
     @staticmethod

@@ -161,13 +161,14 @@ class ColumnParallelLinear(nn.Module):
                 return ops.bias_gelu(fp8_linear(x, self.weight,
                                                 module=self), self.bias)
             return fp8_linear(x, self.weight, self.bias, module=self)
-        y = torch.matmul(x, self.weight.t())
         if self.gelu:
             assert self.bias is not None
-            return ops.bias_gelu(y, self.bias)
+            return ops.bias_gelu(torch.matmul(x, self.weight.t()),
+                                 self.bias)
         if self.bias is not None:
-            y = ops.bias_add(y, self.bias)
-        return y
+            # bias fused into the hipBLASLt epilogue (addmm)
+            return ops.linear_bias(x, self.weight, self.bias)
+        return torch.matmul(x, self.weight.t())
 
     def _use_fp8(self, x):
         return _fp8_ok(x, self.weight, self)
@@ -205,11 +206,16 @@ class RowParallelLinear(nn.Module):
         if _fp8_ok(x, self.weight, self):
             from ..ops.fp8 import fp8_linear
             y = fp8_linear(x, self.weight, module=self)
+        elif (self.mesh is None or self.mesh.axis_size(self.axis) == 1) \
+                and self.bias is not None:
+            # tp degenerate: no all-reduce between GEMM and bias, so the
+            # bias fuses into the hipBLASLt epilogue
+            return ops.linear_bias(x, self.weight, self.bias)
         else:
             y = torch.matmul(x, self.weight.t())
         y = reduce_from_tp(y, self.mesh, self.axis)
         if self.bias is not None:
-            y = ops.bias_add(y, self.bias)
+            y = ops.bias_add(y, self.bias)  # after the tp all-reduce
         return y
 
 
