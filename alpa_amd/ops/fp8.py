@@ -159,7 +159,9 @@ class _Fp8Linear(torch.autograd.Function):
                                   scale_b=sx, out_dtype=dy.dtype)
         else:
             dw = dy2.t() @ xsaved     # exact bf16 weight grad
-        db = dy2.sum(0) if ctx.has_bias else None
+        db = None
+        if ctx.has_bias:
+            db = hip_ops().colsum_bf16(dy2).to(dy.dtype)
         return (dx.reshape(*dy.shape[:-1], wqt.shape[0]), dw, db, None)
 
 
