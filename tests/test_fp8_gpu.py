@@ -105,3 +105,30 @@ def test_fp8_gpt_block_trains():
         assert losses[-1] < losses[0], losses
     finally:
         global_config.fp8_gemm = old
+
+
+def test_fp8_with_zero2_trains():
+    """fp8 GEMMs + ZeRO-2 sharded optimizer (single-rank world): the
+    quantized-weight caches must invalidate from ZeroOptimizer.step
+    (r1 advisor finding) and the loss must keep decreasing."""
+    import alpa_amd as aa
+    from alpa_amd.global_env import global_config
+    from alpa_amd.models.gpt import GPTConfig, GPTModel
+    old = global_config.fp8_gemm
+    global_config.fp8_gemm = True
+    try:
+        cfg = GPTConfig(hidden_size=256, num_layers=2, num_heads=4,
+                        seq_len=128, vocab_size=1024)
+        method = aa.Zero2Parallel()
+        state = aa.TrainState.create(
+            lambda mesh=None, axis=1, dtype=torch.bfloat16, device=None:
+            GPTModel(cfg, mesh, axis, torch.bfloat16,
+                     torch.device("cuda"), init_seed=1),
+            method, lr=1e-3)
+        step = aa.parallelize(lambda m, b: m.loss(*b), method=method)
+        ids = torch.randint(0, 1024, (4, 128), device="cuda")
+        losses = [float(step(state, (ids, ids))) for _ in range(6)]
+        assert all(l == l for l in losses), losses
+        assert losses[-1] < losses[0], losses
+    finally:
+        global_config.fp8_gemm = old
