@@ -88,7 +88,13 @@ def apply_captured_plan(model: nn.Module, cap: CapturedGraph,
         return i
 
     def real_consumers(i: int) -> List[Tuple[int, int]]:
-        """Consumers reachable through elementwise follow chains."""
+        """Consumers reachable through SINGLE-INPUT elementwise chains.
+        Multi-input elementwise joins (residual adds) count as real
+        consumers: their value follows their FIRST input's owner, so a
+        sharded tensor flowing into a join whose other operand is laid
+        out differently must be resolved at the edge (gathered unless
+        the join's own spec matches) — traversing through them could
+        mix shardings."""
         out, stack, seen = [], [i], set()
         while stack:
             x = stack.pop()
@@ -96,7 +102,7 @@ def apply_captured_plan(model: nn.Module, cap: CapturedGraph,
                 if j in seen:
                     continue
                 seen.add(j)
-                if ops[j].kind == "elemwise":
+                if ops[j].kind == "elemwise" and len(ops[j].inputs) <= 1:
                     stack.append(j)
                 else:
                     out.append((j, slot))
