@@ -39,6 +39,14 @@ class GlobalConfig:
     fp8_wgrad: bool = field(
         default_factory=lambda: os.environ.get("ALPA_AMD_FP8_WGRAD",
                                                "1") == "1")
+    #: compute dX with the bf16 MASTER weight instead of the fp8 wqt
+    #: cache: halves the fp8 weight-cache footprint (the +30 GB dual
+    #: cache at 15B pushes hipGraph capture OOM -> eager, see
+    #: profiles/gpt15b_fp8_eager_diag_r02).  bench.py flips this on for
+    #: >= 8B-param models; ALPA_AMD_FP8_DX_BF16=1 forces it
+    fp8_dx_bf16: bool = field(
+        default_factory=lambda: os.environ.get("ALPA_AMD_FP8_DX_BF16",
+                                               "0") == "1")
     #: scatter-allgather resharding rewrite at replicated stage
     #: boundaries (ship 1/R per replica + intra-group all-gather;
     #: reference use_local_allgather, global_env.py:72)

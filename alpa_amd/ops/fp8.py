@@ -98,7 +98,8 @@ def quantize(x2: torch.Tensor, state: _RoleState, dual: bool):
     return q, qt, used
 
 
-def quantize_weight_cached(module, weight: torch.Tensor):
+def quantize_weight_cached(module, weight: torch.Tensor,
+                           dual: bool = True):
     """(wq [n,k], wqt [k,n], ws) re-quantized once per optimizer epoch in
     eager mode; captured in-step under hipGraphs (see module doc).
     Weight scaling is delayed like activations (running amax from the
@@ -112,7 +113,7 @@ def quantize_weight_cached(module, weight: torch.Tensor):
         return cached[1], cached[2], cached[3]
     st = _role(module, "w")
     with torch.no_grad():
-        wq, wqt, ws = quantize(weight.contiguous(), st, dual=True)
+        wq, wqt, ws = quantize(weight.contiguous(), st, dual=dual)
     module._fp8_cache = (_EPOCH, wq, wqt, ws)
     return wq, wqt, ws
 
@@ -129,29 +130,38 @@ class _Fp8Linear(torch.autograd.Function):
         x2 = x.reshape(-1, shape[-1]).contiguous()
         sx = _role(module, "x")
         wgrad_fp8 = global_config.fp8_wgrad
+        dx_bf16 = global_config.fp8_dx_bf16
         xq, xqt, x_scale = quantize(x2, sx, dual=wgrad_fp8)
-        wq, wqt, ws = quantize_weight_cached(module, w)
+        wq, wqt, ws = quantize_weight_cached(module, w, dual=not dx_bf16)
         y = torch._scaled_mm(xq, wq.t(), scale_a=x_scale, scale_b=ws,
                              bias=bias, out_dtype=x.dtype)
         ctx.module = module
         ctx.wgrad_fp8 = wgrad_fp8
+        ctx.dx_bf16 = dx_bf16
+        wb = w if dx_bf16 else wqt
         if wgrad_fp8:
-            ctx.save_for_backward(xqt, wqt, x_scale, ws)
+            ctx.save_for_backward(xqt, wb, x_scale, ws)
         else:
-            ctx.save_for_backward(x2, wqt, x_scale, ws)
+            ctx.save_for_backward(x2, wb, x_scale, ws)
         ctx.has_bias = bias is not None
         return y.reshape(*shape[:-1], w.shape[0])
 
     @staticmethod
     def backward(ctx, dy):
-        xsaved, wqt, sx, ws = ctx.saved_tensors
+        xsaved, wb, sx, ws = ctx.saved_tensors
         module = ctx.module
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
-        sg = _role(module, "g")
-        dyq, dyqt, g_scale = quantize(dy2, sg, dual=ctx.wgrad_fp8)
-        # dX [M,k] = dY [M,n] @ W [n,k]; B col-major = wqt.t()
-        dx = torch._scaled_mm(dyq, wqt.t(), scale_a=g_scale, scale_b=ws,
-                              out_dtype=dy.dtype)
+        if ctx.wgrad_fp8 or not ctx.dx_bf16:
+            sg = _role(module, "g")
+            dyq, dyqt, g_scale = quantize(dy2, sg, dual=ctx.wgrad_fp8)
+        if ctx.dx_bf16:
+            # huge-model mode: dX on the bf16 master weight (no wqt
+            # cache — halves the fp8 weight footprint)
+            dx = dy2 @ wb
+        else:
+            # dX [M,k] = dY [M,n] @ W [n,k]; B col-major = wqt.t()
+            dx = torch._scaled_mm(dyq, wb.t(), scale_a=g_scale,
+                                  scale_b=ws, out_dtype=dy.dtype)
         if ctx.wgrad_fp8:
             # dW [n,k] = dY^T [n,M] @ X [M,k]; A row-major = dyqt,
             # B col-major = xqt.t()
@@ -162,7 +172,8 @@ class _Fp8Linear(torch.autograd.Function):
         db = None
         if ctx.has_bias:
             db = hip_ops().colsum_bf16(dy2).to(dy.dtype)
-        return (dx.reshape(*dy.shape[:-1], wqt.shape[0]), dw, db, None)
+        k = wb.shape[1] if ctx.dx_bf16 else wb.shape[0]
+        return (dx.reshape(*dy.shape[:-1], k), dw, db, None)
 
 
 def fp8_linear(x: torch.Tensor, weight: torch.Tensor,

@@ -90,6 +90,15 @@ def main():
     if args.fp8:
         from alpa_amd.global_env import global_config as _gc0
         _gc0.fp8_gemm = True
+        # huge models: dX on the bf16 master weight so the fp8 weight
+        # cache halves and hipGraph capture fits (see global_env)
+        try:
+            from alpa_amd.models.gpt import gpt_config
+            if torch.cuda.is_available() and args.model != "auto" and \
+                    gpt_config(args.model).num_params() > 8e9:
+                _gc0.fp8_dx_bf16 = True
+        except Exception:
+            pass
     aa.init()
     on_gpu = torch.cuda.is_available()
     n = aa.world_size()

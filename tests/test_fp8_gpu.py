@@ -132,3 +132,39 @@ def test_fp8_with_zero2_trains():
         assert losses[-1] < losses[0], losses
     finally:
         global_config.fp8_gemm = old
+
+def test_fp8_dx_bf16_mode():
+    """Huge-model mode (global_config.fp8_dx_bf16): dX on the bf16
+    master weight, no transposed weight cache — grads stay close to the
+    bf16 reference and dX is EXACT w.r.t. a bf16 matmul of dY @ W."""
+    from alpa_amd.global_env import global_config
+    from alpa_amd.ops.fp8 import fp8_linear
+    old = global_config.fp8_dx_bf16
+    global_config.fp8_dx_bf16 = True
+    try:
+        torch.manual_seed(3)
+        M, K, N = 512, 384, 256
+        x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        w = (torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+             * 0.02).requires_grad_(True)
+
+        class Anchor(torch.nn.Module):
+            pass
+        mod = Anchor().cuda()
+        y = fp8_linear(x, w, None, module=mod)
+        dy = torch.randn_like(y)
+        y.backward(dy)
+        # dX must be exactly dY @ W in bf16 (no fp8 in that path)
+        torch.testing.assert_close(x.grad, dy @ w.detach(), rtol=0, atol=0)
+        # the weight cache must be single-layout (no wqt)
+        assert mod._fp8_cache[2] is None
+        # dW still fp8 — close to the bf16 reference
+        wr = w.detach().clone().requires_grad_(True)
+        xr = x.detach().clone().requires_grad_(True)
+        torch.nn.functional.linear(xr, wr).backward(dy)
+        rel = (w.grad.float() - wr.grad.float()).abs().mean() / \
+            wr.grad.float().abs().mean().clamp_min(1e-8)
+        assert rel < 0.08, rel.item()
+    finally:
+        global_config.fp8_dx_bf16 = old
