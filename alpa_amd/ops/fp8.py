@@ -109,7 +109,8 @@ def quantize_weight_cached(module, weight: torch.Tensor,
     cached = getattr(module, "_fp8_cache", None)
     graphing = torch.cuda.is_current_stream_capturing() \
         if weight.is_cuda else False
-    if cached is not None and cached[0] == _EPOCH and not graphing:
+    if cached is not None and cached[0] == _EPOCH and not graphing \
+            and (not dual or cached[2] is not None):
         return cached[1], cached[2], cached[3]
     st = _role(module, "w")
     with torch.no_grad():
