@@ -93,7 +93,8 @@ def main():
         # huge models: dX on the bf16 master weight so the fp8 weight
         # cache halves and hipGraph capture fits (see global_env)
         try:
-            from alpa_amd.models.gpt import gpt_config
+            # gpt_config is the module-level import — re-importing it
+            # here would shadow it for the whole function
             if torch.cuda.is_available() and args.model != "auto" and \
                     gpt_config(args.model).num_params() > 8e9:
                 _gc0.fp8_dx_bf16 = True
