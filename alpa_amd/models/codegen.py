@@ -165,6 +165,7 @@ class CodeGenModel(nn.Module, GenerationMixin):
                                             dtype=dtype, device=device,
                                             init_seed=init_seed,
                                             init_tag="lm_head")
+        self.lm_head._fp8_exclude = True  # logits GEMM stays bf16
         sin, cos = rotary_tables(cfg.rotary_dim, cfg.max_seq_len, dtype,
                                  device)
         self.register_buffer("rot_sin", sin, persistent=False)

@@ -216,6 +216,7 @@ class OPTModel(nn.Module, GenerationMixin):
                                             dtype=dtype, device=device,
                                             init_seed=init_seed,
                                             init_tag="lm_head")
+        self.lm_head._fp8_exclude = True  # logits GEMM stays bf16
         self.dtype = dtype
         self.device_ = device
 

@@ -132,19 +132,26 @@ class _Fp8Linear(torch.autograd.Function):
         sx = _role(module, "x")
         wgrad_fp8 = global_config.fp8_wgrad
         dx_bf16 = global_config.fp8_dx_bf16
-        xq, xqt, x_scale = quantize(x2, sx, dual=wgrad_fp8)
-        wq, wqt, ws = quantize_weight_cached(module, w, dual=not dx_bf16)
+        # inference (serving decode/prefill under no_grad): no backward
+        # operands — single-layout quantize for both x and w, and the
+        # per-module weight cache holds ONE fp8 copy read at half the
+        # bf16 bytes/step (decode is weight-bandwidth-bound)
+        infer = not torch.is_grad_enabled()
+        xq, xqt, x_scale = quantize(x2, sx, dual=wgrad_fp8 and not infer)
+        wq, wqt, ws = quantize_weight_cached(
+            module, w, dual=(not dx_bf16) and not infer)
         y = torch._scaled_mm(xq, wq.t(), scale_a=x_scale, scale_b=ws,
                              bias=bias, out_dtype=x.dtype)
-        ctx.module = module
-        ctx.wgrad_fp8 = wgrad_fp8
-        ctx.dx_bf16 = dx_bf16
-        wb = w if dx_bf16 else wqt
-        if wgrad_fp8:
-            ctx.save_for_backward(xqt, wb, x_scale, ws)
-        else:
-            ctx.save_for_backward(x2, wb, x_scale, ws)
-        ctx.has_bias = bias is not None
+        if not infer:
+            ctx.module = module
+            ctx.wgrad_fp8 = wgrad_fp8
+            ctx.dx_bf16 = dx_bf16
+            wb = w if dx_bf16 else wqt
+            if wgrad_fp8:
+                ctx.save_for_backward(xqt, wb, x_scale, ws)
+            else:
+                ctx.save_for_backward(x2, wb, x_scale, ws)
+            ctx.has_bias = bias is not None
         return y.reshape(*shape[:-1], w.shape[0])
 
     @staticmethod
