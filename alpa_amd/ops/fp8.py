@@ -125,18 +125,16 @@ class _Fp8Linear(torch.autograd.Function):
     halves the saved-activation bytes as a side effect."""
 
     @staticmethod
-    def forward(ctx, x, w, bias, module):
+    def forward(ctx, x, w, bias, module, infer):
+        # `infer` is computed by the fp8_linear wrapper: grad mode is
+        # ALWAYS disabled inside Function.forward, so is_grad_enabled()
+        # here would misread training as inference.
         from ..global_env import global_config
         shape = x.shape
         x2 = x.reshape(-1, shape[-1]).contiguous()
         sx = _role(module, "x")
         wgrad_fp8 = global_config.fp8_wgrad
         dx_bf16 = global_config.fp8_dx_bf16
-        # inference (serving decode/prefill under no_grad): no backward
-        # operands — single-layout quantize for both x and w, and the
-        # per-module weight cache holds ONE fp8 copy read at half the
-        # bf16 bytes/step (decode is weight-bandwidth-bound)
-        infer = not torch.is_grad_enabled()
         xq, xqt, x_scale = quantize(x2, sx, dual=wgrad_fp8 and not infer)
         wq, wqt, ws = quantize_weight_cached(
             module, w, dual=(not dx_bf16) and not infer)
@@ -181,7 +179,7 @@ class _Fp8Linear(torch.autograd.Function):
         if ctx.has_bias:
             db = hip_ops().colsum_bf16(dy2).to(dy.dtype)
         k = wb.shape[1] if ctx.dx_bf16 else wb.shape[0]
-        return (dx.reshape(*dy.shape[:-1], k), dw, db, None)
+        return (dx.reshape(*dy.shape[:-1], k), dw, db, None, None)
 
 
 def fp8_linear(x: torch.Tensor, weight: torch.Tensor,
@@ -192,4 +190,9 @@ def fp8_linear(x: torch.Tensor, weight: torch.Tensor,
     `fp8_available`.  `module` anchors the per-role delayed-scaling
     state and the per-epoch weight cache."""
     assert module is not None, "fp8_linear needs the owning module"
-    return _Fp8Linear.apply(x, weight, bias, module)
+    # inference (serving decode/prefill under no_grad): no backward
+    # operands — single-layout quantize for both x and w, and the
+    # per-module weight cache holds ONE fp8 copy read at half the
+    # bf16 bytes/step (decode is weight-bandwidth-bound)
+    infer = not torch.is_grad_enabled()
+    return _Fp8Linear.apply(x, weight, bias, module, infer)
