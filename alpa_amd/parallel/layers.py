@@ -113,6 +113,13 @@ def _fp8_ok(x, weight, module=None) -> bool:
         return False
     if module is not None and getattr(module, "_fp8_exclude", False):
         return False
+    if not torch.is_grad_enabled():
+        # inference: hipBLASLt fp8 only beats bf16 from M>=128 (measured
+        # tools/fp8_decode_probe.py: 0.92x at M<=64, 1.15x at 128, 1.81x
+        # at 256) — so PREFILL goes fp8, per-token decode stays on the
+        # skinny bf16 path (also skips the per-linear quantize launches)
+        if x.numel() // x.shape[-1] < 128:
+            return False
     from ..ops.fp8 import fp8_available
     return (fp8_available(x) and x.shape[-1] % 16 == 0 and
             weight.shape[0] % 16 == 0)
