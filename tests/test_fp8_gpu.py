@@ -195,7 +195,7 @@ def test_fp8_serving_decode():
         assert rel < 0.12, rel.item()
         assert out.shape == (2, 72) and bool(out.lt(1024).all())
         # prefill used fp8 with a single-layout (inference) cache
-        qkv = m.blocks[0].attn.qkv
+        qkv = m.blocks[0].qkv
         assert getattr(qkv, "_fp8_cache")[2] is None
         # lm_head stays bf16
         assert not hasattr(m.lm_head, "_fp8_cache")
