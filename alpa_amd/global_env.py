@@ -39,6 +39,13 @@ class GlobalConfig:
     fp8_wgrad: bool = field(
         default_factory=lambda: os.environ.get("ALPA_AMD_FP8_WGRAD",
                                                "1") == "1")
+    #: hand-written decode GEMV (ops/csrc/skinny_gemm.hip) for
+    #: inference GEMMs with <= 64 tokens; ALPA_AMD_SKINNY=0 falls back
+    #: to hipBLASLt
+    skinny_gemm: bool = field(
+        default_factory=lambda: os.environ.get("ALPA_AMD_SKINNY",
+                                               "1") == "1")
+
     #: compute dX with the bf16 MASTER weight instead of the fp8 wqt
     #: cache: halves the fp8 weight-cache footprint (the +30 GB dual
     #: cache at 15B pushes hipGraph capture OOM -> eager, see

@@ -447,3 +447,80 @@ def fused_adamw(params: List[torch.Tensor], grads: List[torch.Tensor],
     else:
         ref.adamw_step(params, grads, exp_avgs, exp_avg_sqs, step, lr, beta1,
                        beta2, eps, weight_decay, grad_scale)
+
+
+# ----------------------------- skinny decode GEMM --------------------------
+
+def _skinny_splits(N: int, K: int) -> int:
+    """Split-K factor: enough workgroups to fill 256 CUs (grid is
+    N/64 x S) while keeping >= 8 k-rounds per workgroup."""
+    target = max(1, 1024 // max(1, N // 64))
+    kg = K // 8
+    s = 1
+    for cand in (2, 4, 8, 16, 32):
+        if cand > target or kg % cand or kg // cand < 8:
+            break
+        s = cand
+    return s
+
+
+def _skinny_cache(module, weight: torch.Tensor, fp8: bool):
+    """Packed [K/8, N, 8] weight (+ e4m3 scale), cached on the module and
+    keyed on the parameter version (in-place optimizer updates bump it)."""
+    key = (weight._version, fp8)
+    cached = getattr(module, "_skinny_pack", None)
+    if cached is not None and cached[0] == key:
+        return cached[1], cached[2]
+    N, K = weight.shape
+    with torch.no_grad():
+        w = weight.contiguous()
+        if fp8:
+            amax = w.abs().amax().float().clamp_min(1e-12)
+            scale = (amax / 448.0).reshape(1)
+            q = (w.float() / scale).clamp(-448.0, 448.0).to(
+                torch.float8_e4m3fn)
+            wp = q.view(torch.uint8).view(N, K // 8, 8) \
+                .permute(1, 0, 2).contiguous()
+        else:
+            scale = None
+            wp = w.view(N, K // 8, 8).permute(1, 0, 2).contiguous()
+    module._skinny_pack = (key, wp, scale)
+    return wp, scale
+
+
+def skinny_ok(x: torch.Tensor, weight: torch.Tensor, module=None) -> bool:
+    """Gate for the decode GEMV kernel (csrc/skinny_gemm.hip): inference
+    only, M <= 64 tokens, 64-aligned dims.  Measured: hipBLASLt runs
+    these shapes at ~2.8 TB/s; the packed-layout kernel streams weights
+    near the HBM roofline (tools/skinny_bench.py)."""
+    from ..global_env import global_config
+    if not global_config.skinny_gemm or torch.is_grad_enabled():
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16):
+        return False
+    if torch.cuda.is_current_stream_capturing():
+        return False
+    N, K = weight.shape
+    m = x.numel() // x.shape[-1]
+    return m <= 64 and N % 64 == 0 and K % 64 == 0
+
+
+def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
+                  bias: Optional[torch.Tensor], module) -> torch.Tensor:
+    """y = x @ W^T (+bias) through the packed decode GEMV.  fp8 weights
+    (half the bytes/token) when global_config.fp8_gemm is on and the
+    module isn't _fp8_exclude; split-K fp32 accumulation (atomicAdd
+    order nondeterministic — inference only)."""
+    from ..global_env import global_config
+    fp8 = (global_config.fp8_gemm and
+           not getattr(module, "_fp8_exclude", False))
+    wp, scale = _skinny_cache(module, weight, fp8)
+    N, K = weight.shape
+    shape = x.shape
+    x2 = x.reshape(-1, K).contiguous()
+    y32 = hip_ops().skinny_gemm(wp, x2, scale, N, K,
+                                _skinny_splits(N, K))
+    y = y32.to(x.dtype)
+    if bias is not None:
+        y = y + bias
+    return y.reshape(*shape[:-1], N)

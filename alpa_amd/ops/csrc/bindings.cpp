@@ -50,6 +50,9 @@ hipError_t launch_fp8_quantize(const void*, void*, const float*, float*,
 hipError_t launch_fp8_quantize_dual(const void*, void*, void*, const float*,
                                     float*, int64_t, int64_t, int,
                                     hipStream_t);
+hipError_t launch_skinny_gemm(const void*, const void*, float*,
+                              const float*, int64_t, int64_t, int64_t, int,
+                              int, hipStream_t);
 hipError_t launch_ce_fwd(const void*, const int64_t*, float*, float*, int64_t,
                          int64_t, hipStream_t);
 hipError_t launch_ce_bwd(const float*, const void*, const int64_t*,
@@ -269,6 +272,36 @@ at::Tensor colsum_bf16(const at::Tensor& t) {
                                kStripes, cur_stream()));
   at::sum_out(out, part, {0});
   return out;
+}
+
+// ----------------------------- skinny GEMM -------------------------------
+
+// y[M,N] fp32 = x[M,K] @ Wp^T for decode-shaped M<=64; wp is the packed
+// [K/8, N, 8] layout (bf16 or e4m3 — pass wscale for e4m3).  x rows
+// beyond M must be absent (the binding pads internally).
+at::Tensor skinny_gemm(const at::Tensor& wp, const at::Tensor& x,
+                       const c10::optional<at::Tensor>& wscale,
+                       int64_t N, int64_t K, int64_t splits) {
+  TORCH_CHECK(wp.is_cuda() && x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(x.dim() == 2 && x.size(1) == K, "x must be [M, K]");
+  const int64_t M = x.size(0);
+  TORCH_CHECK(M >= 1 && M <= 64, "skinny path is for M <= 64");
+  const bool fp8 = wscale.has_value();
+  // pad x rows up to the kernel's MT tier so the kernel needs no guards
+  int64_t MT = 4;
+  while (MT < M) MT *= 2;
+  at::Tensor xp = x;
+  if (MT != M) {
+    xp = at::zeros({MT, K}, x.options());
+    xp.narrow(0, 0, M).copy_(x);
+  }
+  auto y32 = at::zeros({MT, N}, x.options().dtype(at::kFloat));
+  HIP_OK(launch_skinny_gemm(
+      wp.const_data_ptr(), xp.const_data_ptr(),
+      (float*)y32.mutable_data_ptr(),
+      fp8 ? (const float*)wscale->const_data_ptr() : nullptr, M, N, K,
+      (int)splits, fp8 ? 1 : 0, cur_stream()));
+  return MT == M ? y32 : y32.narrow(0, 0, M);
 }
 
 // -------------------------------- AdamW ----------------------------------
@@ -665,6 +698,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "bias+GeLU forward (gfx950)");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "bias+GeLU backward (gfx950)");
   m.def("colsum_bf16", &colsum_bf16, "striped column sum (gfx950)");
+  m.def("skinny_gemm", &skinny_gemm,
+        "decode GEMV over packed weights (gfx950)");
   m.def("adamw_step_raw", &adamw_step_raw, "multi-tensor AdamW (gfx950)");
   m.def("fp8_quantize", &fp8_quantize,
         "one-pass fp8 amax+cast, optional transposed twin (gfx950)");

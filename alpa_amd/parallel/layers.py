@@ -162,6 +162,13 @@ class ColumnParallelLinear(nn.Module):
 
     def forward(self, x):
         x = copy_to_tp(x, self.mesh, self.axis)
+        if ops.skinny_ok(x, self.weight, self):
+            # decode-shaped inference (M <= 64): packed GEMV kernel
+            if self.gelu:
+                return ops.bias_gelu(
+                    ops.skinny_linear(x, self.weight, None, self),
+                    self.bias)
+            return ops.skinny_linear(x, self.weight, self.bias, self)
         if self._use_fp8(x):
             from ..ops.fp8 import fp8_linear
             if self.gelu:
@@ -210,7 +217,9 @@ class RowParallelLinear(nn.Module):
             self.register_parameter("bias", None)
 
     def forward(self, x):
-        if _fp8_ok(x, self.weight, self):
+        if ops.skinny_ok(x, self.weight, self):
+            y = ops.skinny_linear(x, self.weight, None, self)
+        elif _fp8_ok(x, self.weight, self):
             from ..ops.fp8 import fp8_linear
             y = fp8_linear(x, self.weight, module=self)
         elif (self.mesh is None or self.mesh.axis_size(self.axis) == 1) \

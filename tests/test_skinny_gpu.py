@@ -1,0 +1,81 @@
+"""Decode GEMV kernel (csrc/skinny_gemm.hip) numerics against a plain
+fp32 PyTorch reference, plus the layer-level dispatch."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.mark.parametrize("M", [1, 5, 16, 33, 64])
+def test_skinny_bf16_matches_reference(M):
+    from alpa_amd.ops import _skinny_splits
+    from alpa_amd.ops._backend import hip_ops
+    torch.manual_seed(M)
+    N, K = 640, 512
+    x = (torch.randn(M, K, device="cuda") * 0.5).to(torch.bfloat16)
+    w = (torch.randn(N, K, device="cuda") * 0.05).to(torch.bfloat16)
+    wp = w.view(N, K // 8, 8).permute(1, 0, 2).contiguous()
+    y = hip_ops().skinny_gemm(wp, x, None, N, K, _skinny_splits(N, K))
+    ref = x.float() @ w.float().t()
+    rel = (y - ref).abs().mean() / ref.abs().mean()
+    assert rel < 0.01, rel.item()
+
+
+def test_skinny_fp8_matches_quantized_reference():
+    from alpa_amd.ops import _skinny_splits
+    from alpa_amd.ops._backend import hip_ops
+    torch.manual_seed(0)
+    M, N, K = 16, 1280, 1024
+    x = (torch.randn(M, K, device="cuda") * 0.5).to(torch.bfloat16)
+    w = (torch.randn(N, K, device="cuda") * 0.05).to(torch.bfloat16)
+    amax = w.abs().amax().float().clamp_min(1e-12)
+    scale = (amax / 448.0).reshape(1)
+    q = (w.float() / scale).clamp(-448, 448).to(torch.float8_e4m3fn)
+    wp = q.view(torch.uint8).view(N, K // 8, 8).permute(1, 0, 2) \
+        .contiguous()
+    y = hip_ops().skinny_gemm(wp, x, scale, N, K, _skinny_splits(N, K))
+    # reference against the DEQUANTIZED weight (isolates kernel error
+    # from quantization error)
+    ref = x.float() @ (q.float() * scale).t()
+    rel = (y - ref).abs().mean() / ref.abs().mean()
+    assert rel < 0.01, rel.item()
+
+
+def test_skinny_layer_dispatch_and_decode_parity():
+    """ColumnParallelLinear under no_grad with M<=64 must route to the
+    skinny kernel (packed cache appears) and match the hipBLASLt path;
+    a full OPT decode step under the kernel stays close to torch."""
+    from alpa_amd.global_env import global_config
+    from alpa_amd.models.opt import OPTConfig, OPTModel
+    from alpa_amd.parallel.layers import ColumnParallelLinear
+    lin = ColumnParallelLinear(512, 640, None, 1, dtype=torch.bfloat16,
+                               device=torch.device("cuda"), init_seed=0,
+                               init_tag="t")
+    x = (torch.randn(4, 1, 512, device="cuda") * 0.5).to(torch.bfloat16)
+    with torch.no_grad():
+        y = lin(x)
+        assert hasattr(lin, "_skinny_pack")
+        global_config.skinny_gemm = False
+        try:
+            ref = lin(x)
+        finally:
+            global_config.skinny_gemm = True
+    rel = (y.float() - ref.float()).abs().mean() / ref.float().abs().mean()
+    assert rel < 0.01, rel.item()
+
+    cfg = OPTConfig(hidden_size=256, num_layers=2, num_heads=4,
+                    ffn_mult=4, vocab_size=1024, max_seq_len=64)
+    m = OPTModel(cfg, None, 1, torch.bfloat16, torch.device("cuda"),
+                 init_seed=5)
+    ids = torch.randint(0, 1024, (2, 8), device="cuda")
+    with torch.no_grad():
+        out_k = m.generate(ids, max_new_tokens=8)
+        global_config.skinny_gemm = False
+        try:
+            out_t = m.generate(ids, max_new_tokens=8)
+        finally:
+            global_config.skinny_gemm = True
+    # random-init logits are argmax-noise-sensitive; require a strong
+    # majority of identical greedy tokens
+    match = (out_k[:, 8:] == out_t[:, 8:]).float().mean().item()
+    assert match > 0.6, match
