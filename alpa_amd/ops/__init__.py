@@ -451,17 +451,23 @@ def fused_adamw(params: List[torch.Tensor], grads: List[torch.Tensor],
 
 # ----------------------------- skinny decode GEMM --------------------------
 
-def _skinny_splits(N: int, K: int) -> int:
-    """Split-K factor: enough workgroups to fill 256 CUs (grid is
-    N/64 x S) while keeping >= 8 k-rounds per workgroup."""
-    target = max(1, 1024 // max(1, N // 64))
-    kg = K // 8
-    s = 1
-    for cand in (2, 4, 8, 16, 32):
-        if cand > target or kg % cand or kg // cand < 8:
-            break
-        s = cand
-    return s
+def _skinny_splits(N: int, K: int, M: int) -> Optional[int]:
+    """Split-K factor for the decode GEMV: s must divide K/64 (keeps
+    rounds a multiple of the kernel's 8-deep pipeline), the LDS x-slice
+    (rounds * 16 * MT bytes) must fit 64 KB, and the grid (N/64 x s)
+    should fill the 256 CUs.  None = shapes unsupported."""
+    MT = 4
+    while MT < M:
+        MT *= 2
+    q = K // 64
+    lds_ok = [s for s in range(1, q + 1)
+              if q % s == 0 and (K // 8 // s) * 16 * MT <= 65536]
+    if not lds_ok:
+        return None
+    for s in lds_ok:
+        if (N // 64) * s >= 1024:
+            return s
+    return lds_ok[-1]
 
 
 def _skinny_cache(module, weight: torch.Tensor, fp8: bool):
@@ -502,7 +508,8 @@ def skinny_ok(x: torch.Tensor, weight: torch.Tensor, module=None) -> bool:
         return False
     N, K = weight.shape
     m = x.numel() // x.shape[-1]
-    return m <= 64 and N % 64 == 0 and K % 64 == 0
+    return (m <= 64 and N % 64 == 0 and K % 64 == 0
+            and _skinny_splits(N, K, m) is not None)
 
 
 def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
@@ -519,7 +526,7 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
     shape = x.shape
     x2 = x.reshape(-1, K).contiguous()
     y32 = hip_ops().skinny_gemm(wp, x2, scale, N, K,
-                                _skinny_splits(N, K))
+                                _skinny_splits(N, K, x2.shape[0]))
     y = y32.to(x.dtype)
     if bias is not None:
         y = y + bias

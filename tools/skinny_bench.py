@@ -24,7 +24,7 @@ def run(M, N, K, fp8):
         scale = None
         wp = w.view(N, K // 8, 8).permute(1, 0, 2).contiguous()
         wref = w
-    s = _skinny_splits(N, K)
+    s = _skinny_splits(N, K, M)
     y = hip_ops().skinny_gemm(wp, x, scale, N, K, s)
     ref = x.float() @ wref.float().t()
     rel = (y - ref).abs().mean() / ref.abs().mean().clamp_min(1e-9)
@@ -44,7 +44,7 @@ def run(M, N, K, fp8):
     bytes_w = N * K * (1 if fp8 else 2)
     print(f"M={M:3d} N={N:6d} K={K:6d} {'fp8' if fp8 else 'bf16'} "
           f"splits={s:2d}: rel={rel.item():.4f}  kernel {t_k:7.1f} us "
-          f"({bytes_w / t_k / 1e3:5.2f} TB/s)  torch {t_t:7.1f} us  "
+          f"({bytes_w / t_k / 1e6:5.2f} TB/s)  torch {t_t:7.1f} us  "
           f"speedup {t_t / t_k:4.2f}x")
 
 
