@@ -44,6 +44,13 @@ __device__ __forceinline__ float dot2_bf16(bf16x2 a, bf16x2 b, float acc) {
   return __builtin_amdgcn_fdot2_f32_bf16(a, b, acc, false);
 }
 
+__device__ __forceinline__ bf16x2 mk2(short a, short b) {
+  bf16x2 r;
+  r[0] = a;
+  r[1] = b;
+  return r;
+}
+
 __device__ __forceinline__ bf16x2 pair_lo(int v) {
   bf16x2 r;
   r[0] = (short)(v & 0xffff);
@@ -76,6 +83,8 @@ __device__ __forceinline__ void stage_x(const short* __restrict__ x,
 }
 
 template <int MT>
+__launch_bounds__(64, 1)  // 64-thread blocks; let big-MT tiers use the
+                          // full VGPR file instead of spilling
 __global__ void skinny_gemm_bf16_kernel(
     const short* __restrict__ wp,   // [K/8, N, 8] packed bf16
     const short* __restrict__ x,    // [MT, K] bf16 (padded rows zero)
@@ -96,22 +105,27 @@ __global__ void skinny_gemm_bf16_kernel(
     for (int u = 0; u < UNR; ++u)
       w[u] = *reinterpret_cast<const bf16x8*>(wrow + (int64_t)u * wstep);
     wrow += (int64_t)UNR * wstep;
+    // m-outer: the UNR*2 LDS reads for one m are independent — they
+    // issue back-to-back and wait ONCE (the u-outer variant paid the
+    // full ds_read latency per read); 4 accumulator chains keep the
+    // 4*UNR dot2 out of one serial dependency.
 #pragma unroll
-    for (int u = 0; u < UNR; ++u) {
-      const int r = rb + u;
-      bf16x2 w01 = {w[u][0], w[u][1]}, w23 = {w[u][2], w[u][3]};
-      bf16x2 w45 = {w[u][4], w[u][5]}, w67 = {w[u][6], w[u][7]};
+    for (int m = 0; m < MT; ++m) {
+      i32x2 a[UNR], b[UNR];
 #pragma unroll
-      for (int m = 0; m < MT; ++m) {
-        i32x2 a = xs[(r * 2 + 0) * MT + m];
-        i32x2 b = xs[(r * 2 + 1) * MT + m];
-        float s = acc[m];
-        s = dot2_bf16(w01, pair_lo(a[0]), s);
-        s = dot2_bf16(w23, pair_lo(a[1]), s);
-        s = dot2_bf16(w45, pair_lo(b[0]), s);
-        s = dot2_bf16(w67, pair_lo(b[1]), s);
-        acc[m] = s;
+      for (int u = 0; u < UNR; ++u) {
+        a[u] = xs[((rb + u) * 2 + 0) * MT + m];
+        b[u] = xs[((rb + u) * 2 + 1) * MT + m];
       }
+      float s0 = acc[m], s1 = 0.f, s2 = 0.f, s3 = 0.f;
+#pragma unroll
+      for (int u = 0; u < UNR; ++u) {
+        s0 = dot2_bf16(mk2(w[u][0], w[u][1]), pair_lo(a[u][0]), s0);
+        s1 = dot2_bf16(mk2(w[u][2], w[u][3]), pair_lo(a[u][1]), s1);
+        s2 = dot2_bf16(mk2(w[u][4], w[u][5]), pair_lo(b[u][0]), s2);
+        s3 = dot2_bf16(mk2(w[u][6], w[u][7]), pair_lo(b[u][1]), s3);
+      }
+      acc[m] = (s0 + s1) + (s2 + s3);
     }
   }
 #pragma unroll
@@ -122,6 +136,7 @@ __global__ void skinny_gemm_bf16_kernel(
 // fp8 weights: Wp[K/8][N][8] e4m3 bytes, 8-B loads; dequant scale
 // folded in once at the end (per-tensor).
 template <int MT>
+__launch_bounds__(64, 1)
 __global__ void skinny_gemm_fp8_kernel(
     const unsigned char* __restrict__ wp,  // [K/8, N, 8] packed e4m3
     const short* __restrict__ x,           // [MT, K] bf16
@@ -143,9 +158,10 @@ __global__ void skinny_gemm_fp8_kernel(
     for (int u = 0; u < UNR; ++u)
       wb[u] = *reinterpret_cast<const u8x8*>(wrow + (int64_t)u * wstep);
     wrow += (int64_t)UNR * wstep;
+    // convert the whole block's weights to bf16 pairs up front
+    bf16x8 w[UNR];
 #pragma unroll
     for (int u = 0; u < UNR; ++u) {
-      const int r = rb + u;
       unsigned int lo = (unsigned int)wb[u][0] |
                         ((unsigned int)wb[u][1] << 8) |
                         ((unsigned int)wb[u][2] << 16) |
@@ -154,27 +170,32 @@ __global__ void skinny_gemm_fp8_kernel(
                         ((unsigned int)wb[u][5] << 8) |
                         ((unsigned int)wb[u][6] << 16) |
                         ((unsigned int)wb[u][7] << 24);
-      // e4m3 -> f32 pairs -> repack as bf16 pairs so the inner loop
-      // stays on v_dot2 like the bf16 path
       f32x2_t p0 = __builtin_amdgcn_cvt_pk_f32_fp8(lo, false);
       f32x2_t p1 = __builtin_amdgcn_cvt_pk_f32_fp8(lo, true);
       f32x2_t p2 = __builtin_amdgcn_cvt_pk_f32_fp8(hi, false);
       f32x2_t p3 = __builtin_amdgcn_cvt_pk_f32_fp8(hi, true);
-      bf16x2 w01 = {f2bf(p0[0]), f2bf(p0[1])};
-      bf16x2 w23 = {f2bf(p1[0]), f2bf(p1[1])};
-      bf16x2 w45 = {f2bf(p2[0]), f2bf(p2[1])};
-      bf16x2 w67 = {f2bf(p3[0]), f2bf(p3[1])};
+      w[u][0] = f2bf(p0[0]); w[u][1] = f2bf(p0[1]);
+      w[u][2] = f2bf(p1[0]); w[u][3] = f2bf(p1[1]);
+      w[u][4] = f2bf(p2[0]); w[u][5] = f2bf(p2[1]);
+      w[u][6] = f2bf(p3[0]); w[u][7] = f2bf(p3[1]);
+    }
 #pragma unroll
-      for (int m = 0; m < MT; ++m) {
-        i32x2 a = xs[(r * 2 + 0) * MT + m];
-        i32x2 b = xs[(r * 2 + 1) * MT + m];
-        float s = acc[m];
-        s = dot2_bf16(w01, pair_lo(a[0]), s);
-        s = dot2_bf16(w23, pair_lo(a[1]), s);
-        s = dot2_bf16(w45, pair_lo(b[0]), s);
-        s = dot2_bf16(w67, pair_lo(b[1]), s);
-        acc[m] = s;
+    for (int m = 0; m < MT; ++m) {
+      i32x2 a[UNR], b[UNR];
+#pragma unroll
+      for (int u = 0; u < UNR; ++u) {
+        a[u] = xs[((rb + u) * 2 + 0) * MT + m];
+        b[u] = xs[((rb + u) * 2 + 1) * MT + m];
       }
+      float s0 = acc[m], s1 = 0.f, s2 = 0.f, s3 = 0.f;
+#pragma unroll
+      for (int u = 0; u < UNR; ++u) {
+        s0 = dot2_bf16(mk2(w[u][0], w[u][1]), pair_lo(a[u][0]), s0);
+        s1 = dot2_bf16(mk2(w[u][2], w[u][3]), pair_lo(a[u][1]), s1);
+        s2 = dot2_bf16(mk2(w[u][4], w[u][5]), pair_lo(b[u][0]), s2);
+        s3 = dot2_bf16(mk2(w[u][6], w[u][7]), pair_lo(b[u][1]), s3);
+      }
+      acc[m] = (s0 + s1) + (s2 + s3);
     }
   }
   const float ws = *wscale;

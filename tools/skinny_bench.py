@@ -24,7 +24,9 @@ def run(M, N, K, fp8):
         scale = None
         wp = w.view(N, K // 8, 8).permute(1, 0, 2).contiguous()
         wref = w
+    import os
     s = _skinny_splits(N, K, M)
+    sweep = os.environ.get("SKINNY_SWEEP") == "1"
     y = hip_ops().skinny_gemm(wp, x, scale, N, K, s)
     ref = x.float() @ wref.float().t()
     rel = (y - ref).abs().mean() / ref.abs().mean().clamp_min(1e-9)
@@ -40,6 +42,17 @@ def run(M, N, K, fp8):
         return (time.perf_counter() - t0) / iters * 1e6
 
     t_k = bench(lambda: hip_ops().skinny_gemm(wp, x, scale, N, K, s))
+    if sweep:
+        q = K // 64
+        MT = 4
+        while MT < M:
+            MT *= 2
+        for s2 in [d for d in range(1, q + 1) if q % d == 0
+                   and (K // 8 // d) * 16 * MT <= 65536]:
+            t2 = bench(lambda: hip_ops().skinny_gemm(wp, x, scale, N, K,
+                                                     s2), 100)
+            if t2 < t_k:
+                t_k, s = t2, s2
     t_t = bench(lambda: x @ w.t())
     bytes_w = N * K * (1 if fp8 else 2)
     print(f"M={M:3d} N={N:6d} K={K:6d} {'fp8' if fp8 else 'bf16'} "
