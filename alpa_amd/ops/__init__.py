@@ -465,7 +465,7 @@ def _skinny_splits(N: int, K: int, M: int) -> Optional[int]:
     if not lds_ok:
         return None
     for s in lds_ok:
-        if (N // 64) * s >= 1024:
+        if (N // 64) * s >= 2560:   # sweep-tuned (tools/skinny_bench.py)
             return s
     return lds_ok[-1]
 
@@ -496,9 +496,11 @@ def _skinny_cache(module, weight: torch.Tensor, fp8: bool):
 
 def skinny_ok(x: torch.Tensor, weight: torch.Tensor, module=None) -> bool:
     """Gate for the decode GEMV kernel (csrc/skinny_gemm.hip): inference
-    only, M <= 64 tokens, 64-aligned dims.  Measured: hipBLASLt runs
-    these shapes at ~2.8 TB/s; the packed-layout kernel streams weights
-    near the HBM roofline (tools/skinny_bench.py)."""
+    only, M <= 8 tokens, 64-aligned dims.  Measured (tools/
+    skinny_bench.py sweep): the kernel wins 1.5-3.1x at M<=4-8 (up to
+    6.2 TB/s bf16 / 4.6 TB/s-of-fp8-bytes) but the per-m VALU work makes
+    M >= 16 issue-bound (SIMD-32 units: 2 cyc/VALU instr for wave64) —
+    hipBLASLt keeps those; an MFMA 16x16x32 variant is the next step."""
     from ..global_env import global_config
     if not global_config.skinny_gemm or torch.is_grad_enabled():
         return False
@@ -508,7 +510,7 @@ def skinny_ok(x: torch.Tensor, weight: torch.Tensor, module=None) -> bool:
         return False
     N, K = weight.shape
     m = x.numel() // x.shape[-1]
-    return (m <= 64 and N % 64 == 0 and K % 64 == 0
+    return (m <= 8 and N % 64 == 0 and K % 64 == 0
             and _skinny_splits(N, K, m) is not None)
 
 

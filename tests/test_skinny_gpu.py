@@ -42,7 +42,7 @@ def test_skinny_fp8_matches_quantized_reference():
 
 
 def test_skinny_layer_dispatch_and_decode_parity():
-    """ColumnParallelLinear under no_grad with M<=64 must route to the
+    """ColumnParallelLinear under no_grad with M<=8 must route to the
     skinny kernel (packed cache appears) and match the hipBLASLt path;
     a full OPT decode step under the kernel stays close to torch."""
     from alpa_amd.global_env import global_config
