@@ -527,9 +527,7 @@ def skinny_linear(x: torch.Tensor, weight: torch.Tensor,
     N, K = weight.shape
     shape = x.shape
     x2 = x.reshape(-1, K).contiguous()
-    y32 = hip_ops().skinny_gemm(wp, x2, scale, N, K,
-                                _skinny_splits(N, K, x2.shape[0]))
-    y = y32.to(x.dtype)
-    if bias is not None:
-        y = y + bias
+    b = bias.contiguous() if bias is not None else None
+    y = hip_ops().skinny_gemm(wp, x2, scale, b, N, K,
+                              _skinny_splits(N, K, x2.shape[0]))
     return y.reshape(*shape[:-1], N)

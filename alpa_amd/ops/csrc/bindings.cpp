@@ -51,8 +51,9 @@ hipError_t launch_fp8_quantize_dual(const void*, void*, void*, const float*,
                                     float*, int64_t, int64_t, int,
                                     hipStream_t);
 hipError_t launch_skinny_gemm(const void*, const void*, float*,
-                              const float*, int64_t, int64_t, int64_t, int,
-                              int, hipStream_t);
+                              const void*, void*, const float*, int64_t,
+                              int64_t, int64_t, int64_t, int, int,
+                              hipStream_t);
 hipError_t launch_ce_fwd(const void*, const int64_t*, float*, float*, int64_t,
                          int64_t, hipStream_t);
 hipError_t launch_ce_bwd(const float*, const void*, const int64_t*,
@@ -276,18 +277,19 @@ at::Tensor colsum_bf16(const at::Tensor& t) {
 
 // ----------------------------- skinny GEMM -------------------------------
 
-// y[M,N] fp32 = x[M,K] @ Wp^T for decode-shaped M<=64; wp is the packed
-// [K/8, N, 8] layout (bf16 or e4m3 — pass wscale for e4m3).  x rows
-// beyond M must be absent (the binding pads internally).
+// y[M,N] bf16 = x[M,K] @ Wp^T (+ bias) for decode-shaped M<=64; wp is
+// the packed [K/8, N, 8] layout (bf16 or e4m3 — pass wscale for e4m3).
+// Two launches: split-K partials then a deterministic finalize
+// (reduce + bias + cast) — no zero-fill, no atomics.
 at::Tensor skinny_gemm(const at::Tensor& wp, const at::Tensor& x,
                        const c10::optional<at::Tensor>& wscale,
+                       const c10::optional<at::Tensor>& bias,
                        int64_t N, int64_t K, int64_t splits) {
   TORCH_CHECK(wp.is_cuda() && x.is_cuda() && x.is_contiguous());
   TORCH_CHECK(x.dim() == 2 && x.size(1) == K, "x must be [M, K]");
   const int64_t M = x.size(0);
   TORCH_CHECK(M >= 1 && M <= 64, "skinny path is for M <= 64");
   const bool fp8 = wscale.has_value();
-  // pad x rows up to the kernel's MT tier so the kernel needs no guards
   int64_t MT = 4;
   while (MT < M) MT *= 2;
   at::Tensor xp = x;
@@ -295,13 +297,16 @@ at::Tensor skinny_gemm(const at::Tensor& wp, const at::Tensor& x,
     xp = at::zeros({MT, K}, x.options());
     xp.narrow(0, 0, M).copy_(x);
   }
-  auto y32 = at::zeros({MT, N}, x.options().dtype(at::kFloat));
+  auto part = at::empty({splits, MT, N}, x.options().dtype(at::kFloat));
+  auto y = at::empty({M, N}, x.options());
   HIP_OK(launch_skinny_gemm(
       wp.const_data_ptr(), xp.const_data_ptr(),
-      (float*)y32.mutable_data_ptr(),
-      fp8 ? (const float*)wscale->const_data_ptr() : nullptr, M, N, K,
+      (float*)part.mutable_data_ptr(),
+      bias.has_value() ? bias->const_data_ptr() : nullptr,
+      y.mutable_data_ptr(),
+      fp8 ? (const float*)wscale->const_data_ptr() : nullptr, M, MT, N, K,
       (int)splits, fp8 ? 1 : 0, cur_stream()));
-  return MT == M ? y32 : y32.narrow(0, 0, M);
+  return y;
 }
 
 // -------------------------------- AdamW ----------------------------------

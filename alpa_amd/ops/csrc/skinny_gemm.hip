@@ -22,8 +22,13 @@
 //     one load outstanding and ran at HBM latency (374 ns/round), not
 //     bandwidth.
 //   - Each lane accumulates acc[m] (<= MT VGPRs) for ITS row — no
-//     cross-lane reduction.  Split-K workgroups atomicAdd fp32 partials
-//     into y32[M,N]; order nondeterministic — inference only.
+//     cross-lane reduction.  Split-K workgroups STORE fp32 partials to
+//     part[S, MT, N] (no zeroing, no atomics) and a finalize kernel
+//     reduces the S slices + adds bias + casts to bf16 in fixed order —
+//     2 launches total and deterministic (the atomicAdd variant needed
+//     an at::zeros + cast + bias_add per call: ~3 extra launches per
+//     linear, which erased the GEMV win at 320 linears/decode-token on
+//     OPT-66B).
 //   - fp8 (e4m3) weight variant reads HALF the bytes and dequantizes
 //     in-kernel via the packed hardware converter; x stays bf16.
 //
@@ -88,7 +93,7 @@ __launch_bounds__(64, 1)  // 64-thread blocks; let big-MT tiers use the
 __global__ void skinny_gemm_bf16_kernel(
     const short* __restrict__ wp,   // [K/8, N, 8] packed bf16
     const short* __restrict__ x,    // [MT, K] bf16 (padded rows zero)
-    float* __restrict__ y32,        // [MT, N] fp32 (zeroed)
+    float* __restrict__ part,       // [S, MT, N] fp32 (overwritten)
     int64_t N, int64_t K, int rounds) {
   extern __shared__ i32x2 xs[];
   const int64_t row = (int64_t)blockIdx.x * 64 + threadIdx.x;
@@ -128,9 +133,9 @@ __global__ void skinny_gemm_bf16_kernel(
       acc[m] = (s0 + s1) + (s2 + s3);
     }
   }
+  float* out = part + ((int64_t)blockIdx.y * MT) * N + row;
 #pragma unroll
-  for (int m = 0; m < MT; ++m)
-    atomicAdd(y32 + (int64_t)m * N + row, acc[m]);
+  for (int m = 0; m < MT; ++m) out[(int64_t)m * N] = acc[m];
 }
 
 // fp8 weights: Wp[K/8][N][8] e4m3 bytes, 8-B loads; dequant scale
@@ -140,7 +145,7 @@ __launch_bounds__(64, 1)
 __global__ void skinny_gemm_fp8_kernel(
     const unsigned char* __restrict__ wp,  // [K/8, N, 8] packed e4m3
     const short* __restrict__ x,           // [MT, K] bf16
-    float* __restrict__ y32,               // [MT, N] fp32 (zeroed)
+    float* __restrict__ part,              // [S, MT, N] fp32 (overwritten)
     const float* __restrict__ wscale,      // [1]
     int64_t N, int64_t K, int rounds) {
   extern __shared__ i32x2 xs[];
@@ -199,41 +204,60 @@ __global__ void skinny_gemm_fp8_kernel(
     }
   }
   const float ws = *wscale;
+  float* out = part + ((int64_t)blockIdx.y * MT) * N + row;
 #pragma unroll
-  for (int m = 0; m < MT; ++m)
-    atomicAdd(y32 + (int64_t)m * N + row, acc[m] * ws);
+  for (int m = 0; m < MT; ++m) out[(int64_t)m * N] = acc[m] * ws;
+}
+
+// Reduce the S split slices in fixed order, add bias, cast to bf16.
+__global__ void skinny_finalize_kernel(const float* __restrict__ part,
+                                       const short* __restrict__ bias,
+                                       short* __restrict__ y, int64_t MT,
+                                       int64_t M, int64_t N, int splits) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= M * N) return;
+  const int64_t m = i / N, n = i - m * N;
+  float s = 0.f;
+  for (int k = 0; k < splits; ++k) s += part[((int64_t)k * MT + m) * N + n];
+  if (bias) s += bf2f(bias[n]);
+  y[i] = f2bf(s);
 }
 
 extern "C" {
 
 // dispatch over padded M tier; returns hipErrorInvalidValue on bad dims
-hipError_t launch_skinny_gemm(const void* wp, const void* x, float* y32,
+hipError_t launch_skinny_gemm(const void* wp, const void* x, float* part,
+                              const void* bias_or_null, void* y,
                               const float* wscale_or_null, int64_t M,
-                              int64_t N, int64_t K, int splits, int fp8,
-                              hipStream_t stream) {
+                              int64_t MT, int64_t N, int64_t K, int splits,
+                              int fp8, hipStream_t stream) {
   if (N % 64 || K % 8 || (K / 8) % splits) return hipErrorInvalidValue;
   int rounds = (int)(K / 8 / splits);
   if (rounds % UNR) return hipErrorInvalidValue;
   dim3 grid((uint32_t)(N / 64), (uint32_t)splits);
   dim3 blk(64);
-#define DISPATCH(MT)                                                       \
+#define DISPATCH(MTC)                                                      \
   do {                                                                     \
-    size_t lds = (size_t)rounds * 2 * MT * sizeof(i32x2);                  \
+    size_t lds = (size_t)rounds * 2 * MTC * sizeof(i32x2);                 \
     if (fp8)                                                               \
-      skinny_gemm_fp8_kernel<MT><<<grid, blk, lds, stream>>>(              \
-          (const unsigned char*)wp, (const short*)x, y32,                  \
+      skinny_gemm_fp8_kernel<MTC><<<grid, blk, lds, stream>>>(             \
+          (const unsigned char*)wp, (const short*)x, part,                 \
           wscale_or_null, N, K, rounds);                                   \
     else                                                                   \
-      skinny_gemm_bf16_kernel<MT><<<grid, blk, lds, stream>>>(             \
-          (const short*)wp, (const short*)x, y32, N, K, rounds);           \
+      skinny_gemm_bf16_kernel<MTC><<<grid, blk, lds, stream>>>(            \
+          (const short*)wp, (const short*)x, part, N, K, rounds);          \
   } while (0)
-  if (M <= 4) DISPATCH(4);
-  else if (M <= 8) DISPATCH(8);
-  else if (M <= 16) DISPATCH(16);
-  else if (M <= 32) DISPATCH(32);
-  else if (M <= 64) DISPATCH(64);
+  if (MT == 4) DISPATCH(4);
+  else if (MT == 8) DISPATCH(8);
+  else if (MT == 16) DISPATCH(16);
+  else if (MT == 32) DISPATCH(32);
+  else if (MT == 64) DISPATCH(64);
   else return hipErrorInvalidValue;
 #undef DISPATCH
+  int64_t total = M * N;
+  int blocks = (int)((total + 255) / 256);
+  skinny_finalize_kernel<<<dim3(blocks), dim3(256), 0, stream>>>(
+      part, (const short*)bias_or_null, (short*)y, MT, M, N, splits);
   return hipGetLastError();
 }
 

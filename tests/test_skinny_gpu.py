@@ -15,7 +15,8 @@ def test_skinny_bf16_matches_reference(M):
     x = (torch.randn(M, K, device="cuda") * 0.5).to(torch.bfloat16)
     w = (torch.randn(N, K, device="cuda") * 0.05).to(torch.bfloat16)
     wp = w.view(N, K // 8, 8).permute(1, 0, 2).contiguous()
-    y = hip_ops().skinny_gemm(wp, x, None, N, K, _skinny_splits(N, K, M))
+    y = hip_ops().skinny_gemm(wp, x, None, None, N, K,
+                              _skinny_splits(N, K, M)).float()
     ref = x.float() @ w.float().t()
     rel = (y - ref).abs().mean() / ref.abs().mean()
     assert rel < 0.01, rel.item()
@@ -33,7 +34,7 @@ def test_skinny_fp8_matches_quantized_reference():
     q = (w.float() / scale).clamp(-448, 448).to(torch.float8_e4m3fn)
     wp = q.view(torch.uint8).view(N, K // 8, 8).permute(1, 0, 2) \
         .contiguous()
-    y = hip_ops().skinny_gemm(wp, x, scale, N, K, _skinny_splits(N, K, M))
+    y = hip_ops().skinny_gemm(wp, x, scale, None, N, K, _skinny_splits(N, K, M))
     # reference against the DEQUANTIZED weight (isolates kernel error
     # from quantization error)
     ref = x.float() @ (q.float() * scale).t()

@@ -27,7 +27,7 @@ def run(M, N, K, fp8):
     import os
     s = _skinny_splits(N, K, M)
     sweep = os.environ.get("SKINNY_SWEEP") == "1"
-    y = hip_ops().skinny_gemm(wp, x, scale, N, K, s)
+    y = hip_ops().skinny_gemm(wp, x, scale, None, N, K, s)
     ref = x.float() @ wref.float().t()
     rel = (y - ref).abs().mean() / ref.abs().mean().clamp_min(1e-9)
 
@@ -41,7 +41,7 @@ def run(M, N, K, fp8):
         torch.cuda.synchronize()
         return (time.perf_counter() - t0) / iters * 1e6
 
-    t_k = bench(lambda: hip_ops().skinny_gemm(wp, x, scale, N, K, s))
+    t_k = bench(lambda: hip_ops().skinny_gemm(wp, x, scale, None, N, K, s))
     if sweep:
         q = K // 64
         MT = 4
@@ -49,7 +49,7 @@ def run(M, N, K, fp8):
             MT *= 2
         for s2 in [d for d in range(1, q + 1) if q % d == 0
                    and (K // 8 // d) * 16 * MT <= 65536]:
-            t2 = bench(lambda: hip_ops().skinny_gemm(wp, x, scale, N, K,
+            t2 = bench(lambda: hip_ops().skinny_gemm(wp, x, scale, None, N, K,
                                                      s2), 100)
             if t2 < t_k:
                 t_k, s = t2, s2
