@@ -451,6 +451,10 @@ def fused_adamw(params: List[torch.Tensor], grads: List[torch.Tensor],
 
 # ----------------------------- skinny decode GEMM --------------------------
 
+import functools
+
+
+@functools.lru_cache(maxsize=None)
 def _skinny_splits(N: int, K: int, M: int) -> Optional[int]:
     """Split-K factor for the decode GEMV: s must divide K/64 (keeps
     rounds a multiple of the kernel's 8-deep pipeline), the LDS x-slice
