@@ -40,11 +40,16 @@ class GlobalConfig:
         default_factory=lambda: os.environ.get("ALPA_AMD_FP8_WGRAD",
                                                "1") == "1")
     #: hand-written decode GEMV (ops/csrc/skinny_gemm.hip) for
-    #: inference GEMMs with <= 64 tokens; ALPA_AMD_SKINNY=0 falls back
-    #: to hipBLASLt
+    #: inference GEMMs with <= 8 tokens.  OFF by default: the kernel
+    #: beats hipBLASLt 1.5-3.1x in isolation (tools/skinny_bench.py)
+    #: but eager decode is HOST-bound (~23 us/op measured on OPT-66B:
+    #: same box, skinny off 28.1 -> on 33.8 ms/token) and the path adds
+    #: ~2 launches + allocations per linear.  Becomes profitable once
+    #: the decode step is hipGraph-captured (r3).  ALPA_AMD_SKINNY=1
+    #: enables.
     skinny_gemm: bool = field(
         default_factory=lambda: os.environ.get("ALPA_AMD_SKINNY",
-                                               "1") == "1")
+                                               "0") == "1")
 
     #: compute dX with the bf16 MASTER weight instead of the fp8 wqt
     #: cache: halves the fp8 weight-cache footprint (the +30 GB dual
