@@ -510,7 +510,11 @@ def skinny_ok(x: torch.Tensor, weight: torch.Tensor, module=None) -> bool:
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return False
-    if torch.cuda.is_current_stream_capturing():
+    if torch.cuda.is_current_stream_capturing() and \
+            getattr(module, "_skinny_pack", None) is None:
+        # packing mustn't be captured into the graph; the warmup pass
+        # before capture builds the cache, after which replay just
+        # re-reads the packed weights in place
         return False
     N, K = weight.shape
     m = x.numel() // x.shape[-1]

@@ -46,6 +46,10 @@ def main():
                         "(lm_head stays bf16): halves the weight bytes "
                         "read per token — decode is weight-bandwidth-"
                         "bound, so this is the serving fp8 lever")
+    p.add_argument("--graph", action="store_true",
+                   help="hipGraph-captured decode step (OPT only): one "
+                        "graph replay per token instead of ~15 eager op "
+                        "dispatches per layer")
     p.add_argument("--check", action="store_true",
                    help="with --fp8: also greedy-decode in bf16 from "
                         "the same prompts and report the token match "
@@ -106,21 +110,35 @@ def main():
     prefill_ms = (time.perf_counter() - t0) * 1e3
 
     # decode timing
-    tok = model.greedy_token(logits).unsqueeze(1)
-    sync()
-    t0 = time.perf_counter()
-    with torch.no_grad():
-        for _ in range(args.gen):
-            logits = model.forward_step(tok, cache)
-            tok = model.greedy_token(logits).unsqueeze(1)
-    sync()
-    decode_s = time.perf_counter() - t0
+    if args.graph:
+        # build (prefill + one-time capture) outside the timed region,
+        # then time the replay loop only
+        with torch.no_grad():
+            gd = model.graphed_decoder(ids, max_new_tokens=args.gen)
+        sync()
+        t0 = time.perf_counter()
+        gd.run()
+        sync()
+        decode_s = time.perf_counter() - t0
+    else:
+        tok = model.greedy_token(logits).unsqueeze(1)
+        sync()
+        t0 = time.perf_counter()
+        with torch.no_grad():
+            for _ in range(args.gen):
+                logits = model.forward_step(tok, cache)
+                tok = model.greedy_token(logits).unsqueeze(1)
+        sync()
+        decode_s = time.perf_counter() - t0
 
     if args.beams > 1:
         with torch.no_grad():
             model.beam_search(ids, max_new_tokens=8, num_beams=args.beams)
         sync()
 
+    # graphed decode emits gen-1 tokens in the timed replay loop (the
+    # first token comes from prefill)
+    n_dec = (args.gen - 1) if args.graph else args.gen
     if aa.rank() == 0:
         print(json.dumps({
             "model": f"{args.family}-{args.model}", "batch": args.batch,
@@ -128,12 +146,13 @@ def main():
             "gen_tokens": args.gen,
             "dtype": ("bf16+fp8gemm" if args.fp8
                       else str(dtype).split(".")[-1]),
+            "graphed_decode": bool(args.graph),
             **({"fp8_vs_bf16_token_match": round(match_rate, 4),
                 "fp8_logit_rel_err": round(logit_rel, 5)}
                if match_rate is not None else {}),
             "prefill_ms": round(prefill_ms, 2),
-            "decode_ms_per_token": round(decode_s / args.gen * 1e3, 3),
-            "decode_tokens_per_s": round(args.batch * args.gen / decode_s,
+            "decode_ms_per_token": round(decode_s / n_dec * 1e3, 3),
+            "decode_tokens_per_s": round(args.batch * n_dec / decode_s,
                                          1),
             "data": "synthetic/random-init",
         }))
