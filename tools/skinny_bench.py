@@ -61,8 +61,15 @@ def run(M, N, K, fp8):
           f"speedup {t_t / t_k:4.2f}x")
 
 
+import os as _os
+
+_shapes = _os.environ.get("SKINNY_SHAPES")
+SHAPES = ([tuple(map(int, t.split("x"))) for t in _shapes.split(",")]
+          if _shapes else [(5120, 5120), (15360, 5120), (20480, 5120),
+                           (5120, 20480)])
+MS = tuple(int(m) for m in
+           _os.environ.get("SKINNY_MS", "4,16,64").split(","))
 for fp8 in (False, True):
-    for (N, K) in ((5120, 5120), (15360, 5120), (20480, 5120),
-                   (5120, 20480)):
-        for M in (4, 16, 64):
+    for (N, K) in SHAPES:
+        for M in MS:
             run(M, N, K, fp8)
