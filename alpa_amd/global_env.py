@@ -40,16 +40,16 @@ class GlobalConfig:
         default_factory=lambda: os.environ.get("ALPA_AMD_FP8_WGRAD",
                                                "1") == "1")
     #: hand-written decode GEMV (ops/csrc/skinny_gemm.hip) for
-    #: inference GEMMs with <= 8 tokens.  OFF by default: the kernel
-    #: beats hipBLASLt 1.5-3.1x in isolation (tools/skinny_bench.py)
-    #: but eager decode is HOST-bound (~23 us/op measured on OPT-66B:
-    #: same box, skinny off 28.1 -> on 33.8 ms/token) and the path adds
-    #: ~2 launches + allocations per linear.  Becomes profitable once
-    #: the decode step is hipGraph-captured (r3).  ALPA_AMD_SKINNY=1
-    #: enables.
-    skinny_gemm: bool = field(
+    #: inference GEMMs with <= 8 tokens.  "auto" (default) enables it
+    #: only inside the MEASURED end-to-end win envelope — hidden-5120-
+    #: class shapes (K <= 5120 or the matching 4x ffn K), where OPT-13B
+    #: decode drops 11.85 -> 8.58 ms/token bf16 / 7.70 fp8.  hipBLASLt
+    #: already streams >= 5 TB/s at hidden >= 7168 (30B/66B measured
+    #: parity-to-loss end-to-end), so those stay on the library.
+    #: ALPA_AMD_SKINNY=1 forces it everywhere, =0 disables.
+    skinny_gemm: str = field(
         default_factory=lambda: os.environ.get("ALPA_AMD_SKINNY",
-                                               "0") == "1")
+                                               "auto"))
 
     #: compute dX with the bf16 MASTER weight instead of the fp8 wqt
     #: cache: halves the fp8 weight-cache footprint (the +30 GB dual

@@ -468,8 +468,14 @@ def _skinny_splits(N: int, K: int, M: int) -> Optional[int]:
               if q % s == 0 and (K // 8 // s) * 16 * MT <= 65536]
     if not lds_ok:
         return None
+    # sweep-tuned (tools/skinny_bench.py): ~16 k-rounds per workgroup is
+    # the sweet spot — take the LARGEST split with rounds >= 16 that
+    # still fills the chip, else the most-parallel legal config
+    for s in sorted(lds_ok, reverse=True):
+        if K // 8 // s >= 16 and (N // 64) * s >= 1024:
+            return s
     for s in lds_ok:
-        if (N // 64) * s >= 2560:   # sweep-tuned (tools/skinny_bench.py)
+        if (N // 64) * s >= 2560:
             return s
     return lds_ok[-1]
 
@@ -506,7 +512,8 @@ def skinny_ok(x: torch.Tensor, weight: torch.Tensor, module=None) -> bool:
     M >= 16 issue-bound (SIMD-32 units: 2 cyc/VALU instr for wave64) —
     hipBLASLt keeps those; an MFMA 16x16x32 variant is the next step."""
     from ..global_env import global_config
-    if not global_config.skinny_gemm or torch.is_grad_enabled():
+    mode = global_config.skinny_gemm
+    if mode in ("0", False) or torch.is_grad_enabled():
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16):
         return False
@@ -517,6 +524,9 @@ def skinny_ok(x: torch.Tensor, weight: torch.Tensor, module=None) -> bool:
         # re-reads the packed weights in place
         return False
     N, K = weight.shape
+    if mode == "auto" and (K > 6144 and N > 6144):
+        # outside the measured win envelope (see global_env.skinny_gemm)
+        return False
     m = x.numel() // x.shape[-1]
     return (m <= 8 and N % 64 == 0 and K % 64 == 0
             and _skinny_splits(N, K, m) is not None)
