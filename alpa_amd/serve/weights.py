@@ -164,3 +164,73 @@ def load_bloom_hf(src, mesh: Optional[DeviceMesh] = None, axis: int = 1,
                                    tp, idx))
         _set(blk.fc2.bias, sd[p + "mlp.dense_4h_to_h.bias"])
     return model
+
+
+# ------------------------- fp8 serving checkpoints -------------------------
+
+E4M3_MAX = 448.0
+
+
+def save_fp8_checkpoint(model: torch.nn.Module, path: str) -> None:
+    """Serving-weight checkpoint with every shardable linear weight
+    stored as OCP e4m3 bytes + per-output-row fp32 scales (reference
+    analog: the
+    llm_serving weight-conversion scripts, ``scripts/step_*.py`` —
+    here the conversion is a first-class format).  Halves the weight
+    file; modules marked ``_fp8_exclude`` (lm_head) and every
+    non-linear parameter stay at full precision.
+
+    The file holds THIS RANK's (possibly TP-sharded) state — save and
+    load under the same tp degree (metadata-checked).
+    """
+    from ..mesh import world_size
+    blob = {"format": "alpa_amd-fp8-v1", "tp": world_size(), "tensors": {}}
+    quant_owner = {}
+    for mname, mod in model.named_modules():
+        w = getattr(mod, "weight", None)
+        if (isinstance(w, torch.nn.Parameter) and w.dim() == 2
+                and not getattr(mod, "_fp8_exclude", False)
+                and type(mod).__name__.endswith("ParallelLinear")):
+            quant_owner[f"{mname}.weight" if mname else "weight"] = True
+    for name, p in model.named_parameters():
+        t = p.detach().cpu()
+        if quant_owner.get(name):
+            # per-output-row scales: much tighter than per-tensor for a
+            # storage format (the runtime re-quantizes per-tensor with
+            # its own delayed scaling anyway)
+            scale = (t.abs().amax(dim=1, keepdim=True).float()
+                     .clamp_min(1e-12) / E4M3_MAX)
+            q = (t.float() / scale).clamp(-E4M3_MAX, E4M3_MAX).to(
+                torch.float8_e4m3fn)
+            blob["tensors"][name] = {"q": q.view(torch.uint8),
+                                     "scale": scale,
+                                     "shape": list(t.shape)}
+        else:
+            blob["tensors"][name] = {"raw": t}
+    torch.save(blob, path)
+
+
+def load_fp8_checkpoint(model: torch.nn.Module, path: str) -> None:
+    """Load a ``save_fp8_checkpoint`` file: e4m3 weights dequantize into
+    the module dtype (re-quantization by the fp8/skinny serving paths is
+    idempotent under the same per-tensor scale)."""
+    from ..mesh import world_size
+    blob = torch.load(path, map_location="cpu", weights_only=False)
+    assert blob.get("format") == "alpa_amd-fp8-v1", "not an fp8 checkpoint"
+    assert blob["tp"] == world_size(), (
+        f"checkpoint saved at tp={blob['tp']}, loading at "
+        f"tp={world_size()}")
+    params = dict(model.named_parameters())
+    missing = set(params) - set(blob["tensors"])
+    assert not missing, f"checkpoint missing params: {sorted(missing)[:5]}"
+    with torch.no_grad():
+        for name, entry in blob["tensors"].items():
+            p = params.get(name)
+            if p is None:
+                continue
+            if "raw" in entry:
+                p.copy_(entry["raw"].to(p.dtype))
+            else:
+                q = entry["q"].view(torch.float8_e4m3fn)
+                p.copy_((q.float() * entry["scale"]).reshape(
+                    entry["shape"]).to(p.dtype))
